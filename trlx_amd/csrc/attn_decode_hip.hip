@@ -1,0 +1,151 @@
+#include "hip/hip_runtime.h"
+// Fused decode attention — the generation hot kernel.
+//
+// q [B, Hq, 1, D] vs KV cache [B, Hkv, S, D] (bf16, contiguous, GQA-aware),
+// per-row valid lengths.  Memory-bound flash-decode: online softmax, zero
+// [B, H, S] score materialization, KV loads vectorized at 16 B/lane (guide
+// Guideline 13: 64 lanes x 8 bf16 = one 1 KiB transaction covering 512/D keys
+// per instruction).  One 256-thread block per (b, h); the 4 waves x
+// (512/D key-groups) partial accumulators are merged through LDS.
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int NWAVES = BLOCK / WAVE;
+
+template <int D>
+__global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ kc,
+                                   const bf16_t* __restrict__ vc, const int* __restrict__ seq_lens,
+                                   bf16_t* __restrict__ out, int Hq, int Hkv, int S, float scale) {
+  constexpr int G = D / 8;          // lanes cooperating on one key
+  constexpr int KPW = WAVE / G;     // keys per wave per iteration
+  constexpr int NPART = NWAVES * KPW;  // independent (m, s, o) partials
+
+  const int b = blockIdx.x / Hq;
+  const int hq = blockIdx.x % Hq;
+  const int hkv = hq / (Hq / Hkv);
+  const int len = seq_lens[b];
+
+  const int lane = threadIdx.x % WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int kgrp = lane / G;        // which key within the wave's group
+  const int d0 = (lane % G) * 8;    // this lane's slice of D
+
+  const bf16_t* qp = q + ((size_t)b * Hq + hq) * D;
+  const bf16_t* kbase = kc + ((size_t)b * Hkv + hkv) * S * D;
+  const bf16_t* vbase = vc + ((size_t)b * Hkv + hkv) * S * D;
+
+  // q fragment for this lane's d-slice
+  float qf[8];
+  load8<bf16_t>(qp + d0, qf);
+#pragma unroll
+  for (int i = 0; i < 8; ++i) qf[i] *= scale;
+
+  float m = -INFINITY, s = 0.f;
+  float o[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) o[i] = 0.f;
+
+  for (int sbase = wid * KPW; sbase < len; sbase += NWAVES * KPW) {
+    const int key = sbase + kgrp;
+    float score = -INFINITY;
+    float vf[8];
+    if (key < len) {
+      float kf[8];
+      load8<bf16_t>(kbase + (size_t)key * D + d0, kf);
+      load8<bf16_t>(vbase + (size_t)key * D + d0, vf);
+      float partial = 0.f;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) partial += qf[i] * kf[i];
+      // reduce across the G lanes of this key group (contiguous lanes)
+#pragma unroll
+      for (int off = G / 2; off > 0; off >>= 1) partial += __shfl_xor(partial, off);
+      score = partial;
+    }
+    // online softmax update for this group's accumulator
+    if (score > m) {
+      const float corr = expf(m - score);
+      s = s * corr + 1.f;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) o[i] = o[i] * corr + vf[i];
+      m = score;
+    } else if (score != -INFINITY) {
+      const float p = expf(score - m);
+      s += p;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) o[i] += p * vf[i];
+    }
+  }
+
+  // merge the NPART partial accumulators via LDS
+  __shared__ float ms_buf[NPART][2];
+  __shared__ float o_buf[NPART][D];
+  const int part = wid * KPW + kgrp;
+  if (lane % G == 0) {
+    ms_buf[part][0] = m;
+    ms_buf[part][1] = s;
+  }
+#pragma unroll
+  for (int i = 0; i < 8; ++i) o_buf[part][d0 + i] = o[i];
+  __syncthreads();
+
+  // final combine: thread t handles output dim t (D <= BLOCK)
+  if (threadIdx.x < D) {
+    float mstar = -INFINITY;
+#pragma unroll
+    for (int p = 0; p < NPART; ++p) mstar = fmaxf(mstar, ms_buf[p][0]);
+    float sstar = 0.f, acc = 0.f;
+#pragma unroll
+    for (int p = 0; p < NPART; ++p) {
+      const float mp = ms_buf[p][0];
+      if (mp == -INFINITY) continue;
+      const float w = expf(mp - mstar);
+      sstar += ms_buf[p][1] * w;
+      acc += o_buf[p][threadIdx.x] * w;
+    }
+    const float res = (sstar > 0.f) ? acc / sstar : 0.f;
+    bf16_t* op = out + ((size_t)b * Hq + hq) * D;
+    op[threadIdx.x].u = f2bf(res);
+  }
+}
+
+}  // namespace
+
+at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
+                            const at::Tensor& seq_lens, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16 && q.is_contiguous());
+  TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous());
+  TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attention_decode: q must be [B,Hq,1,D]");
+  const int B = q.size(0), Hq = q.size(1), D = q.size(3);
+  const int Hkv = kc.size(1), S = kc.size(2);
+  TORCH_CHECK(kc.size(3) == D && Hq % Hkv == 0);
+  TORCH_CHECK(seq_lens.dtype() == at::kInt && seq_lens.numel() == B);
+  auto out = at::empty_like(q);
+  if (B == 0) return out;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int grid = B * Hq;
+  auto qp = reinterpret_cast<const bf16_t*>(q.data_ptr());
+  auto kp = reinterpret_cast<const bf16_t*>(kc.data_ptr());
+  auto vp = reinterpret_cast<const bf16_t*>(vc.data_ptr());
+  auto op = reinterpret_cast<bf16_t*>(out.data_ptr());
+  auto sl = seq_lens.data_ptr<int>();
+  switch (D) {
+    case 64:
+     hipLaunchKernelGGL(( attn_decode_kernel<64>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, op, Hq, Hkv, S, (float)scale);
+      break;
+    case 128:
+     hipLaunchKernelGGL(( attn_decode_kernel<128>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, op, Hq, Hkv, S, (float)scale);
+      break;
+    case 256:
+     hipLaunchKernelGGL(( attn_decode_kernel<256>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, op, Hq, Hkv, S, (float)scale);
+      break;
+    default:
+      TORCH_CHECK(false, "attention_decode: head dim must be 64/128/256, got ", D);
+  }
+  HIP_CHECK_LAST();
+  return out;
+}

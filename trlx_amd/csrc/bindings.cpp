@@ -1,0 +1,50 @@
+// Python bindings for the trlx_amd gfx950 kernels (trlx_amd._C).
+#include <torch/extension.h>
+
+std::vector<at::Tensor> logprobs_fwd(const at::Tensor& logits, const at::Tensor& labels);
+at::Tensor logprobs_bwd(const at::Tensor& logits, const at::Tensor& labels, const at::Tensor& lse,
+                        const at::Tensor& gout);
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps);
+std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
+                                    const at::Tensor& invr, const at::Tensor& dy);
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& w,
+                                      const c10::optional<at::Tensor>& b, double eps);
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x, const at::Tensor& w,
+                                      const at::Tensor& mean, const at::Tensor& invstd,
+                                      const at::Tensor& dy);
+at::Tensor rope_fwd(const at::Tensor& x, const at::Tensor& cos, const at::Tensor& sin,
+                    const at::Tensor& pos, bool interleaved, bool inverse, long rot);
+std::vector<at::Tensor> gae(const at::Tensor& values, const at::Tensor& rewards, double gamma,
+                            double lam);
+at::Tensor sum_count(const at::Tensor& x);
+at::Tensor normalize(const at::Tensor& x, const at::Tensor& mean, const at::Tensor& var,
+                     bool shift_mean);
+at::Tensor gumbel_sample(const at::Tensor& logits, double temperature,
+                         const c10::optional<at::Tensor>& thresholds, long seed, long offset);
+at::Tensor causal_softmax_fwd(const at::Tensor& scores, long start_pos,
+                              const c10::optional<at::Tensor>& key_starts);
+at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs);
+at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
+                            const at::Tensor& seq_lens, double scale,
+                            const c10::optional<at::Tensor>& seq_starts);
+void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
+                 at::Tensor& v, long step, double lr, double beta1, double beta2, double eps,
+                 double weight_decay, double grad_scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("logprobs_fwd", &logprobs_fwd, "fused logsumexp + label gather fwd");
+  mod.def("logprobs_bwd", &logprobs_bwd, "logprobs backward");
+  mod.def("rmsnorm_fwd", &rmsnorm_fwd);
+  mod.def("rmsnorm_bwd", &rmsnorm_bwd);
+  mod.def("layernorm_fwd", &layernorm_fwd);
+  mod.def("layernorm_bwd", &layernorm_bwd);
+  mod.def("rope_fwd", &rope_fwd);
+  mod.def("gae", &gae);
+  mod.def("sum_count", &sum_count);
+  mod.def("normalize", &normalize);
+  mod.def("gumbel_sample", &gumbel_sample);
+  mod.def("causal_softmax_fwd", &causal_softmax_fwd);
+  mod.def("causal_softmax_bwd", &causal_softmax_bwd);
+  mod.def("attention_decode", &attention_decode);
+  mod.def("fused_adamw", &fused_adamw);
+}

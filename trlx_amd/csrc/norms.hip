@@ -1,0 +1,359 @@
+// Fused RMSNorm / LayerNorm fwd+bwd.
+//
+// One block per row (grid-strided), vectorized 16 B/lane bf16 loads (guide
+// Guideline 13), fp32 accumulation, saved row statistics for backward.
+// Weight-grad accumulation goes through an LDS per-block partial (each column
+// owned by exactly one thread, race-free) followed by one global atomicAdd per
+// column per block — not per row.
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int NWAVES = BLOCK / WAVE;
+
+template <typename T>
+__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                   T* __restrict__ y, float* __restrict__ invr, int H, float eps,
+                                   long N) {
+  __shared__ float rbuf[NWAVES];
+  const int H8 = H & ~7;
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + (size_t)row * H;
+    T* yr = y + (size_t)row * H;
+    float ss = 0.f;
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float v[8];
+      load8<T>(xr + base, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) ss += v[i] * v[i];
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      float xi = ScalarIO<T>::load(xr + i);
+      ss += xi * xi;
+    }
+    ss = block_sum<NWAVES>(ss, rbuf);
+    const float r = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0) invr[row] = r;
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float v[8], wv[8];
+      load8<T>(xr + base, v);
+      load8<T>(w + base, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) v[i] = v[i] * r * wv[i];
+      store8<T>(yr + base, v);
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      ScalarIO<T>::store(yr + i, ScalarIO<T>::load(xr + i) * r * ScalarIO<T>::load(w + i));
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                   const float* __restrict__ invr, const T* __restrict__ dy,
+                                   T* __restrict__ dx, float* __restrict__ dw, int H, long N) {
+  extern __shared__ float dwacc[];  // H floats
+  __shared__ float rbuf[NWAVES];
+  for (int i = threadIdx.x; i < H; i += BLOCK) dwacc[i] = 0.f;
+  __syncthreads();
+  const int H8 = H & ~7;
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + (size_t)row * H;
+    const T* dyr = dy + (size_t)row * H;
+    T* dxr = dx + (size_t)row * H;
+    const float r = invr[row];
+    float c = 0.f;
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float xv[8], dv[8], wv[8];
+      load8<T>(xr + base, xv);
+      load8<T>(dyr + base, dv);
+      load8<T>(w + base, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) c += dv[i] * wv[i] * xv[i];
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK)
+      c += ScalarIO<T>::load(dyr + i) * ScalarIO<T>::load(w + i) * ScalarIO<T>::load(xr + i);
+    c = block_sum<NWAVES>(c, rbuf);
+    const float k = c * r * r * r / H;
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float xv[8], dv[8], wv[8], o[8];
+      load8<T>(xr + base, xv);
+      load8<T>(dyr + base, dv);
+      load8<T>(w + base, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        o[i] = r * wv[i] * dv[i] - k * xv[i];
+        dwacc[base + i] += dv[i] * xv[i] * r;
+      }
+      store8<T>(dxr + base, o);
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      float xv = ScalarIO<T>::load(xr + i);
+      float dv = ScalarIO<T>::load(dyr + i);
+      ScalarIO<T>::store(dxr + i, r * ScalarIO<T>::load(w + i) * dv - k * xv);
+      dwacc[i] += dv * xv * r;
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < H; i += BLOCK) atomicAdd(&dw[i], dwacc[i]);
+}
+
+template <typename T, bool HAS_BIAS>
+__global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                     const T* __restrict__ b, T* __restrict__ y,
+                                     float* __restrict__ mean, float* __restrict__ invstd, int H,
+                                     float eps, long N) {
+  __shared__ float rbuf[NWAVES];
+  const int H8 = H & ~7;
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + (size_t)row * H;
+    T* yr = y + (size_t)row * H;
+    float s = 0.f, ss = 0.f;
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float v[8];
+      load8<T>(xr + base, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        s += v[i];
+        ss += v[i] * v[i];
+      }
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      float xi = ScalarIO<T>::load(xr + i);
+      s += xi;
+      ss += xi * xi;
+    }
+    s = block_sum<NWAVES>(s, rbuf);
+    ss = block_sum<NWAVES>(ss, rbuf);
+    const float mu = s / H;
+    const float var = fmaxf(ss / H - mu * mu, 0.f);
+    const float istd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean[row] = mu;
+      invstd[row] = istd;
+    }
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float v[8], wv[8], bv[8];
+      load8<T>(xr + base, v);
+      load8<T>(w + base, wv);
+      if (HAS_BIAS) load8<T>(b + base, bv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float o = (v[i] - mu) * istd * wv[i];
+        if (HAS_BIAS) o += bv[i];
+        v[i] = o;
+      }
+      store8<T>(yr + base, v);
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      float o = (ScalarIO<T>::load(xr + i) - mu) * istd * ScalarIO<T>::load(w + i);
+      if (HAS_BIAS) o += ScalarIO<T>::load(b + i);
+      ScalarIO<T>::store(yr + i, o);
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void layernorm_bwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd, const T* __restrict__ dy,
+                                     T* __restrict__ dx, float* __restrict__ dw,
+                                     float* __restrict__ db, int H, long N) {
+  extern __shared__ float acc[];  // 2*H floats: [dw | db]
+  __shared__ float rbuf[NWAVES];
+  float* dwacc = acc;
+  float* dbacc = acc + H;
+  for (int i = threadIdx.x; i < H; i += BLOCK) {
+    dwacc[i] = 0.f;
+    dbacc[i] = 0.f;
+  }
+  __syncthreads();
+  const int H8 = H & ~7;
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + (size_t)row * H;
+    const T* dyr = dy + (size_t)row * H;
+    T* dxr = dx + (size_t)row * H;
+    const float mu = mean[row];
+    const float istd = invstd[row];
+    float s1 = 0.f, s2 = 0.f;  // sum(dy*w), sum(dy*w*xhat)
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float xv[8], dv[8], wv[8];
+      load8<T>(xr + base, xv);
+      load8<T>(dyr + base, dv);
+      load8<T>(w + base, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float dyw = dv[i] * wv[i];
+        s1 += dyw;
+        s2 += dyw * (xv[i] - mu) * istd;
+      }
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      float dyw = ScalarIO<T>::load(dyr + i) * ScalarIO<T>::load(w + i);
+      s1 += dyw;
+      s2 += dyw * (ScalarIO<T>::load(xr + i) - mu) * istd;
+    }
+    s1 = block_sum<NWAVES>(s1, rbuf) / H;
+    s2 = block_sum<NWAVES>(s2, rbuf) / H;
+    for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
+      float xv[8], dv[8], wv[8], o[8];
+      load8<T>(xr + base, xv);
+      load8<T>(dyr + base, dv);
+      load8<T>(w + base, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float xhat = (xv[i] - mu) * istd;
+        o[i] = istd * (dv[i] * wv[i] - s1 - xhat * s2);
+        dwacc[base + i] += dv[i] * xhat;
+        dbacc[base + i] += dv[i];
+      }
+      store8<T>(dxr + base, o);
+    }
+    for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
+      float dv = ScalarIO<T>::load(dyr + i);
+      float xhat = (ScalarIO<T>::load(xr + i) - mu) * istd;
+      ScalarIO<T>::store(dxr + i, istd * (dv * ScalarIO<T>::load(w + i) - s1 - xhat * s2));
+      dwacc[i] += dv * xhat;
+      dbacc[i] += dv;
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < H; i += BLOCK) {
+    atomicAdd(&dw[i], dwacc[i]);
+    atomicAdd(&db[i], dbacc[i]);
+  }
+}
+
+int pick_grid(long n) { return (int)std::min<long>(n, 2048); }
+
+}  // namespace
+
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.dtype() == x.dtype());
+  const long N = x.size(0);
+  const int H = x.size(1);
+  auto y = at::empty_like(x);
+  auto invr = at::empty({N}, x.options().dtype(at::kFloat));
+  if (N == 0) return {y, invr};
+  auto stream = c10::hip::getCurrentHIPStream();
+  auto wc = w.contiguous();
+  const int grid = pick_grid(N);
+  if (x.dtype() == at::kBFloat16) {
+    rmsnorm_fwd_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(
+        reinterpret_cast<const bf16_t*>(x.data_ptr()), reinterpret_cast<const bf16_t*>(wc.data_ptr()),
+        reinterpret_cast<bf16_t*>(y.data_ptr()), invr.data_ptr<float>(), H, (float)eps, N);
+  } else {
+    rmsnorm_fwd_kernel<float><<<grid, BLOCK, 0, stream>>>(x.data_ptr<float>(), wc.data_ptr<float>(),
+                                                          y.data_ptr<float>(),
+                                                          invr.data_ptr<float>(), H, (float)eps, N);
+  }
+  HIP_CHECK_LAST();
+  return {y, invr};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
+                                    const at::Tensor& invr, const at::Tensor& dy) {
+  const long N = x.size(0);
+  const int H = x.size(1);
+  TORCH_CHECK(H <= 16384, "rmsnorm_bwd: H too large for LDS accumulation");
+  auto dx = at::empty_like(x);
+  auto dwf = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream();
+  auto wc = w.contiguous();
+  const int grid = pick_grid(N);
+  const size_t lds = (size_t)H * sizeof(float);
+  if (N > 0) {
+    if (x.dtype() == at::kBFloat16) {
+      rmsnorm_bwd_kernel<bf16_t><<<grid, BLOCK, lds, stream>>>(
+          reinterpret_cast<const bf16_t*>(x.data_ptr()),
+          reinterpret_cast<const bf16_t*>(wc.data_ptr()), invr.data_ptr<float>(),
+          reinterpret_cast<const bf16_t*>(dy.data_ptr()), reinterpret_cast<bf16_t*>(dx.data_ptr()),
+          dwf.data_ptr<float>(), H, N);
+    } else {
+      rmsnorm_bwd_kernel<float><<<grid, BLOCK, lds, stream>>>(
+          x.data_ptr<float>(), wc.data_ptr<float>(), invr.data_ptr<float>(), dy.data_ptr<float>(),
+          dx.data_ptr<float>(), dwf.data_ptr<float>(), H, N);
+    }
+    HIP_CHECK_LAST();
+  }
+  return {dx, dwf.to(w.dtype())};
+}
+
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& w,
+                                      const c10::optional<at::Tensor>& b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  const long N = x.size(0);
+  const int H = x.size(1);
+  auto y = at::empty_like(x);
+  auto mean = at::empty({N}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty({N}, x.options().dtype(at::kFloat));
+  if (N == 0) return {y, mean, invstd};
+  auto stream = c10::hip::getCurrentHIPStream();
+  auto wc = w.contiguous();
+  const bool has_bias = b.has_value();
+  at::Tensor bc;
+  if (has_bias) bc = b->contiguous();
+  const int grid = pick_grid(N);
+#define LAUNCH_LN_FWD(T, HB, XP, WP, BP, YP)                                                   \
+  layernorm_fwd_kernel<T, HB><<<grid, BLOCK, 0, stream>>>(XP, WP, BP, YP, mean.data_ptr<float>(), \
+                                                          invstd.data_ptr<float>(), H, (float)eps, N)
+  if (x.dtype() == at::kBFloat16) {
+    auto xp = reinterpret_cast<const bf16_t*>(x.data_ptr());
+    auto wp = reinterpret_cast<const bf16_t*>(wc.data_ptr());
+    auto yp = reinterpret_cast<bf16_t*>(y.data_ptr());
+    if (has_bias)
+      LAUNCH_LN_FWD(bf16_t, true, xp, wp, reinterpret_cast<const bf16_t*>(bc.data_ptr()), yp);
+    else
+      LAUNCH_LN_FWD(bf16_t, false, xp, wp, nullptr, yp);
+  } else {
+    auto xp = x.data_ptr<float>();
+    auto wp = wc.data_ptr<float>();
+    auto yp = y.data_ptr<float>();
+    if (has_bias)
+      LAUNCH_LN_FWD(float, true, xp, wp, bc.data_ptr<float>(), yp);
+    else
+      LAUNCH_LN_FWD(float, false, xp, wp, nullptr, yp);
+  }
+#undef LAUNCH_LN_FWD
+  HIP_CHECK_LAST();
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x, const at::Tensor& w,
+                                      const at::Tensor& mean, const at::Tensor& invstd,
+                                      const at::Tensor& dy) {
+  const long N = x.size(0);
+  const int H = x.size(1);
+  TORCH_CHECK(H <= 16384, "layernorm_bwd: H too large for LDS accumulation");
+  auto dx = at::empty_like(x);
+  auto dwf = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto dbf = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream();
+  auto wc = w.contiguous();
+  const int grid = pick_grid(N);
+  const size_t lds = 2 * (size_t)H * sizeof(float);
+  if (N > 0) {
+    if (x.dtype() == at::kBFloat16) {
+      layernorm_bwd_kernel<bf16_t><<<grid, BLOCK, lds, stream>>>(
+          reinterpret_cast<const bf16_t*>(x.data_ptr()),
+          reinterpret_cast<const bf16_t*>(wc.data_ptr()), mean.data_ptr<float>(),
+          invstd.data_ptr<float>(), reinterpret_cast<const bf16_t*>(dy.data_ptr()),
+          reinterpret_cast<bf16_t*>(dx.data_ptr()), dwf.data_ptr<float>(), dbf.data_ptr<float>(),
+          H, N);
+    } else {
+      layernorm_bwd_kernel<float><<<grid, BLOCK, lds, stream>>>(
+          x.data_ptr<float>(), wc.data_ptr<float>(), mean.data_ptr<float>(),
+          invstd.data_ptr<float>(), dy.data_ptr<float>(), dx.data_ptr<float>(),
+          dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
+    }
+    HIP_CHECK_LAST();
+  }
+  return {dx, dwf.to(w.dtype()), dbf.to(w.dtype())};
+}

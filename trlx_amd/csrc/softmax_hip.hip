@@ -1,0 +1,143 @@
+#include "hip/hip_runtime.h"
+// Fused causal-masked softmax for training attention (SURVEY.md K2's
+// softmax/mask half; score and PV GEMMs ride rocBLAS batched MFMA GEMMs).
+//
+// Replaces Apex's scaled_upper_triang_masked_softmax: the causal mask is
+// never materialized — each row's valid prefix length is computed from the
+// row index, the online max/sum runs in fp32 over vectorized bf16 loads, and
+// masked positions are written as exact zeros (so the PV GEMM sees zeros).
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int BLOCK = 256;
+constexpr int NWAVES = BLOCK / WAVE;
+
+template <typename T>
+__global__ void causal_softmax_fwd_kernel(const T* __restrict__ scores, T* __restrict__ probs,
+                                          const int* __restrict__ key_starts, int HTq, int Tq,
+                                          int Tk, int start_pos, long rows) {
+  __shared__ MS wbuf[NWAVES];
+  __shared__ float bshare[2];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int tq = row % Tq;
+    const int kstart = key_starts ? key_starts[row / HTq] : 0;
+    const int valid = min(Tk, start_pos + tq + 1);
+    const T* xr = scores + (size_t)row * Tk;
+    T* pr = probs + (size_t)row * Tk;
+    MS ms{-INFINITY, 0.f};
+    for (int i = kstart + threadIdx.x; i < valid; i += BLOCK) {
+      const float xi = ScalarIO<T>::load(xr + i);
+      if (xi > ms.m) {
+        ms.s = ms.s * expf(ms.m - xi) + 1.f;
+        ms.m = xi;
+      } else {
+        ms.s += expf(xi - ms.m);
+      }
+    }
+    ms = wave_ms(ms);
+    const int wid = threadIdx.x / WAVE;
+    if (threadIdx.x % WAVE == 0) wbuf[wid] = ms;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      MS total = wbuf[0];
+#pragma unroll
+      for (int i = 1; i < NWAVES; ++i) total = ms_combine(total, wbuf[i]);
+      bshare[0] = total.m;
+      bshare[1] = (total.s > 0.f) ? 1.0f / total.s : 0.f;
+    }
+    __syncthreads();
+    const float m = bshare[0];
+    const float rs = bshare[1];
+    for (int i = threadIdx.x; i < Tk; i += BLOCK) {
+      float p = 0.f;
+      if (i >= kstart && i < valid) p = expf(ScalarIO<T>::load(xr + i) - m) * rs;
+      ScalarIO<T>::store(pr + i, p);
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void causal_softmax_bwd_kernel(const T* __restrict__ probs, const T* __restrict__ dprobs,
+                                          T* __restrict__ dscores, int Tk, long rows) {
+  __shared__ float rbuf[NWAVES];
+  __shared__ float bshare;
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* pr = probs + (size_t)row * Tk;
+    const T* dpr = dprobs + (size_t)row * Tk;
+    T* dsr = dscores + (size_t)row * Tk;
+    float dot = 0.f;
+    for (int i = threadIdx.x; i < Tk; i += BLOCK) {
+      dot += ScalarIO<T>::load(pr + i) * ScalarIO<T>::load(dpr + i);
+    }
+    dot = block_sum<NWAVES>(dot, rbuf);
+    if (threadIdx.x == 0) bshare = dot;
+    __syncthreads();
+    const float d = bshare;
+    for (int i = threadIdx.x; i < Tk; i += BLOCK) {
+      const float p = ScalarIO<T>::load(pr + i);
+      ScalarIO<T>::store(dsr + i, p * (ScalarIO<T>::load(dpr + i) - d));
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+at::Tensor causal_softmax_fwd(const at::Tensor& scores, long start_pos,
+                              const c10::optional<at::Tensor>& key_starts) {
+  TORCH_CHECK(scores.is_cuda() && scores.dim() == 4 && scores.is_contiguous());
+  const long B = scores.size(0), H = scores.size(1), Tq = scores.size(2), Tk = scores.size(3);
+  auto probs = at::empty_like(scores);
+  const long rows = B * H * Tq;
+  if (rows == 0) return probs;
+  const int grid = (int)std::min<long>(rows, 8192);
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int* ks = nullptr;
+  at::Tensor ksc;
+  if (key_starts.has_value()) {
+    ksc = key_starts->contiguous();
+    TORCH_CHECK(ksc.numel() == B && ksc.dtype() == at::kInt);
+    ks = ksc.data_ptr<int>();
+  }
+  const int HTq = (int)(H * Tq);
+  if (scores.dtype() == at::kBFloat16) {
+   hipLaunchKernelGGL(( causal_softmax_fwd_kernel<bf16_t>), dim3(grid), dim3(BLOCK), 0, stream, 
+        reinterpret_cast<const bf16_t*>(scores.data_ptr()),
+        reinterpret_cast<bf16_t*>(probs.data_ptr()), ks, HTq, (int)Tq, (int)Tk, (int)start_pos,
+        rows);
+  } else if (scores.dtype() == at::kFloat) {
+   hipLaunchKernelGGL(( causal_softmax_fwd_kernel<float>), dim3(grid), dim3(BLOCK), 0, stream, 
+        scores.data_ptr<float>(), probs.data_ptr<float>(), ks, HTq, (int)Tq, (int)Tk,
+        (int)start_pos, rows);
+  } else {
+    TORCH_CHECK(false, "causal_softmax: unsupported dtype");
+  }
+  HIP_CHECK_LAST();
+  return probs;
+}
+
+at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs) {
+  const long B = probs.size(0), H = probs.size(1), Tq = probs.size(2), Tk = probs.size(3);
+  auto dscores = at::empty_like(probs);
+  const long rows = B * H * Tq;
+  if (rows == 0) return dscores;
+  const int grid = (int)std::min<long>(rows, 8192);
+  auto stream = c10::hip::getCurrentHIPStream();
+  if (probs.dtype() == at::kBFloat16) {
+   hipLaunchKernelGGL(( causal_softmax_bwd_kernel<bf16_t>), dim3(grid), dim3(BLOCK), 0, stream, 
+        reinterpret_cast<const bf16_t*>(probs.data_ptr()),
+        reinterpret_cast<const bf16_t*>(dprobs.data_ptr()),
+        reinterpret_cast<bf16_t*>(dscores.data_ptr()), (int)Tk, rows);
+  } else {
+   hipLaunchKernelGGL(( causal_softmax_bwd_kernel<float>), dim3(grid), dim3(BLOCK), 0, stream, 
+        probs.data_ptr<float>(), dprobs.data_ptr<float>(), dscores.data_ptr<float>(), (int)Tk,
+        rows);
+  }
+  HIP_CHECK_LAST();
+  return dscores;
+}

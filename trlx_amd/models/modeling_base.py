@@ -1,0 +1,144 @@
+"""Base model wrapper.
+
+Parity target: reference trlx/models/modeling_base.py (PreTrainedModelWrapper:
+from_pretrained / from_config / save_pretrained with separately-prefixed head
+weights).  The wrapped model here is the native CausalTransformer; checkpoints
+are HF-format directories (config.json + model.safetensors with HF weight
+names) so vanilla transformers can load the base model, with the wrapper's
+head weights stored alongside under their ``v_head.`` / ``ilql_heads.`` /
+``frozen_head.`` prefixes (reference modeling_ppo.py:354-368 key scheme).
+"""
+
+import json
+import os
+from typing import Any, Dict, List, Optional, Union
+
+import torch
+import torch.nn as nn
+
+from ..utils import logging
+from .nn.config import PRESETS, TransformerConfig, preset
+from .nn.convert import config_to_hf, load_hf_dir, save_hf_dir, state_dict_to_hf
+from .nn.transformer import CausalTransformer
+
+logger = logging.get_logger(__name__)
+
+WRAPPER_HEADS_NAME = "wrapper_heads.pt"
+
+
+class PreTrainedModelWrapper(nn.Module):
+    """A wrapper around the native transformer plus method-specific heads."""
+
+    _auto_model_parent_class = CausalTransformer
+    _supported_modules: List[str] = []  # head module names, e.g. ["v_head"]
+    _supported_args: List[str] = []
+
+    def __init__(self, base_model: CausalTransformer, **kwargs):
+        super().__init__()
+        self.base_model = base_model
+        self.config = base_model.config
+
+    # --- construction -------------------------------------------------------
+
+    @classmethod
+    def _split_kwargs(cls, kwargs: Dict[str, Any]):
+        supported = {}
+        unsupported = {}
+        for k, v in kwargs.items():
+            if k in cls._supported_args:
+                supported[k] = v
+            else:
+                unsupported[k] = v
+        return supported, unsupported
+
+    @classmethod
+    def from_config(cls, config: Union[TransformerConfig, str], **kwargs):
+        """Build with randomly initialized weights."""
+        if isinstance(config, str):
+            config = preset(config)
+        wrapped_kwargs, _ = cls._split_kwargs(kwargs)
+        base = CausalTransformer(config)
+        return cls(base, **wrapped_kwargs)
+
+    @classmethod
+    def from_pretrained(cls, pretrained_model_name_or_path: Union[str, CausalTransformer], *,
+                        revision=None, **kwargs):
+        """Load from a local HF-format directory (or wrap an existing model).
+
+        With no network, a bare model name (e.g. "gpt2") falls back to a
+        random-init preset of the same architecture with a warning.
+        """
+        wrapped_kwargs, _ = cls._split_kwargs(kwargs)
+        heads_sd = {}
+        if isinstance(pretrained_model_name_or_path, CausalTransformer):
+            base = pretrained_model_name_or_path
+        elif isinstance(pretrained_model_name_or_path, PreTrainedModelWrapper):
+            base = pretrained_model_name_or_path.base_model
+        elif os.path.isdir(pretrained_model_name_or_path):
+            cfg, sd = load_hf_dir(pretrained_model_name_or_path)
+            # wrapper head weights (if this directory was saved by a wrapper)
+            heads_path = os.path.join(pretrained_model_name_or_path, WRAPPER_HEADS_NAME)
+            if os.path.exists(heads_path):
+                heads_sd = torch.load(heads_path, map_location="cpu", weights_only=True)
+            base = CausalTransformer(cfg)
+            missing, unexpected = base.load_state_dict(sd, strict=False)
+            missing = [m for m in missing if not m.startswith("rope_")]
+            if missing:
+                logger.warning(f"missing keys loading {pretrained_model_name_or_path}: {missing}")
+        elif pretrained_model_name_or_path in PRESETS:
+            logger.warning(
+                f"'{pretrained_model_name_or_path}' is not a local directory; building a "
+                f"randomly-initialized '{pretrained_model_name_or_path}' preset (no network access)."
+            )
+            base = CausalTransformer(preset(pretrained_model_name_or_path))
+        else:
+            raise OSError(
+                f"'{pretrained_model_name_or_path}' is neither a local HF directory nor a known "
+                f"preset ({sorted(PRESETS)}); there is no network access to fetch it."
+            )
+        model = cls(base, **wrapped_kwargs)
+        if heads_sd:
+            model.post_init(heads_sd)
+        else:
+            model.post_init({})
+        return model
+
+    def post_init(self, state_dict: Dict[str, torch.Tensor]):
+        """Load head weights saved by ``save_pretrained`` (no-op otherwise)."""
+        if not state_dict:
+            return
+        own = {}
+        for k, v in state_dict.items():
+            own[k] = v
+        missing, unexpected = self.load_state_dict(own, strict=False)
+        if unexpected:
+            logger.warning(f"unexpected head keys: {unexpected}")
+
+    # --- persistence --------------------------------------------------------
+
+    def heads_state_dict(self) -> Dict[str, torch.Tensor]:
+        """State dict of everything except the base model (prefixed keys)."""
+        full = self.state_dict()
+        return {k: v for k, v in full.items() if not k.startswith("base_model.")}
+
+    def save_pretrained(self, save_directory: str, **kwargs):
+        """Write an HF directory for the base model + wrapper head weights."""
+        os.makedirs(save_directory, exist_ok=True)
+        base_sd = {k: v.cpu() for k, v in self.base_model.state_dict().items()}
+        save_hf_dir(save_directory, self.config, base_sd)
+        heads = {k: v.cpu() for k, v in self.heads_state_dict().items()}
+        if heads:
+            torch.save(heads, os.path.join(save_directory, WRAPPER_HEADS_NAME))
+
+    # --- misc ---------------------------------------------------------------
+
+    @property
+    def device(self):
+        return next(self.parameters()).device
+
+    @property
+    def dtype(self):
+        return next(self.base_model.parameters()).dtype
+
+    def num_parameters(self) -> int:
+        return sum(p.numel() for p in self.parameters())

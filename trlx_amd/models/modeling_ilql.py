@@ -1,0 +1,249 @@
+"""ILQL method config, loss, heads, and wrapper with shaped generation.
+
+Parity target: reference trlx/models/modeling_ilql.py — topk_mask (29),
+batched_index_select (36), ILQLConfig loss (94-166: twin-Q TD loss vs
+r + gamma*V', expectile-tau V loss, CQL cross-entropy, AWAC weighted CE),
+ILQLHeads with Polyak-synced frozen target Q heads (169-227), and the custom
+token-by-token generate that samples from log pi + beta*(minQ - V) with a
+top-k mask (325-412).
+
+MI355X redesign: generation rides the native KV-cached loop
+(nn/generation.py) with the ILQL shaping as a ``shaping_fn`` hook, so decode
+attention and sampling still hit the fused HIP kernels; heads compute fp32.
+"""
+
+from copy import deepcopy
+from dataclasses import dataclass, field
+from functools import reduce
+from typing import Any, Dict, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..data.ilql_types import ILQLBatch
+from ..data.method_configs import MethodConfig, register_method
+from ..utils.modeling import flatten_dict, get_tensor_stats, make_head
+from .modeling_base import PreTrainedModelWrapper
+from .nn.generation import GenerateConfig, generate
+from .nn.transformer import CausalTransformer
+
+
+def topk_mask(xs: torch.Tensor, k: int) -> torch.Tensor:
+    """Keep the top-k entries, -inf the rest (reference modeling_ilql.py:29)."""
+    if k > xs.shape[-1]:
+        return xs
+    mintop = torch.topk(xs, k)[0][:, -1].unsqueeze(-1)
+    return torch.where(xs < mintop, torch.full_like(xs, float("-inf")), xs)
+
+
+def batched_index_select(x: torch.Tensor, idxs: torch.Tensor, dim: int = 1) -> torch.Tensor:
+    """Gather vectors at ``idxs`` along ``dim`` (reference modeling_ilql.py:36)."""
+    idxs = idxs.unsqueeze(-1).expand(idxs.shape[0], idxs.shape[1], x.shape[-1])
+    return x.gather(dim=dim, index=idxs)
+
+
+@dataclass
+@register_method
+class ILQLConfig(MethodConfig):
+    """ILQL hyperparameters (reference modeling_ilql.py:48-166).
+
+    :param tau: expectile for the V regression (0.5 = MSE, ->1 = max over Q)
+    :param gamma: discount
+    :param cql_scale: conservative-Q (CQL) loss scale
+    :param awac_scale: AWAC weighted-CE loss scale
+    :param alpha: Polyak coefficient for target-Q sync
+    :param beta: advantage weighting in AWAC / generation shaping
+    :param steps_for_target_q_sync: steps between target-Q syncs
+    :param two_qs: twin Q-heads with min-combine
+    """
+
+    name: str = "ILQLConfig"
+    tau: float = 0.7
+    gamma: float = 0.99
+    cql_scale: float = 0.1
+    awac_scale: float = 1.0
+    alpha: float = 0.001
+    beta: float = 0.0
+    steps_for_target_q_sync: int = 5
+    two_qs: bool = True
+    gen_kwargs: Dict[str, Any] = field(default_factory=lambda: dict(max_new_tokens=56, top_k=20, beta=4, temperature=1.0))
+
+    def loss(self, outputs, labels: ILQLBatch):
+        """The ILQL objective: TD-Q + expectile-V + CQL + AWAC
+        (semantics of reference modeling_ilql.py:94-166)."""
+        logits, (qs, target_qs, vs) = outputs
+        terminal_mask = labels.dones[:, :-1]
+        n_nonterminal = max(1, terminal_mask.sum())
+
+        if isinstance(labels, ILQLBatch):
+            actions = labels.input_ids[:, 1:].gather(dim=1, index=labels.actions_ixs).unsqueeze(-1)
+        else:
+            actions = labels.decoder_input_ids[:, 1:].unsqueeze(-1)
+        nactions = actions.shape[1]
+        bsize, _, dsize = logits.shape
+
+        Q = [q.gather(-1, actions).squeeze(-1) for q in qs]
+        targetQs = [q.gather(-1, actions).squeeze(-1).detach() for q in target_qs]
+        targetQ = reduce(torch.minimum, targetQs)
+
+        # len(states) == len(rewards) + 1: V of current states / next states
+        V = vs[:, :-1, 0]
+        Vnext = vs[:, 1:, 0] * labels.dones[:, 1:].to(vs.dtype)
+        Q_target = labels.rewards + self.gamma * Vnext.detach()
+
+        loss_q = sum(((Qi - Q_target) * terminal_mask).pow(2).sum() / n_nonterminal for Qi in Q)
+
+        targetQ = targetQ.detach()
+        expectile_w = torch.where(targetQ >= V, self.tau, 1 - self.tau)
+        loss_v = (expectile_w * (targetQ - V).pow(2) * terminal_mask).sum() / n_nonterminal
+
+        def cql_loss(q):
+            ce = F.cross_entropy(q.reshape(-1, dsize), actions.reshape(-1), reduction="none")
+            return (ce.reshape(bsize, nactions) * terminal_mask).sum() / n_nonterminal
+
+        loss_cql = sum(cql_loss(q) for q in qs)
+
+        action_logits = batched_index_select(logits, labels.actions_ixs, dim=1)
+        cross_entropy = F.cross_entropy(
+            action_logits.reshape(-1, dsize).float(), actions.reshape(-1), reduction="none"
+        ).reshape(bsize, nactions)
+        with torch.no_grad():
+            awac_weight = torch.exp(self.beta * (targetQ - V))
+        loss_awac = torch.sum(cross_entropy * awac_weight * terminal_mask) / n_nonterminal
+
+        loss = loss_q + loss_v + self.cql_scale * loss_cql + self.awac_scale * loss_awac
+
+        stats = dict(
+            losses=dict(
+                loss=loss.item(),
+                loss_q=float(loss_q),
+                loss_v=float(loss_v),
+                loss_cql=float(loss_cql),
+                loss_awac=float(loss_awac),
+            ),
+            values=get_tensor_stats(V, terminal_mask, n_nonterminal),
+            qvalues={str(ix): get_tensor_stats(Q[ix], terminal_mask, n_nonterminal) for ix in range(len(Q))},
+            awac_weight=get_tensor_stats(awac_weight, terminal_mask, n_nonterminal),
+        )
+        return loss, flatten_dict(stats)
+
+
+class ILQLHeads(nn.Module):
+    """V head + 1-2 Q heads + Polyak-synced frozen target Q heads
+    (reference modeling_ilql.py:169-227)."""
+
+    def __init__(self, hidden_size: int, vocab_size: int, two_qs: bool, alpha: float,
+                 dtype: torch.dtype = torch.float32):
+        super().__init__()
+        self.hidden_size = hidden_size
+        self.vocab_size = vocab_size
+        self.two_qs = two_qs
+        self.alpha = alpha
+        self.v_head = make_head(hidden_size, 1, dtype)
+        n_qs = 2 if two_qs else 1
+        self.q_heads = nn.ModuleList(make_head(hidden_size, vocab_size, dtype) for _ in range(n_qs))
+        self.target_q_heads = nn.ModuleList(deepcopy(q) for q in self.q_heads)
+        for t in self.target_q_heads:
+            t.requires_grad_(False)
+
+    def forward(self, hs: torch.Tensor, states_ixs: Optional[torch.Tensor] = None,
+                actions_ixs: Optional[torch.Tensor] = None):
+        hs = hs.float()
+        if states_ixs is not None:
+            states_hs = batched_index_select(hs, states_ixs, 1)
+            actions_hs = batched_index_select(hs, actions_ixs, 1)
+        else:
+            states_hs = actions_hs = hs
+        qs = tuple(q(actions_hs) for q in self.q_heads)
+        target_qs = tuple(q(actions_hs) for q in self.target_q_heads)
+        vs = self.v_head(states_hs)
+        return qs, target_qs, vs
+
+    def sync_target_q_heads(self):
+        """Polyak update: target <- alpha*q + (1-alpha)*target."""
+        with torch.no_grad():
+            for target_q, q in zip(self.target_q_heads, self.q_heads):
+                for tp, p in zip(target_q.parameters(), q.parameters()):
+                    tp.data.copy_(self.alpha * p.data + (1.0 - self.alpha) * tp.data)
+
+
+@dataclass
+class CausalILQLOutput:
+    logits: Optional[torch.Tensor] = None
+    qs: Optional[Tuple[torch.Tensor, ...]] = None
+    target_qs: Optional[Tuple[torch.Tensor, ...]] = None
+    vs: Optional[torch.Tensor] = None
+    last_hidden_state: Optional[torch.Tensor] = None
+
+
+class AutoModelForCausalLMWithILQLHeads(PreTrainedModelWrapper):
+    """Native LM + ILQL heads with shaped generation
+    (reference modeling_ilql.py:262-442)."""
+
+    _supported_modules = ["ilql_heads"]
+    _supported_args = ["two_qs", "alpha", "peft_config"]
+
+    def __init__(self, base_model: CausalTransformer, two_qs: bool = True, alpha: float = 0.99,
+                 peft_config=None):
+        super().__init__(base_model)
+        self.two_qs = two_qs
+        self.alpha = alpha
+        self.peft_config = peft_config
+        self.ilql_heads = ILQLHeads(self.config.hidden_size, self.config.vocab_size, two_qs, alpha)
+
+    def forward(self, input_ids, attention_mask=None, position_ids=None,
+                actions_ixs=None, states_ixs=None, **kwargs):
+        out = self.base_model(input_ids, attention_mask=attention_mask, position_ids=position_ids)
+        qs, target_qs, vs = self.ilql_heads(out.last_hidden_state, states_ixs=states_ixs,
+                                            actions_ixs=actions_ixs)
+        return CausalILQLOutput(out.logits, qs, target_qs, vs, out.last_hidden_state)
+
+    def sync_target_q_heads(self):
+        self.ilql_heads.sync_target_q_heads()
+
+    @torch.no_grad()
+    def generate(
+        self,
+        input_ids,
+        attention_mask=None,
+        beta: float = 1.0,
+        max_new_tokens: int = 32,
+        max_length: int = 1024,
+        temperature: float = 1.0,
+        top_k: int = 20,
+        logit_mask: Optional[torch.Tensor] = None,
+        pad_token_id: Optional[int] = None,
+        eos_token_id: Optional[int] = None,
+        **kwargs,
+    ):
+        """Sample from softmax(topk(log pi + beta*(minQ - V)) / T) per step
+        (reference modeling_ilql.py:325-412), on the KV-cached native loop."""
+        if attention_mask is None and pad_token_id is not None:
+            attention_mask = input_ids.not_equal(pad_token_id).long()
+
+        def shaping_fn(logits, hidden, last_tokens):
+            hs = hidden.unsqueeze(1)
+            qs, target_qs, vs = self.ilql_heads(hs)
+            if self.two_qs:
+                q = torch.minimum(target_qs[0][:, -1, :], target_qs[1][:, -1, :])
+            else:
+                q = target_qs[0][:, -1, :]
+            v = vs[:, -1, :]
+            if logit_mask is not None:
+                mask = logit_mask[last_tokens.to(logit_mask.device)]
+                logits = logits.masked_fill(mask.to(logits.device), float("-inf"))
+            adv = (q - v).to(logits.dtype)
+            pi_beta = F.log_softmax(logits, -1)
+            return pi_beta + beta * adv
+
+        max_new_tokens = min(max_new_tokens, max_length - input_ids.shape[1])
+        gen = GenerateConfig(
+            max_new_tokens=max_new_tokens,
+            do_sample=temperature > 0,
+            temperature=temperature if temperature > 0 else 1.0,
+            top_k=top_k,
+            eos_token_id=eos_token_id,
+            pad_token_id=eos_token_id if eos_token_id is not None else pad_token_id,
+        )
+        return generate(self.base_model, input_ids, attention_mask, gen=gen, shaping_fn=shaping_fn)

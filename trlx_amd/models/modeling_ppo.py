@@ -1,0 +1,304 @@
+"""PPO method config, losses, KL controllers, and value-head/hydra wrappers.
+
+Parity target: reference trlx/models/modeling_ppo.py —
+AdaptiveKLController/FixedKLController (35-67), PPOConfig with GAE (136-173)
+and the clipped PPO loss (175-238), AutoModelForCausalLMWithValueHead
+(266-382), AutoModelForCausalLMWithHydraValueHead + ModelBranch (385-544).
+
+MI355X redesign notes:
+- GAE runs as the wave-parallel affine-scan HIP kernel (ops.gae) instead of a
+  Python reverse loop; whitening uses the fused stats kernels + one RCCL
+  all-reduce when distributed.
+- The hydra frozen branch re-runs only the top unfrozen blocks from the
+  trunk's stashed hidden state: ``forward(..., return_ref_logits=True)`` gives
+  policy logits, values, AND reference logits in ONE trunk pass (the reference
+  needs forward + forward_hydra = two).  The frozen bottom layers are shared
+  storage — 288 GB HBM keeps the whole hydra resident.
+- No per-arch branch classes: the native Block is arch-agnostic.
+"""
+
+import copy
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..data.method_configs import MethodConfig, register_method
+from ..utils.modeling import flatten_dict, get_tensor_stats, make_head, whiten
+from .modeling_base import PreTrainedModelWrapper
+from .nn.generation import generate
+from .nn.transformer import CausalTransformer
+
+
+class AdaptiveKLController:
+    """Adaptive KL controller per Ziegler et al. "Fine-Tuning Language Models
+    from Human Preferences" (reference modeling_ppo.py:35-53)."""
+
+    def __init__(self, init_kl_coef: float, target: float, horizon: int):
+        self.value = init_kl_coef
+        self.target = target
+        self.horizon = horizon
+
+    def update(self, current: float, n_steps: int):
+        proportional_error = max(min(current / self.target - 1, 0.2), -0.2)
+        mult = 1 + proportional_error * n_steps / self.horizon
+        self.value *= mult
+
+
+class FixedKLController:
+    """Fixed KL coefficient (reference modeling_ppo.py:56-67)."""
+
+    def __init__(self, kl_coef: float):
+        self.value = kl_coef
+
+    def update(self, current: float, n_steps: int):
+        pass
+
+
+@dataclass
+@register_method
+class PPOConfig(MethodConfig):
+    """PPO hyperparameters (reference modeling_ppo.py:73-238).
+
+    :param ppo_epochs: optimization epochs per rollout batch
+    :param num_rollouts: experiences to collect per outer epoch
+    :param chunk_size: prompts per generate() call during rollouts
+    :param init_kl_coef: initial KL penalty coefficient
+    :param target: adaptive-KL target (None -> fixed coefficient)
+    :param horizon: adaptive-KL horizon
+    :param gamma, lam: GAE discount / lambda
+    :param cliprange, cliprange_value: PPO ratio / value clipping
+    :param vf_coef: value loss coefficient
+    :param scale_reward: None | "ref" | "running" reward scaling
+    :param cliprange_reward: reward clip bound
+    :param gen_kwargs: generation settings for rollouts/eval
+    :param gen_experience_kwargs: override generation settings for experience
+    """
+
+    name: str = "PPOConfig"
+    ppo_epochs: int = 4
+    num_rollouts: int = 128
+    chunk_size: int = 128
+    init_kl_coef: float = 0.001
+    target: Optional[float] = None
+    horizon: int = 10000
+    gamma: float = 1.0
+    lam: float = 0.95
+    cliprange: float = 0.2
+    cliprange_value: float = 0.2
+    vf_coef: float = 1.0
+    scale_reward: Optional[str] = "ignored"
+    ref_mean: Optional[float] = None
+    ref_std: Optional[float] = None
+    cliprange_reward: float = 10.0
+    num_value_layers_unfrozen: int = 0
+    gen_kwargs: Dict[str, Any] = field(default_factory=lambda: dict(max_new_tokens=40, top_k=0, top_p=1.0, do_sample=True))
+    gen_experience_kwargs: Optional[Dict[str, Any]] = None
+
+    def get_advantages_and_returns(
+        self,
+        values: torch.Tensor,
+        rewards: torch.Tensor,
+        response_length: int,
+        use_whitening: bool = True,
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """GAE over the response tokens (reference modeling_ppo.py:136-173);
+        the reverse scan is the affine-scan HIP kernel on GPU."""
+        return ops.gae_advantages_and_returns(
+            values[:, :response_length].float(),
+            rewards[:, :response_length].float(),
+            self.gamma,
+            self.lam,
+            use_whitening=use_whitening,
+        )
+
+    def loss(
+        self,
+        logprobs: torch.Tensor,
+        values: torch.Tensor,
+        old_logprobs: torch.Tensor,
+        old_values: torch.Tensor,
+        advantages: torch.Tensor,
+        returns: torch.Tensor,
+        mask: torch.Tensor,
+    ):
+        """Clipped surrogate PPO loss (reference modeling_ppo.py:175-238)."""
+        values_clipped = torch.clamp(
+            values, old_values - self.cliprange_value, old_values + self.cliprange_value
+        )
+        n = mask.sum()
+
+        vf_loss1 = (values - returns) ** 2
+        vf_loss2 = (values_clipped - returns) ** 2
+        vf_loss = 0.5 * torch.sum(torch.max(vf_loss1, vf_loss2) * mask) / n
+        vf_clipfrac = torch.sum((vf_loss2 > vf_loss1).float() * mask) / n
+
+        log_ratio = (logprobs - old_logprobs) * mask
+        ratio = torch.exp(log_ratio)
+        # k3 KL estimator
+        with torch.no_grad():
+            approx_kl = torch.mean((ratio - 1) - log_ratio)
+
+        pg_loss1 = -advantages * ratio
+        pg_loss2 = -advantages * torch.clamp(ratio, 1.0 - self.cliprange, 1.0 + self.cliprange)
+        pg_loss = torch.sum(torch.max(pg_loss1, pg_loss2) * mask) / n
+        pg_clipfrac = torch.sum((pg_loss2 > pg_loss1).float() * mask) / n
+
+        loss = pg_loss + self.vf_coef * vf_loss
+
+        stats = dict(
+            losses=dict(
+                total_loss=loss.item(),
+                policy_loss=pg_loss.item(),
+                value_loss=vf_loss.item(),
+            ),
+            values=dict(
+                get_tensor_stats(values, mask, n),
+                values_error=torch.sum(((values - returns) * mask) ** 2) / n,
+                clipfrac=vf_clipfrac,
+            ),
+            old_values=get_tensor_stats(old_values, mask, n),
+            returns=get_tensor_stats(returns, mask, n),
+            policy=dict(approx_kl=approx_kl.item(), clipfrac=pg_clipfrac.item()),
+            ratio=(ratio * mask).sum() / n,
+            padding_percentage=1 - n / mask.numel(),
+        )
+        return loss, flatten_dict(stats)
+
+
+@dataclass
+class CausalLMOutputWithValue:
+    logits: Optional[torch.Tensor] = None
+    values: Optional[torch.Tensor] = None
+    ref_logits: Optional[torch.Tensor] = None
+    last_hidden_state: Optional[torch.Tensor] = None
+
+
+class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
+    """Native LM + scalar value head (reference modeling_ppo.py:266-382)."""
+
+    _supported_modules = ["v_head"]
+    _supported_args = ["peft_config", "num_value_layers_unfrozen"]
+
+    def __init__(self, base_model: CausalTransformer, peft_config=None, num_value_layers_unfrozen: int = 0):
+        super().__init__(base_model)
+        self.peft_config = peft_config
+        self.num_value_layers_unfrozen = num_value_layers_unfrozen
+        # value head computes in fp32 regardless of trunk dtype
+        self.v_head = make_head(self.config.hidden_size, 1, dtype=torch.float32)
+
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: Optional[torch.Tensor] = None,
+        position_ids: Optional[torch.Tensor] = None,
+        return_ref_logits: bool = False,
+        **kwargs,
+    ) -> CausalLMOutputWithValue:
+        out = self.base_model(
+            input_ids, attention_mask=attention_mask, position_ids=position_ids
+        )
+        values = self.v_head(out.last_hidden_state.float()).squeeze(-1)
+        return CausalLMOutputWithValue(
+            logits=out.logits, values=values, last_hidden_state=out.last_hidden_state
+        )
+
+    def generate(self, input_ids, attention_mask=None, **kwargs):
+        return generate(self.base_model, input_ids, attention_mask, **kwargs)
+
+    def generate_eval(self, input_ids, attention_mask=None, **kwargs):
+        return self.generate(input_ids, attention_mask, **kwargs)
+
+
+class FrozenBranch(nn.Module):
+    """Frozen copies of the top ``num_layers_unfrozen`` blocks + final norm +
+    lm_head — the hydra reference head (reference ModelBranch,
+    modeling_ppo.py:502-544), arch-agnostic here."""
+
+    def __init__(self, base_model: CausalTransformer, num_layers_unfrozen: int):
+        super().__init__()
+        self.num_layers_unfrozen = num_layers_unfrozen
+        self.blocks = nn.ModuleList(
+            copy.deepcopy(block) for block in base_model.layers[-num_layers_unfrozen:]
+        )
+        self.final_norm = copy.deepcopy(base_model.final_norm)
+        self.lm_head = copy.deepcopy(base_model.lm_head)
+        for p in self.parameters():
+            p.requires_grad_(False)
+
+    def forward(self, hidden: torch.Tensor, ctx, rope_tables) -> torch.Tensor:
+        with torch.no_grad():
+            h = hidden
+            for block in self.blocks:
+                h = block(h, ctx, rope_tables)
+            return self.lm_head(self.final_norm(h))
+
+
+class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
+    """Value-head LM + frozen reference branch sharing the trunk
+    (reference modeling_ppo.py:385-499)."""
+
+    _supported_modules = ["v_head", "frozen_head"]
+    _supported_args = ["peft_config", "num_layers_unfrozen", "num_value_layers_unfrozen"]
+
+    def __init__(self, base_model: CausalTransformer, peft_config=None,
+                 num_layers_unfrozen: int = -1, num_value_layers_unfrozen: int = 0):
+        super().__init__(base_model, peft_config=peft_config,
+                         num_value_layers_unfrozen=num_value_layers_unfrozen)
+        self.num_layers_unfrozen = num_layers_unfrozen
+        self.frozen_head = None
+        if num_layers_unfrozen > 0 and peft_config is None:
+            self.frozen_head = FrozenBranch(base_model, num_layers_unfrozen)
+
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: Optional[torch.Tensor] = None,
+        position_ids: Optional[torch.Tensor] = None,
+        return_ref_logits: bool = False,
+        **kwargs,
+    ) -> CausalLMOutputWithValue:
+        """One trunk pass; optionally also the reference logits via the frozen
+        branch on the stashed pre-branch hidden state (MI355X fusion of the
+        reference's forward + forward_hydra pair)."""
+        stash = None
+        if return_ref_logits and self.frozen_head is not None:
+            stash = -self.num_layers_unfrozen
+        out = self.base_model(
+            input_ids, attention_mask=attention_mask, position_ids=position_ids,
+            hidden_at_layer=stash,
+        )
+        values = self.v_head(out.last_hidden_state.float()).squeeze(-1)
+        ref_logits = None
+        if return_ref_logits and self.frozen_head is not None:
+            ctx = self.base_model.make_context(input_ids, attention_mask, 0)
+            if position_ids is not None:
+                ctx.position_ids = position_ids.to(torch.int32)
+            ref_logits = self.frozen_head(out.hidden_at_layer, ctx, self.base_model.rope_tables)
+        return CausalLMOutputWithValue(
+            logits=out.logits, values=values, ref_logits=ref_logits,
+            last_hidden_state=out.last_hidden_state,
+        )
+
+    def forward_hydra(
+        self,
+        input_ids: torch.Tensor,
+        attention_mask: Optional[torch.Tensor] = None,
+        position_ids: Optional[torch.Tensor] = None,
+        **kwargs,
+    ) -> CausalLMOutputWithValue:
+        """Reference-branch-only forward (parity with reference
+        modeling_ppo.py:410-453); prefer forward(return_ref_logits=True)."""
+        if self.frozen_head is None:
+            raise RuntimeError("forward_hydra requires num_layers_unfrozen > 0")
+        out = self.base_model(
+            input_ids, attention_mask=attention_mask, position_ids=position_ids,
+            hidden_at_layer=-self.num_layers_unfrozen, return_logits=False,
+        )
+        ctx = self.base_model.make_context(input_ids, attention_mask, 0)
+        if position_ids is not None:
+            ctx.position_ids = position_ids.to(torch.int32)
+        ref_logits = self.frozen_head(out.hidden_at_layer, ctx, self.base_model.rope_tables)
+        return CausalLMOutputWithValue(logits=ref_logits)

@@ -1,0 +1,430 @@
+"""Custom-op dispatch: HIP/CDNA4 kernels on GPU, torch reference on CPU.
+
+Policy (enforced): on a GPU (ROCm) device the hand-written gfx950 kernels are
+the ONLY path — if the compiled extension is missing the op raises instead of
+silently falling back to eager PyTorch.  On CPU the pure-torch reference
+implementations run so the full framework is testable without hardware.
+
+Set ``TRLX_AMD_ALLOW_EAGER=1`` to permit the eager fallback on GPU (debugging
+only; never in benchmarks).
+"""
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference
+
+_EXT = None
+_EXT_TRIED = False
+_EXT_ERR: Optional[str] = None
+
+
+def _load_ext():
+    """Import the in-tree compiled extension (trlx_amd._C)."""
+    global _EXT, _EXT_TRIED, _EXT_ERR
+    if _EXT_TRIED:
+        return _EXT
+    _EXT_TRIED = True
+    try:
+        from trlx_amd import _C  # built by setup.py build_ext --inplace
+
+        _EXT = _C
+    except ImportError as e:  # pragma: no cover - exercised on GPU boxes
+        _EXT = None
+        _EXT_ERR = str(e)
+    return _EXT
+
+
+def extension_available() -> bool:
+    return _load_ext() is not None
+
+
+def _require_ext(op_name: str):
+    ext = _load_ext()
+    if ext is None:
+        if os.environ.get("TRLX_AMD_ALLOW_EAGER") == "1":
+            return None
+        raise RuntimeError(
+            f"trlx_amd op '{op_name}' needs the HIP extension (trlx_amd._C) on GPU "
+            f"but it is not built (import error: {_EXT_ERR}). "
+            f"Run `python setup.py build_ext --inplace` (or __graft_entry__.build()). "
+            f"Set TRLX_AMD_ALLOW_EAGER=1 to debug with the eager fallback."
+        )
+    return ext
+
+
+# --------------------------------------------------------------------------
+# logprobs_of_labels
+# --------------------------------------------------------------------------
+
+
+class _LogprobsOfLabels(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels, ext):
+        logprobs, lse = ext.logprobs_fwd(logits, labels)
+        ctx.save_for_backward(logits, labels, lse)
+        return logprobs
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        logits, labels, lse = ctx.saved_tensors
+        ext = _require_ext("logprobs_of_labels")
+        grad_logits = ext.logprobs_bwd(logits, labels, lse, grad_out.contiguous())
+        return grad_logits, None, None
+
+
+def logprobs_of_labels(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """log softmax(logits)[..., labels] without materializing [., V] logprobs.
+
+    Replaces reference trlx/utils/modeling.py:213-219 — the fused HIP kernel
+    does the row max/logsumexp reduction and the single-label gather in one
+    pass."""
+    if logits.is_cuda:
+        ext = _require_ext("logprobs_of_labels")
+        if ext is not None:
+            flat_logits = logits.reshape(-1, logits.shape[-1]).contiguous()
+            flat_labels = labels.reshape(-1).contiguous()
+            out = _LogprobsOfLabels.apply(flat_logits, flat_labels, ext)
+            return out.view(labels.shape)
+    return reference.logprobs_of_labels(logits, labels)
+
+
+# --------------------------------------------------------------------------
+# fused LM-head + logprob gather (no [B,T,V] logits materialization)
+# --------------------------------------------------------------------------
+
+
+def lm_logprobs(hidden: torch.Tensor, weight: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """log p(labels) from final hidden states + lm_head weight, computed
+    tile-by-tile over the vocab so the [N, V] logits never hit HBM.
+
+    Inference-only (no grad).  hidden [N,H] (bf16), weight [V,H], labels [N].
+    """
+    if hidden.is_cuda:
+        ext = _require_ext("lm_logprobs")
+        if ext is not None and hasattr(ext, "lm_logprobs"):
+            return ext.lm_logprobs(hidden.contiguous(), weight.contiguous(), labels.contiguous())
+    logits = hidden.float() @ weight.float().t()
+    return reference.logprobs_of_labels(logits, labels)
+
+
+# --------------------------------------------------------------------------
+# norms
+# --------------------------------------------------------------------------
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps, ext):
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        out, inv_rms = ext.rmsnorm_fwd(x2d, weight, eps)
+        ctx.save_for_backward(x2d, weight, inv_rms)
+        ctx.x_shape = x.shape
+        return out.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x2d, weight, inv_rms = ctx.saved_tensors
+        ext = _require_ext("rmsnorm")
+        dx, dw = ext.rmsnorm_bwd(x2d, weight, inv_rms, grad_out.reshape(x2d.shape).contiguous())
+        return dx.view(ctx.x_shape), dw, None, None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("rmsnorm")
+        if ext is not None:
+            return _RMSNorm.apply(x, weight, eps, ext)
+    return reference.rmsnorm(x, weight, eps)
+
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps, ext):
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        out, mean, invstd = ext.layernorm_fwd(x2d, weight, bias, eps)
+        ctx.save_for_backward(x2d, weight, mean, invstd)
+        ctx.x_shape = x.shape
+        ctx.has_bias = bias is not None
+        return out.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x2d, weight, mean, invstd = ctx.saved_tensors
+        ext = _require_ext("layernorm")
+        dx, dw, db = ext.layernorm_bwd(x2d, weight, mean, invstd, grad_out.reshape(x2d.shape).contiguous())
+        return dx.view(ctx.x_shape), dw, (db if ctx.has_bias else None), None, None
+
+
+def layernorm(
+    x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor], eps: float = 1e-5
+) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _require_ext("layernorm")
+        if ext is not None:
+            return _LayerNorm.apply(x, weight, bias, eps, ext)
+    return reference.layernorm(x, weight, bias, eps)
+
+
+# --------------------------------------------------------------------------
+# RoPE
+# --------------------------------------------------------------------------
+
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin, positions, interleaved, rot, ext):
+        out = ext.rope_fwd(x.contiguous(), cos, sin, positions, interleaved, False, rot)
+        ctx.save_for_backward(cos, sin, positions)
+        ctx.interleaved = interleaved
+        ctx.rot = rot
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        cos, sin, positions = ctx.saved_tensors
+        ext = _require_ext("rope")
+        dx = ext.rope_fwd(grad_out.contiguous(), cos, sin, positions, ctx.interleaved, True, ctx.rot)
+        return dx, None, None, None, None, None, None
+
+
+def apply_rope(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    positions: Optional[torch.Tensor] = None,
+    interleaved: bool = False,
+    rot: int = 0,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Rotary embedding on q, k [B, H, T, D].  ``cos/sin``: [P, rot/2] tables
+    (host-precomputed — guide Appendix B); ``positions``: [B, T] int32 row
+    indices into the tables (required when prompts are left-padded);
+    ``rot`` <= D rotates only the first ``rot`` dims (GPT-J/NeoX partial
+    rotary), 0 means full D."""
+    D = q.shape[-1]
+    if rot <= 0:
+        rot = D
+    if positions is None:
+        T = q.shape[2]
+        positions = torch.arange(T, device=q.device, dtype=torch.int32).expand(q.shape[0], T).contiguous()
+    if q.is_cuda:
+        ext = _require_ext("rope")
+        if ext is not None:
+            positions = positions.to(torch.int32).contiguous()
+            qo = _Rope.apply(q, cos, sin, positions, interleaved, rot, ext)
+            ko = _Rope.apply(k, cos, sin, positions, interleaved, rot, ext)
+            return qo, ko
+    # CPU reference: gather per-position tables [B, T, rot/2]
+    c = cos[positions.long()]
+    s = sin[positions.long()]
+    return _rope_ref_positional(q, k, c, s, interleaved, rot)
+
+
+def _rope_ref_positional(q, k, c, s, interleaved, rot):
+    """CPU reference with per-(b,t) tables c,s of [B, T, rot/2]."""
+
+    def rotate(x):
+        xf = x.float()
+        xr = xf[..., :rot]
+        if interleaved:
+            x1, x2 = xr[..., 0::2], xr[..., 1::2]
+        else:
+            half = rot // 2
+            x1, x2 = xr[..., :half], xr[..., half:]
+        cc = c.float().unsqueeze(1)  # [B, 1, T, rot/2]
+        ss = s.float().unsqueeze(1)
+        o1 = x1 * cc - x2 * ss
+        o2 = x2 * cc + x1 * ss
+        if interleaved:
+            out = torch.stack((o1, o2), dim=-1).flatten(-2)
+        else:
+            out = torch.cat((o1, o2), dim=-1)
+        if rot < x.shape[-1]:
+            out = torch.cat((out, xf[..., rot:]), dim=-1)
+        return out.to(x.dtype)
+
+    return rotate(q), rotate(k)
+
+
+# --------------------------------------------------------------------------
+# GAE / whitening
+# --------------------------------------------------------------------------
+
+
+def gae_advantages_and_returns(
+    values: torch.Tensor,
+    rewards: torch.Tensor,
+    gamma: float,
+    lam: float,
+    use_whitening: bool = True,
+    distributed: bool = False,
+    group=None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """GAE reverse scan (+ optional distributed whitening) — SURVEY.md K6."""
+    if values.is_cuda:
+        ext = _require_ext("gae")
+        if ext is not None:
+            adv, ret = ext.gae(values.contiguous().float(), rewards.contiguous().float(), gamma, lam)
+            if use_whitening:
+                adv = whiten(adv, distributed=distributed, group=group)
+            return adv.detach(), ret
+    with torch.no_grad():
+        adv, ret = reference.gae_advantages_and_returns(values, rewards, gamma, lam, use_whitening=False)
+        if use_whitening:
+            adv = whiten(adv, distributed=distributed, group=group)
+        return adv.detach(), ret
+
+
+def whiten(xs: torch.Tensor, shift_mean: bool = True, distributed: bool = False, group=None) -> torch.Tensor:
+    """Zero-mean/unit-var normalization, optionally with global (cross-rank)
+    statistics over ``group`` (reference trlx/utils/modeling.py:200-210)."""
+    if xs.is_cuda:
+        ext = _require_ext("whiten")
+        if ext is not None:
+            stats = ext.sum_count(xs.contiguous().float())  # [sum, sumsq, count]
+            if distributed:
+                import torch.distributed as dist
+
+                dist.all_reduce(stats, group=group)
+            mean = stats[0] / stats[2]
+            var = stats[1] / stats[2] - mean * mean
+            return ext.normalize(xs.contiguous().float(), mean, var, shift_mean)
+    return reference.whiten(xs, shift_mean, distributed, group)
+
+
+# --------------------------------------------------------------------------
+# sampling
+# --------------------------------------------------------------------------
+
+
+def sample_token(
+    logits: torch.Tensor,
+    temperature: float = 1.0,
+    top_k: int = 0,
+    top_p: float = 1.0,
+    generator: Optional[torch.Generator] = None,
+    seed: Optional[int] = None,
+    offset: int = 0,
+) -> torch.Tensor:
+    """One fused sampling step [B, V] -> [B].
+
+    On GPU: Gumbel-max categorical sampling in a single pass (no softmax
+    materialization), counter-based RNG keyed by (seed, offset, row) for
+    per-DP-rank reproducibility.  top_k/top_p pre-filter runs tile-wise in
+    the same kernel via threshold select.
+    """
+    if logits.is_cuda:
+        ext = _require_ext("sample_token")
+        if ext is not None:
+            if temperature == 0.0:
+                return torch.argmax(logits, dim=-1)
+            if seed is None:
+                seed = int(torch.randint(0, 2**31 - 1, (1,), generator=generator).item())
+            if top_p is not None and 0.0 < top_p < 1.0:
+                # top-p needs a sorted scan: do the filter with library sort,
+                # then fused gumbel sampling on the filtered logits
+                lg = _top_p_filter(logits.float() / max(temperature, 1e-6), top_p)
+                return ext.gumbel_sample(lg.contiguous(), 1.0, None, seed, offset)
+            if top_k and 0 < top_k < logits.shape[-1]:
+                lg = (logits.float() / max(temperature, 1e-6)).contiguous()
+                thr = torch.topk(lg, top_k, dim=-1).values[:, -1].contiguous()
+                return ext.gumbel_sample(lg, 1.0, thr, seed, offset)
+            return ext.gumbel_sample(logits.float().contiguous(), float(temperature), None, seed, offset)
+    return reference.sample_token(logits, temperature, top_k, top_p, generator)
+
+
+def _top_p_filter(logits: torch.Tensor, top_p: float) -> torch.Tensor:
+    sorted_logits, sorted_idx = torch.sort(logits, descending=True, dim=-1)
+    probs = torch.softmax(sorted_logits, dim=-1)
+    cum = probs.cumsum(dim=-1)
+    remove = cum - probs > top_p
+    sorted_logits = sorted_logits.masked_fill(remove, float("-inf"))
+    return torch.full_like(logits, float("-inf")).scatter(-1, sorted_idx, sorted_logits)
+
+
+# --------------------------------------------------------------------------
+# attention decode (single new token vs KV cache)
+# --------------------------------------------------------------------------
+
+
+def attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+    seq_starts: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Fused decode attention: q [B, Hq, 1, D] vs cache [B, Hkv, S, D]
+    with per-row valid key range [seq_starts[b], seq_lens[b]) -> [B, Hq, 1, D]
+    (``seq_starts`` handles left-padded prompts).
+
+    Memory-bound flash-decode kernel; online softmax, no
+    [B, H, S] score materialization.
+    """
+    if q.is_cuda:
+        ext = _require_ext("attention_decode")
+        if ext is not None:
+            ss = None if seq_starts is None else seq_starts.to(torch.int32).contiguous()
+            return ext.attention_decode(
+                q.contiguous(), k_cache.contiguous(), v_cache.contiguous(),
+                seq_lens.to(torch.int32).contiguous(), scale, ss,
+            )
+    # reference: masked SDPA
+    B, Hq, _, D = q.shape
+    Hkv = k_cache.shape[1]
+    S = k_cache.shape[2]
+    qf = q.float()
+    kf = k_cache.float()
+    vf = v_cache.float()
+    if Hkv != Hq:
+        rep = Hq // Hkv
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [B, Hq, 1, S]
+    pos = torch.arange(S, device=q.device).view(1, 1, 1, S)
+    mask = pos >= seq_lens.view(B, 1, 1, 1)
+    if seq_starts is not None:
+        mask = mask | (pos < seq_starts.view(B, 1, 1, 1))
+    scores = scores.masked_fill(mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    return torch.matmul(probs, vf).to(q.dtype)
+
+
+# --------------------------------------------------------------------------
+# causal softmax (training attention: rocBLAS GEMM + this fused kernel)
+# --------------------------------------------------------------------------
+
+
+class _CausalSoftmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, scores, start_pos, key_starts, ext):
+        probs = ext.causal_softmax_fwd(scores.contiguous(), start_pos, key_starts)
+        ctx.save_for_backward(probs)
+        return probs
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (probs,) = ctx.saved_tensors
+        ext = _require_ext("causal_softmax")
+        return ext.causal_softmax_bwd(probs, grad_out.contiguous()), None, None, None
+
+
+def causal_softmax(
+    scores: torch.Tensor, start_pos: int = 0, key_starts: Optional[torch.Tensor] = None
+) -> torch.Tensor:
+    """Causal-masked softmax of attention scores [B, H, Tq, Tk] (fused mask +
+    online softmax; SURVEY.md K2's softmax half — GEMMs ride rocBLAS).
+
+    ``key_starts`` [B] marks the first valid key per batch row (left-padding).
+    """
+    if scores.is_cuda:
+        ext = _require_ext("causal_softmax")
+        if ext is not None:
+            if key_starts is not None:
+                key_starts = key_starts.to(torch.int32).contiguous()
+            return _CausalSoftmax.apply(scores, start_pos, key_starts, ext)
+    return reference.causal_softmax(scores, start_pos, key_starts)

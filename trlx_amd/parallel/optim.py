@@ -1,0 +1,168 @@
+"""Arena-based fused AdamW.
+
+Every trainable parameter is repointed into one contiguous arena per dtype:
+- param arena (model dtype, e.g. bf16) — ``p.data`` becomes a view;
+- fp32 master arena (mixed-precision master weights);
+- fp32 m/v moment arenas;
+- grad arena (model dtype) — ``p.grad`` is pre-pinned as a view so autograd
+  accumulates straight into it (gradient_accumulation_fusion, K10).
+
+The whole optimizer step is then ONE HIP kernel launch per arena (vs. one per
+tensor), the DP gradient averaging (1/N) is fused into the step, and the
+bucketed gradient all-reduce (ddp.py) operates on contiguous arena slices.
+"""
+
+import math
+from typing import Dict, Iterable, List, Optional, Tuple
+
+import torch
+
+from .. import ops
+from ..ops import reference
+
+
+class FusedAdamW(torch.optim.Optimizer):
+    """AdamW over flat parameter arenas with fp32 masters.
+
+    On GPU the step is the trlx_amd._C.fused_adamw kernel; on CPU a flat
+    torch implementation with identical math (numerics tests compare them).
+    """
+
+    def __init__(self, params: Iterable[torch.nn.Parameter], lr: float = 1e-3,
+                 betas: Tuple[float, float] = (0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.0, grad_scale: float = 1.0):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self.grad_scale = grad_scale
+        self._step = 0
+        self._arenas: List[Dict] = []
+        self._build_arenas()
+
+    def _build_arenas(self):
+        """Repoint each param group's tensors into contiguous arenas."""
+        for group in self.param_groups:
+            params = [p for p in group["params"] if p.requires_grad]
+            if not params:
+                self._arenas.append(None)
+                continue
+            device = params[0].device
+            dtype = params[0].dtype
+            assert all(p.device == device and p.dtype == dtype for p in params), \
+                "FusedAdamW: one device/dtype per param group"
+            total = sum(p.numel() for p in params)
+            flat_p = torch.empty(total, device=device, dtype=dtype)
+            flat_g = torch.zeros(total, device=device, dtype=dtype)
+            offset = 0
+            offsets = []
+            for p in params:
+                n = p.numel()
+                flat_p[offset : offset + n].copy_(p.data.reshape(-1))
+                p.data = flat_p[offset : offset + n].view(p.shape)
+                p.grad = flat_g[offset : offset + n].view(p.shape)
+                offsets.append((offset, n))
+                offset += n
+            master = flat_p if dtype == torch.float32 else flat_p.float()
+            m = torch.zeros(total, device=device, dtype=torch.float32)
+            v = torch.zeros(total, device=device, dtype=torch.float32)
+            self._arenas.append(dict(
+                params=params, flat_p=flat_p, flat_g=flat_g, master=master, m=m, v=v,
+                offsets=offsets,
+            ))
+
+    # --- optimizer protocol -------------------------------------------------
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        self._step += 1
+        for group, arena in zip(self.param_groups, self._arenas):
+            if arena is None:
+                continue
+            lr = group["lr"]
+            beta1, beta2 = group["betas"]
+            eps = group["eps"]
+            wd = group["weight_decay"]
+            if arena["flat_p"].is_cuda:
+                ext = ops._require_ext("fused_adamw")
+                ext.fused_adamw(arena["flat_p"], arena["master"], arena["flat_g"], arena["m"],
+                                arena["v"], self._step, lr, beta1, beta2, eps, wd, self.grad_scale)
+            else:
+                self._cpu_step(arena, lr, beta1, beta2, eps, wd)
+        return loss
+
+    def _cpu_step(self, arena, lr, beta1, beta2, eps, wd):
+        g = arena["flat_g"].float() * self.grad_scale
+        m, v, master = arena["m"], arena["v"], arena["master"]
+        m.mul_(beta1).add_(g, alpha=1 - beta1)
+        v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        bc1 = 1 - beta1**self._step
+        bc2 = 1 - beta2**self._step
+        master.mul_(1 - lr * wd)
+        denom = (v / bc2).sqrt().add_(eps)
+        master.addcdiv_(m / bc1, denom, value=-lr)
+        if arena["flat_p"].dtype != torch.float32:
+            arena["flat_p"].copy_(master.to(arena["flat_p"].dtype))
+
+    def zero_grad(self, set_to_none: bool = False):
+        # grads are pre-pinned arena views: zero in place, never detach
+        for arena in self._arenas:
+            if arena is not None:
+                arena["flat_g"].zero_()
+
+    # --- persistence ----------------------------------------------------------
+
+    def state_dict(self):
+        return {
+            "step": self._step,
+            "grad_scale": self.grad_scale,
+            "param_groups": [
+                {k: v for k, v in g.items() if k != "params"} for g in self.param_groups
+            ],
+            "arenas": [
+                None if a is None else {"master": a["master"], "m": a["m"], "v": a["v"]}
+                for a in self._arenas
+            ],
+        }
+
+    def load_state_dict(self, state):
+        self._step = state["step"]
+        self.grad_scale = state.get("grad_scale", self.grad_scale)
+        for g, sg in zip(self.param_groups, state["param_groups"]):
+            g.update(sg)
+        for arena, sa in zip(self._arenas, state["arenas"]):
+            if arena is None or sa is None:
+                continue
+            arena["master"].copy_(sa["master"].to(arena["master"].device))
+            arena["m"].copy_(sa["m"].to(arena["m"].device))
+            arena["v"].copy_(sa["v"].to(arena["v"].device))
+            if arena["flat_p"].dtype != torch.float32:
+                arena["flat_p"].copy_(arena["master"].to(arena["flat_p"].dtype))
+
+    # --- arena access for the gradient reducer --------------------------------
+
+    def grad_arenas(self) -> List[Tuple[torch.Tensor, List[Tuple[torch.nn.Parameter, int, int]]]]:
+        """[(flat_grad, [(param, offset, numel), ...]), ...] for bucketing."""
+        out = []
+        for arena in self._arenas:
+            if arena is None:
+                continue
+            entries = [(p, off, n) for p, (off, n) in zip(arena["params"], arena["offsets"])]
+            out.append((arena["flat_g"], entries))
+        return out
+
+
+def build_optimizer(model: torch.nn.Module, name: str, kwargs: dict, world: int = 1):
+    """Construct the optimizer named in the config (reference
+    utils/__init__.py get_optimizer_class registry)."""
+    params = [p for p in model.parameters() if p.requires_grad]
+    kwargs = dict(kwargs)
+    if name in ("fused_adamw", "adamw"):
+        return FusedAdamW(params, grad_scale=1.0 / world, **kwargs)
+    if name == "adam":
+        return torch.optim.Adam(params, **kwargs)
+    if name == "sgd":
+        return torch.optim.SGD(params, **kwargs)
+    raise ValueError(f"Unknown optimizer: {name}")

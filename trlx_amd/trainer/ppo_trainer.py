@@ -1,0 +1,357 @@
+"""PPO trainer.
+
+Parity target: reference trlx/trainer/accelerate_ppo_trainer.py
+(AcceleratePPOTrainer): rollout store + KL controller + RunningMoments ctor
+(42-106), loss with the left-pad/right-pad index bookkeeping (127-204),
+rollout collection make_experience (251-524) with the §2.3 collective
+protocol (pad+gather samples, rank-0 reward, scatter scores, AVG-all-reduce
+of the KL), and the post-epoch store refresh.
+
+MI355X notes: the policy/value/reference forward during experience runs as a
+single trunk pass (hydra ``return_ref_logits=True``); logprob gathers are the
+fused HIP kernel; generation rides the KV-cache decode kernel.
+"""
+
+import json
+import os
+import uuid
+from time import time
+from typing import Any, Dict, List, Tuple
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+from torch.nn.utils.rnn import pad_sequence
+
+from ..data.configs import TRLConfig
+from ..data.ppo_types import PPORLBatch, PPORLElement
+from ..models.modeling_ppo import (
+    AdaptiveKLController,
+    AutoModelForCausalLMWithHydraValueHead,
+    AutoModelForCausalLMWithValueHead,
+    FixedKLController,
+)
+from ..parallel import comm
+from ..pipeline.offline_pipeline import PromptPipeline
+from ..pipeline.ppo_pipeline import PPORolloutStorage
+from ..trainer import register_trainer
+from ..utils import Clock, infinite_dataloader
+from ..utils import logging
+from ..utils.modeling import RunningMoments, gather_dict, logprobs_of_labels
+from .base_trainer import NativeRLTrainer
+
+logger = logging.get_logger(__name__)
+
+
+@register_trainer
+class PPOTrainer(NativeRLTrainer):
+    """PPO RL trainer (also registered under the reference's name)."""
+
+    def __init__(self, config: TRLConfig, **kwargs):
+        super().__init__(config, **kwargs)
+
+        self.log_rollouts = config.train.rollout_logging_dir is not None
+        if self.log_rollouts:
+            self.setup_rollout_logging(config)
+
+        self.store = PPORolloutStorage(self.tokenizer.pad_token_id, self.tokenizer.padding_side)
+        self.store.clear_history()
+
+        # separate full reference model only without a hydra branch
+        if hasattr(self.model, "frozen_head") and self.model.frozen_head is not None:
+            self.ref_model = None
+        else:
+            self.ref_model = self.get_arch(self.config)
+            self.ref_model.base_model.load_state_dict(self.model.base_model.state_dict())
+            self.ref_model = self.ref_model.to(self.device)
+            if self.dtype != torch.float32:
+                self.ref_model.base_model.to(self.dtype)
+            self.ref_model.eval()
+
+        if config.method.target is not None:
+            self.kl_ctl = AdaptiveKLController(config.method.init_kl_coef, config.method.target,
+                                               config.method.horizon)
+        else:
+            self.kl_ctl = FixedKLController(config.method.init_kl_coef)
+
+        self.running_moments = RunningMoments()
+        self.ref_mean = config.method.ref_mean
+        self.ref_std = config.method.ref_std
+        self.mean_kl = 0.0
+
+    def get_arch(self, config: TRLConfig):
+        if config.model.model_arch_type == "seq2seq":
+            raise NotImplementedError("seq2seq PPO lands with the T5 family")
+        model_cls = AutoModelForCausalLMWithHydraValueHead
+        kwargs = dict(
+            num_layers_unfrozen=config.model.num_layers_unfrozen,
+            peft_config=config.model.peft_config,
+        )
+        if config.model.num_layers_unfrozen == -1:
+            model_cls = AutoModelForCausalLMWithValueHead
+            kwargs.pop("num_layers_unfrozen")
+        path = config.model.model_path
+        if isinstance(path, str) and not os.path.isdir(path) and hasattr(config.model, "model_extra_configs") and config.model.model_extra_configs.get("config"):
+            from ..models.nn.config import TransformerConfig
+
+            return model_cls.from_config(
+                TransformerConfig.from_dict(config.model.model_extra_configs["config"]), **kwargs
+            )
+        return model_cls.from_pretrained(path, **kwargs)
+
+    # --- loss ------------------------------------------------------------------
+
+    def loss(self, batch: PPORLBatch) -> Tuple[torch.Tensor, Dict[str, Any]]:
+        """PPO loss on stored rollouts (reference accelerate_ppo_trainer.py:127-204
+        — the start/end index semantics are preserved exactly)."""
+        query_tensors = batch.query_tensors.to(self.device)
+        response_tensors = batch.response_tensors.to(self.device)
+        old_logprobs = batch.logprobs.to(self.device)
+        old_values = batch.values.to(self.device)
+        old_rewards = batch.rewards.to(self.device)
+        response_length = old_rewards.shape[1]
+
+        advantages, returns = self.config.method.get_advantages_and_returns(
+            old_values, old_rewards, response_length
+        )
+
+        tokens = torch.cat((query_tensors, response_tensors), dim=1)
+        attention_mask = tokens.not_equal(self.tokenizer.pad_token_id).long().to(tokens.device)
+        outputs = self.model(tokens, attention_mask)
+        logits = outputs.logits
+        values_pred = outputs.values[:, :-1]
+        logprobs = logprobs_of_labels(logits[:, :-1, :], tokens[:, 1:])
+
+        start = query_tensors.shape[1] - 1
+        end = start + response_length
+        logprobs, values_pred, mask = (
+            logprobs[:, start:end],
+            values_pred[:, start:end],
+            attention_mask[:, start + 1 : end + 1],
+        )
+
+        loss, stats = self.config.method.loss(
+            logprobs=logprobs,
+            values=values_pred,
+            old_logprobs=old_logprobs,
+            old_values=old_values,
+            advantages=advantages,
+            returns=returns,
+            mask=mask,
+        )
+        return loss, stats
+
+    # --- plumbing ----------------------------------------------------------------
+
+    def setup_rollout_logging(self, config):
+        exists = os.path.exists(config.train.rollout_logging_dir)
+        isdir = os.path.isdir(config.train.rollout_logging_dir)
+        assert exists and isdir
+        self.run_id = f"run-{uuid.uuid4()}"
+        self.rollout_logging_dir = os.path.join(config.train.rollout_logging_dir, self.run_id)
+        os.mkdir(self.rollout_logging_dir)
+        with open(os.path.join(self.rollout_logging_dir, "config.json"), "w") as f:
+            f.write(json.dumps(config.to_dict(), indent=2))
+
+    def post_epoch_callback(self):
+        """Clear the store and collect fresh rollouts — the RL outer loop."""
+        if self.log_rollouts:
+            self.store.export_history(location=self.rollout_logging_dir)
+        self.store.clear_history()
+        self.make_experience(self.config.method.num_rollouts, self.iter_count)
+
+    def post_backward_callback(self):
+        self.kl_ctl.update(self.mean_kl, n_steps=self.config.train.batch_size)
+
+    def create_train_dataloader(self):
+        return self.store.create_loader(self.config.train.batch_size, shuffle=True)
+
+    def prepare_learning(self):
+        self.eval_dataloader = self.eval_pipeline.create_loader(self.config.method.chunk_size)
+        self.make_experience(self.config.method.num_rollouts)
+        self.train_dataloader = self.create_train_dataloader()
+        self.n_inner_epochs = self.config.method.ppo_epochs
+        self.total_steps = self.config.train.epochs * self.n_inner_epochs * len(self.train_dataloader)
+        self.total_steps = min(self.total_steps, self.config.train.total_steps)
+
+    def add_prompt_pipeline(self, pipeline: PromptPipeline):
+        prompt_dataloader = pipeline.create_loader(self.config.method.chunk_size, shuffle=True)
+        self.prompt_iterator = infinite_dataloader(prompt_dataloader)
+
+    # --- experience generation ------------------------------------------------------
+
+    def make_experience(self, num_rollouts: int = 1024, iter_count: int = 0):  # noqa: C901
+        """Collect rollouts: generate, score on rank 0, scatter scores, compute
+        per-token KL-penalty rewards, push PPORLElements
+        (reference accelerate_ppo_trainer.py:251-524)."""
+        logger.info("Collecting rollouts")
+        clock = Clock()
+        ppo_rl_elements: List[PPORLElement] = []
+        accumulated_stats: List[Dict] = []
+        device = self.device
+        tbar = logging.tqdm(total=num_rollouts, disable=not comm.is_main_process(),
+                            desc=f"[rollout 0 / {num_rollouts}]")
+
+        while len(ppo_rl_elements) < num_rollouts:
+            stats = {}
+            batch = next(self.prompt_iterator)
+
+            rollout_generate_time = time()
+            samples = self.generate(batch["input_ids"], batch["attention_mask"])
+            stats["time/rollout_generate"] = time() - rollout_generate_time
+
+            prompt_tensors = batch["input_ids"].to(device)
+            prompt_sizes = torch.full((len(prompt_tensors),), prompt_tensors.shape[1],
+                                      device=device, dtype=torch.long)
+
+            padded_samples = comm.pad_across_processes(samples, 1, self.tokenizer.eos_token_id)
+            padded_prompts = comm.pad_across_processes(prompt_tensors, 1, self.tokenizer.eos_token_id)
+            gathered_samples = comm.gather(padded_samples)
+            gathered_prompts = comm.gather(padded_prompts)
+            gathered_prompt_sizes = comm.gather(prompt_sizes)
+            metadata = gather_dict({k: v for k, v in batch.items()
+                                    if k not in ("input_ids", "attention_mask")})
+
+            if comm.is_main_process():
+                all_str_samples, all_str_prompts, all_str_outputs = self.decode(
+                    gathered_prompts, gathered_samples, gathered_prompt_sizes, append_eos_token=True
+                )
+                rollout_score_time = time()
+                all_scores = self.reward_fn(
+                    samples=all_str_samples, prompts=all_str_prompts, outputs=all_str_outputs,
+                    tokenizer=self.tokenizer, **metadata,
+                )
+                all_scores = [torch.tensor(score, dtype=torch.float, device=device).view(-1)
+                              for score in all_scores]
+                all_scores = pad_sequence(all_scores, batch_first=True, padding_value=-np.inf)
+                max_len = torch.tensor(all_scores.shape[1], dtype=torch.long, device=device)
+                stats["time/rollout_score"] = time() - rollout_score_time
+                all_scores = list(all_scores.reshape(self.world_size, -1, max_len).unbind())
+            else:
+                all_scores = None
+                max_len = torch.tensor(0, dtype=torch.long, device=device)
+
+            if torch.distributed.is_initialized():
+                torch.distributed.broadcast(max_len, 0)
+                scores = torch.empty((len(samples), max_len), device=device)
+                torch.distributed.scatter(scores, all_scores)
+            else:
+                scores = all_scores[0].clone().detach()
+            scores_mask = scores != -np.inf
+
+            str_samples, str_prompts, str_outputs = self.decode(prompt_tensors, samples,
+                                                                append_eos_token=True)
+
+            # re-tokenize outputs (stop sequences may have trimmed them)
+            outputs = self.tokenizer(str_outputs).input_ids
+            outputs = list(map(torch.LongTensor, outputs))
+            maxsize = max(max(map(len, outputs)), 1)
+            outputs = [
+                F.pad(output, (0, maxsize - len(output)), value=self.tokenizer.pad_token_id)
+                for output in outputs
+            ]
+            sample_outputs = torch.vstack(outputs).to(device)
+
+            if self.config.method.cliprange_reward:
+                scores = torch.clip(scores, -self.config.method.cliprange_reward,
+                                    self.config.method.cliprange_reward)
+
+            if self.ref_mean is None:
+                self.ref_mean = (scores * scores_mask).sum(dim=1).mean()
+                self.ref_std = (scores * scores_mask).sum(dim=1).std()
+            all_scores_mean, all_scores_std = self.running_moments.update(
+                torch.sum(scores * scores_mask, dim=1))
+            stats["rollout_scores/mean"] = float(all_scores_mean)
+            stats["rollout_scores/std"] = float(all_scores_std)
+            stats["rollout_scores/running_mean"] = float(self.running_moments.mean)
+            stats["rollout_scores/running_std"] = float(self.running_moments.std)
+
+            if self.config.method.scale_reward == "running":
+                scores /= self.running_moments.std
+            elif self.config.method.scale_reward == "ref":
+                scores /= self.ref_std
+
+            # one trunk pass for policy logits + values (+ ref logits via hydra)
+            all_tokens = torch.cat((prompt_tensors, sample_outputs), dim=1)
+            attention_mask = all_tokens.not_equal(self.tokenizer.pad_token_id).long().to(device)
+            with torch.no_grad():
+                outputs = self.model(all_tokens, attention_mask=attention_mask,
+                                     return_ref_logits=True)
+                logits, values = outputs.logits, outputs.values
+                if outputs.ref_logits is not None:
+                    ref_logits = outputs.ref_logits
+                elif self.ref_model is not None:
+                    ref_logits = self.ref_model(all_tokens, attention_mask=attention_mask).logits
+                else:
+                    # num_layers_unfrozen == -1 and no separate ref: KL vs itself
+                    ref_logits = logits
+
+                logprobs = logprobs_of_labels(logits[:, :-1, :], all_tokens[:, 1:])
+                ref_logprobs = logprobs_of_labels(ref_logits[:, :-1, :], all_tokens[:, 1:])
+
+            n_samples = samples.shape[0]
+            start = prompt_tensors.shape[1] - 1
+
+            log_ratio = (logprobs - ref_logprobs) * attention_mask[:, :-1]
+            kl = log_ratio.exp() - 1 - log_ratio
+            mean_kl_per_token = kl.mean()
+            mean_kl = kl.sum(1).mean()
+
+            logprobs = logprobs.cpu()
+            prompt_tensors = prompt_tensors.cpu()
+            sample_outputs = sample_outputs.cpu()
+            values = values.cpu()[:, :-1]
+
+            ends = start + attention_mask[:, start:].sum(1).cpu() + 1
+            all_values = [values[ix, start : ends[ix]] for ix in range(n_samples)]
+            all_logprobs = [logprobs[ix, start : ends[ix]] for ix in range(n_samples)]
+
+            kl_penalty = self.kl_ctl.value * -log_ratio.cpu()
+            kl_penalty = [xs[start : ends[ix]] for ix, xs in enumerate(kl_penalty)]
+
+            rollout_count = 0
+            for sample_idx in range(n_samples):
+                rewards = kl_penalty[sample_idx].clone()
+                if scores.shape[1] == 1:
+                    # terminal reward at the last (eos) token
+                    rewards[-1] += scores[sample_idx][0].cpu()
+                else:
+                    # dense per-token rewards
+                    score = scores[sample_idx]
+                    score_right_padding = torch.sum(scores_mask[sample_idx])
+                    score = score[:score_right_padding].cpu()
+                    p_score = torch.zeros_like(rewards)
+                    p_score[: score.shape[0]] += score
+                    rewards += p_score
+
+                ppo_rl_elements.append(PPORLElement(
+                    query_tensor=prompt_tensors[sample_idx],
+                    response_tensor=sample_outputs[sample_idx],
+                    logprobs=all_logprobs[sample_idx],
+                    values=all_values[sample_idx],
+                    rewards=rewards,
+                ))
+                rollout_count += 1
+
+            if torch.distributed.is_initialized():
+                comm.all_reduce_mean(mean_kl)
+
+            stats["time/rollout_time"] = clock.tick()
+            stats["policy/sqrt_kl"] = torch.sqrt(torch.clamp(mean_kl, min=0)).item()
+            stats["policy/kl_per_token"] = torch.sqrt(torch.clamp(mean_kl_per_token, min=0)).item()
+            accumulated_stats.append(stats)
+            tbar.set_description(f"[rollout {len(ppo_rl_elements)} / {num_rollouts}]")
+            tbar.update(min(rollout_count, num_rollouts))
+        tbar.close()
+
+        stats = {k: sum(xs.get(k, 0.0) for xs in accumulated_stats) / len(accumulated_stats)
+                 for k in accumulated_stats[-1]}
+        stats["kl_ctl_value"] = self.kl_ctl.value
+        self.mean_kl = stats["policy/sqrt_kl"] ** 2
+        self.tracker.log(stats, step=iter_count)
+
+        self.push_to_store(ppo_rl_elements)
+
+    def save_pretrained(self, directory=None, **kwargs):
+        """PPO export saves the base model only (reference ppo_trainer:526-553)."""
+        super().save_pretrained(directory, **kwargs)

@@ -1,0 +1,187 @@
+"""Tokenizer loading with a fully-offline fallback.
+
+Real checkpoints come with tokenizer files (loaded via transformers from a
+local directory).  Without network access and without local files, a
+byte-level tokenizer stands in — deterministic, reversible, and requiring no
+vocab download — which is what the synthetic benchmarks/tests use.
+"""
+
+from typing import List, Optional, Union
+
+import torch
+
+
+class ByteTokenizer:
+    """Minimal HF-compatible byte-level tokenizer (offline stand-in).
+
+    Vocabulary: 256 byte values + BOS/EOS/PAD specials.  Implements the
+    surface the trainers use: __call__, decode, batch_decode, pad/eos/bos
+    attrs, padding_side/truncation_side.
+    """
+
+    def __init__(self, padding_side: str = "left", truncation_side: str = "right",
+                 vocab_size: int = 259):
+        self.byte_offset = 3
+        self.bos_token_id = 0
+        self.eos_token_id = 1
+        self.pad_token_id = 2
+        self.bos_token = "<|bos|>"
+        self.eos_token = "<|eos|>"
+        self.pad_token = "<|pad|>"
+        self.sep_token = ""
+        self.padding_side = padding_side
+        self.truncation_side = truncation_side
+        self.vocab_size = max(vocab_size, 256 + self.byte_offset)
+        self.name_or_path = "byte-tokenizer"
+
+    def __len__(self):
+        return self.vocab_size
+
+    # --- encode ------------------------------------------------------------
+
+    def _encode_one(self, text: str, max_length: Optional[int] = None, truncation: bool = False,
+                    add_special_tokens: bool = False) -> List[int]:
+        specials = {self.bos_token: self.bos_token_id, self.eos_token: self.eos_token_id,
+                    self.pad_token: self.pad_token_id}
+        ids: List[int] = []
+        rest = text
+        while rest:
+            matched = False
+            for tok, tid in specials.items():
+                if rest.startswith(tok):
+                    ids.append(tid)
+                    rest = rest[len(tok):]
+                    matched = True
+                    break
+            if not matched:
+                b = rest[0].encode("utf-8", errors="replace")
+                ids.extend(x + self.byte_offset for x in b)
+                rest = rest[1:]
+        if truncation and max_length is not None and len(ids) > max_length:
+            if self.truncation_side == "left":
+                ids = ids[-max_length:]
+            else:
+                ids = ids[:max_length]
+        return ids
+
+    class _Encoding(dict):
+        @property
+        def input_ids(self):
+            return self["input_ids"]
+
+        @property
+        def attention_mask(self):
+            return self["attention_mask"]
+
+    def __call__(self, text: Union[str, List[str]], truncation: bool = False, padding=False,
+                 max_length: Optional[int] = None, add_special_tokens: bool = False,
+                 return_tensors: Optional[str] = None):
+        single = isinstance(text, str)
+        texts = [text] if single else list(text)
+        all_ids = [self._encode_one(t, max_length, truncation, add_special_tokens) for t in texts]
+        masks = [[1] * len(i) for i in all_ids]
+        if padding:
+            width = max(len(i) for i in all_ids)
+            for i in range(len(all_ids)):
+                n = width - len(all_ids[i])
+                if self.padding_side == "left":
+                    all_ids[i] = [self.pad_token_id] * n + all_ids[i]
+                    masks[i] = [0] * n + masks[i]
+                else:
+                    all_ids[i] = all_ids[i] + [self.pad_token_id] * n
+                    masks[i] = masks[i] + [0] * n
+        if single:
+            out = self._Encoding(input_ids=all_ids[0], attention_mask=masks[0])
+        else:
+            out = self._Encoding(input_ids=all_ids, attention_mask=masks)
+        if return_tensors == "pt":
+            out = self._Encoding(
+                input_ids=torch.tensor(out["input_ids"], dtype=torch.long),
+                attention_mask=torch.tensor(out["attention_mask"], dtype=torch.long),
+            )
+        return out
+
+    def encode(self, text: str, **kwargs) -> List[int]:
+        return self._encode_one(text, kwargs.get("max_length"), kwargs.get("truncation", False))
+
+    # --- decode ------------------------------------------------------------
+
+    def decode(self, ids, skip_special_tokens: bool = True) -> str:
+        if isinstance(ids, torch.Tensor):
+            ids = ids.tolist()
+        if isinstance(ids, int):
+            ids = [ids]
+        out_bytes = bytearray()
+        pieces: List[str] = []
+        for tid in ids:
+            tid = int(tid)
+            if tid < self.byte_offset:
+                if out_bytes:
+                    pieces.append(out_bytes.decode("utf-8", errors="replace"))
+                    out_bytes = bytearray()
+                if not skip_special_tokens:
+                    pieces.append([self.bos_token, self.eos_token, self.pad_token][tid])
+            elif tid < 256 + self.byte_offset:
+                out_bytes.append(tid - self.byte_offset)
+            # ids beyond byte range (random-init sampling) are dropped
+        if out_bytes:
+            pieces.append(out_bytes.decode("utf-8", errors="replace"))
+        return "".join(pieces)
+
+    def batch_decode(self, batch, skip_special_tokens: bool = True) -> List[str]:
+        return [self.decode(x, skip_special_tokens) for x in batch]
+
+    def pad(self, encoded_inputs, return_tensors: str = "pt", **kwargs):
+        ids = [e["input_ids"] for e in encoded_inputs]
+        ids = [i.tolist() if isinstance(i, torch.Tensor) else list(i) for i in ids]
+        width = max(len(i) for i in ids)
+        masks = []
+        for i in range(len(ids)):
+            n = width - len(ids[i])
+            m = [1] * len(ids[i])
+            if self.padding_side == "left":
+                ids[i] = [self.pad_token_id] * n + ids[i]
+                masks.append([0] * n + m)
+            else:
+                ids[i] = ids[i] + [self.pad_token_id] * n
+                masks.append(m + [0] * n)
+        return self._Encoding(
+            input_ids=torch.tensor(ids, dtype=torch.long),
+            attention_mask=torch.tensor(masks, dtype=torch.long),
+        )
+
+    def save_pretrained(self, directory: str):
+        import json
+        import os
+
+        os.makedirs(directory, exist_ok=True)
+        with open(os.path.join(directory, "byte_tokenizer.json"), "w") as f:
+            json.dump({"type": "byte", "padding_side": self.padding_side,
+                       "truncation_side": self.truncation_side, "vocab_size": self.vocab_size}, f)
+
+
+def get_tokenizer(path: str, padding_side: str = "left", truncation_side: str = "right", **kwargs):
+    """Load a tokenizer from a local path; fall back to ByteTokenizer offline."""
+    import os
+
+    if path and os.path.isdir(path):
+        byte_cfg = os.path.join(path, "byte_tokenizer.json")
+        if os.path.exists(byte_cfg):
+            import json
+
+            with open(byte_cfg) as f:
+                cfg = json.load(f)
+            return ByteTokenizer(padding_side=padding_side, truncation_side=truncation_side,
+                                 vocab_size=cfg.get("vocab_size", 259))
+        try:
+            from transformers import AutoTokenizer
+
+            tok = AutoTokenizer.from_pretrained(path, **kwargs)
+            tok.padding_side = padding_side
+            tok.truncation_side = truncation_side
+            if tok.pad_token is None:
+                tok.pad_token = tok.eos_token
+            return tok
+        except Exception:
+            pass
+    return ByteTokenizer(padding_side=padding_side, truncation_side=truncation_side)
