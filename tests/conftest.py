@@ -1,0 +1,36 @@
+import pytest
+import torch
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: needs a ROCm GPU (MI355X)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip_gpu = pytest.mark.skip(reason="no GPU available")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip_gpu)
+
+
+@pytest.fixture
+def byte_tokenizer():
+    from trlx_amd.utils.tokenizer import ByteTokenizer
+
+    return ByteTokenizer()
+
+
+def tiny_config(**overrides):
+    from trlx_amd.models.nn.config import TransformerConfig
+
+    base = dict(vocab_size=300, hidden_size=64, num_layers=2, num_heads=4,
+                max_position_embeddings=256, arch_name="gpt2")
+    base.update(overrides)
+    return TransformerConfig(**base)
+
+
+@pytest.fixture
+def tiny_cfg():
+    return tiny_config()

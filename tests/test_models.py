@@ -1,0 +1,237 @@
+"""Model tests (parity: reference tests/test_models.py — forward/generate,
+save/load round-trip with mutated heads, hydra frozen-branch equivalence —
+re-targeted at the native transformer; HF-equivalence replaces downloaded
+checkpoints since there is no network)."""
+
+import pytest
+import torch
+
+from trlx_amd.models.modeling_ilql import AutoModelForCausalLMWithILQLHeads
+from trlx_amd.models.modeling_ppo import (
+    AdaptiveKLController,
+    AutoModelForCausalLMWithHydraValueHead,
+    AutoModelForCausalLMWithValueHead,
+    FixedKLController,
+)
+from trlx_amd.models.nn.config import TransformerConfig
+from trlx_amd.models.nn.convert import (
+    config_from_hf,
+    config_to_hf,
+    state_dict_from_hf,
+    state_dict_to_hf,
+)
+from trlx_amd.models.nn.generation import GenerateConfig, generate
+from trlx_amd.models.nn.transformer import CausalTransformer
+
+from conftest import tiny_config
+
+
+def test_forward_shapes(tiny_cfg):
+    m = CausalTransformer(tiny_cfg).eval()
+    ids = torch.randint(0, 300, (2, 9))
+    out = m(ids)
+    assert out.logits.shape == (2, 9, 300)
+    assert out.last_hidden_state.shape == (2, 9, 64)
+
+
+@pytest.mark.parametrize(
+    "kwargs",
+    [
+        dict(),  # gpt2-ish
+        dict(norm="rmsnorm", position_encoding="rope", swiglu=True, activation="silu",
+             attn_bias=False, mlp_bias=False, tie_word_embeddings=False, arch_name="llama",
+             num_kv_heads=2),
+        dict(position_encoding="rope", rope_interleaved=True, rope_pct=0.5,
+             parallel_residual=True, arch_name="gptj", attn_bias=False),
+        dict(position_encoding="rope", rope_pct=0.25, parallel_residual=True,
+             arch_name="gpt_neox", tie_word_embeddings=False),
+    ],
+)
+def test_kv_cache_decode_matches_full_forward(kwargs):
+    """Incremental decode with the KV cache must reproduce the full forward."""
+    torch.manual_seed(0)
+    cfg = tiny_config(**kwargs)
+    m = CausalTransformer(cfg).eval()
+    B, T = 2, 10
+    ids = torch.randint(3, 300, (B, T))
+    mask = torch.ones_like(ids)
+    mask[0, :3] = 0
+    with torch.no_grad():
+        full = m(ids, attention_mask=mask).logits
+
+        kv = m.new_kv_cache(B, T)
+        pre = m(ids[:, :5], attention_mask=mask[:, :5], kv_cache=kv, start_pos=0).logits
+        key_starts = (5 - mask[:, :5].sum(-1)).to(torch.int32)
+        step_logits = [pre[:, -1]]
+        for t in range(5, T):
+            pos = (t - key_starts).to(torch.int32).unsqueeze(1)
+            seq_lens = torch.full((B,), t + 1, dtype=torch.int32)
+            out = m(ids[:, t : t + 1], kv_cache=kv, start_pos=t, position_ids=pos,
+                    seq_lens=seq_lens, key_starts=key_starts)
+            step_logits.append(out.logits[:, -1])
+    inc = torch.stack(step_logits, dim=1)  # logits at positions 4..T-1
+    want = full[:, 4:]
+    assert torch.allclose(inc, want, atol=1e-4), (inc - want).abs().max()
+
+
+def test_generate_stops_at_eos(tiny_cfg):
+    torch.manual_seed(0)
+    m = CausalTransformer(tiny_cfg).eval()
+    ids = torch.randint(3, 300, (2, 5))
+    out = generate(m, ids, gen=GenerateConfig(max_new_tokens=8, do_sample=False, eos_token_id=None))
+    assert out.shape == (2, 13)
+    assert torch.equal(out[:, :5], ids)
+
+
+def test_hydra_frozen_branch_matches_base_at_init(tiny_cfg):
+    """At init the frozen branch is an exact copy, so reference logits must
+    equal the base logits (reference tests/test_models.py:105-130)."""
+    torch.manual_seed(0)
+    model = AutoModelForCausalLMWithHydraValueHead.from_config(tiny_cfg, num_layers_unfrozen=1)
+    model.eval()
+    ids = torch.randint(3, 300, (2, 8))
+    mask = torch.ones_like(ids)
+    mask[0, :2] = 0
+    with torch.no_grad():
+        out = model(ids, attention_mask=mask, return_ref_logits=True)
+        hydra_only = model.forward_hydra(ids, attention_mask=mask)
+    assert torch.allclose(out.ref_logits, out.logits, atol=1e-4)
+    assert torch.allclose(hydra_only.logits, out.logits, atol=1e-4)
+
+
+def test_hydra_ref_diverges_after_update(tiny_cfg):
+    torch.manual_seed(0)
+    model = AutoModelForCausalLMWithHydraValueHead.from_config(tiny_cfg, num_layers_unfrozen=1)
+    with torch.no_grad():
+        for p in model.base_model.layers[-1].parameters():
+            p.add_(torch.randn_like(p) * 0.1)
+    ids = torch.randint(3, 300, (1, 6))
+    with torch.no_grad():
+        out = model(ids, return_ref_logits=True)
+    assert not torch.allclose(out.ref_logits, out.logits, atol=1e-3)
+
+
+def test_value_head_wrapper_save_load_roundtrip(tiny_cfg, tmp_path):
+    """Mutate head weights, save, reload, verify heads survived
+    (reference tests/test_models.py:75-94)."""
+    torch.manual_seed(0)
+    model = AutoModelForCausalLMWithValueHead.from_config(tiny_cfg)
+    with torch.no_grad():
+        for p in model.v_head.parameters():
+            p.fill_(0.31)
+    model.save_pretrained(str(tmp_path / "ckpt"))
+    loaded = AutoModelForCausalLMWithValueHead.from_pretrained(str(tmp_path / "ckpt"))
+    for p in loaded.v_head.parameters():
+        assert torch.all(p == 0.31)
+    ids = torch.randint(3, 300, (1, 7))
+    with torch.no_grad():
+        a = model(ids)
+        b = loaded(ids)
+    assert torch.allclose(a.logits, b.logits, atol=1e-5)
+    assert torch.allclose(a.values, b.values, atol=1e-5)
+
+
+def test_ilql_heads_shapes_and_target_sync(tiny_cfg):
+    torch.manual_seed(0)
+    model = AutoModelForCausalLMWithILQLHeads.from_config(tiny_cfg, two_qs=True, alpha=0.5)
+    ids = torch.randint(3, 300, (2, 9))
+    actions_ixs = torch.tensor([[0, 1, 2]] * 2)
+    states_ixs = torch.tensor([[0, 1, 2, 3]] * 2)
+    out = model(ids, actions_ixs=actions_ixs, states_ixs=states_ixs)
+    assert len(out.qs) == 2
+    assert out.qs[0].shape == (2, 3, 300)
+    assert out.vs.shape == (2, 4, 1)
+
+    # Polyak sync moves target towards online
+    q0 = [p.clone() for p in model.ilql_heads.q_heads[0].parameters()]
+    with torch.no_grad():
+        for p in model.ilql_heads.q_heads[0].parameters():
+            p.add_(1.0)
+    model.sync_target_q_heads()
+    for tp, p0 in zip(model.ilql_heads.target_q_heads[0].parameters(), q0):
+        assert torch.allclose(tp, 0.5 * (p0 + 1.0) + 0.5 * p0, atol=1e-5)
+
+
+def test_ilql_shaped_generate(tiny_cfg):
+    torch.manual_seed(0)
+    model = AutoModelForCausalLMWithILQLHeads.from_config(tiny_cfg)
+    model.eval()
+    ids = torch.randint(3, 300, (2, 5))
+    out = model.generate(ids, max_new_tokens=6, beta=2, top_k=10, temperature=1.0,
+                         eos_token_id=1, pad_token_id=2)
+    assert out.shape[1] <= 11 and out.shape[0] == 2
+
+
+def test_kl_controllers():
+    adaptive = AdaptiveKLController(init_kl_coef=0.1, target=1.0, horizon=100)
+    adaptive.update(current=2.0, n_steps=10)  # above target -> coef grows
+    assert adaptive.value > 0.1
+    adaptive2 = AdaptiveKLController(init_kl_coef=0.1, target=1.0, horizon=100)
+    adaptive2.update(current=0.5, n_steps=10)  # below target -> coef shrinks
+    assert adaptive2.value < 0.1
+    fixed = FixedKLController(0.05)
+    fixed.update(3.0, 10)
+    assert fixed.value == 0.05
+
+
+# --- HF interop ------------------------------------------------------------
+
+
+@pytest.mark.parametrize("family", ["gpt2", "llama", "gptj", "gpt_neox", "opt"])
+def test_hf_equivalence(family):
+    """Native forward must match HF transformers bit-for-tolerance on the same
+    random weights — the no-network substitute for real-checkpoint tests."""
+    torch.manual_seed(0)
+    transformers = pytest.importorskip("transformers")
+    V, H, L, NH = 97, 64, 2, 4
+    if family == "gpt2":
+        hf_cfg = transformers.GPT2Config(vocab_size=V, n_embd=H, n_layer=L, n_head=NH,
+                                         n_positions=64, resid_pdrop=0, embd_pdrop=0, attn_pdrop=0)
+        hf = transformers.GPT2LMHeadModel(hf_cfg)
+    elif family == "llama":
+        hf_cfg = transformers.LlamaConfig(vocab_size=V, hidden_size=H, num_hidden_layers=L,
+                                          num_attention_heads=NH, num_key_value_heads=2,
+                                          intermediate_size=128, max_position_embeddings=64)
+        hf = transformers.LlamaForCausalLM(hf_cfg)
+    elif family == "gptj":
+        hf_cfg = transformers.GPTJConfig(vocab_size=V, n_embd=H, n_layer=L, n_head=NH,
+                                         n_positions=64, rotary_dim=8, resid_pdrop=0,
+                                         embd_pdrop=0, attn_pdrop=0)
+        hf = transformers.GPTJForCausalLM(hf_cfg)
+    elif family == "gpt_neox":
+        hf_cfg = transformers.GPTNeoXConfig(vocab_size=V, hidden_size=H, num_hidden_layers=L,
+                                            num_attention_heads=NH, intermediate_size=256,
+                                            max_position_embeddings=64, rotary_pct=0.25,
+                                            hidden_dropout=0.0, attention_dropout=0.0)
+        hf = transformers.GPTNeoXForCausalLM(hf_cfg)
+    elif family == "opt":
+        hf_cfg = transformers.OPTConfig(vocab_size=V, hidden_size=H, num_hidden_layers=L,
+                                        num_attention_heads=NH, ffn_dim=256,
+                                        max_position_embeddings=64, dropout=0.0,
+                                        word_embed_proj_dim=H)
+        hf = transformers.OPTForCausalLM(hf_cfg)
+    hf = hf.eval()
+
+    cfg = config_from_hf(hf_cfg.to_dict())
+    m = CausalTransformer(cfg).eval()
+    sd = state_dict_from_hf(cfg, hf.state_dict())
+    missing, unexpected = m.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    real_missing = [k for k in missing if not k.startswith("rope_")
+                    and not (cfg.tie_word_embeddings and k == "lm_head.weight")]
+    assert not real_missing, real_missing
+
+    ids = torch.randint(0, V, (2, 13))
+    with torch.no_grad():
+        want = hf(ids).logits
+        got = m(ids).logits
+    assert torch.allclose(got, want, atol=2e-4), (got - want).abs().max()
+
+    # state-dict export inverts import
+    back = state_dict_to_hf(cfg, {k: v for k, v in m.state_dict().items() if not k.startswith("rope_")})
+    for k, v in back.items():
+        assert torch.allclose(v, hf.state_dict()[k], atol=1e-6), k
+
+    # config export round-trips through config_from_hf
+    cfg2 = config_from_hf(config_to_hf(cfg))
+    assert cfg2.hidden_size == cfg.hidden_size and cfg2.num_layers == cfg.num_layers

@@ -1,0 +1,347 @@
+"""Numerics tests for the custom ops.
+
+CPU: the torch reference implementations against independent golden math.
+GPU (marked): the gfx950 HIP kernels against the fp32 torch references.
+"""
+
+import math
+
+import pytest
+import torch
+
+from trlx_amd import ops
+from trlx_amd.ops import reference
+
+
+# ---------------------------------------------------------------------------
+# CPU reference correctness (independent golden math)
+# ---------------------------------------------------------------------------
+
+
+def test_logprobs_of_labels_matches_log_softmax():
+    torch.manual_seed(0)
+    logits = torch.randn(4, 7, 33)
+    labels = torch.randint(0, 33, (4, 7))
+    got = reference.logprobs_of_labels(logits, labels)
+    want = torch.log_softmax(logits.float(), -1).gather(-1, labels.unsqueeze(-1)).squeeze(-1)
+    assert torch.allclose(got, want, atol=1e-6)
+
+
+def test_gae_matches_naive_loop():
+    torch.manual_seed(1)
+    B, T = 3, 11
+    values = torch.randn(B, T)
+    rewards = torch.randn(B, T)
+    gamma, lam = 0.98, 0.95
+    adv, ret = reference.gae_advantages_and_returns(values, rewards, gamma, lam, use_whitening=False)
+    # independent naive computation
+    want_adv = torch.zeros(B, T)
+    for b in range(B):
+        last = 0.0
+        for t in reversed(range(T)):
+            nextv = values[b, t + 1] if t < T - 1 else 0.0
+            delta = rewards[b, t] + gamma * nextv - values[b, t]
+            last = delta + gamma * lam * last
+            want_adv[b, t] = last
+    assert torch.allclose(adv, want_adv, atol=1e-5)
+    assert torch.allclose(ret, want_adv + values, atol=1e-5)
+
+
+def test_whiten():
+    torch.manual_seed(2)
+    xs = torch.randn(128) * 3 + 5
+    w = reference.whiten(xs)
+    assert abs(w.mean().item()) < 1e-5
+    assert abs(w.var().item() - 1.0) < 2e-2
+
+
+def test_causal_softmax_masks_future_and_padding():
+    torch.manual_seed(3)
+    B, H, T = 2, 2, 6
+    scores = torch.randn(B, H, T, T)
+    key_starts = torch.tensor([2, 0], dtype=torch.int32)
+    probs = reference.causal_softmax(scores, 0, key_starts)
+    # future masked
+    for i in range(T - 1):
+        assert probs[:, :, i, i + 1 :].abs().max() == 0
+    # left padding masked for row 0
+    assert probs[0, :, 3:, :2].abs().max() == 0
+    # rows sum to 1 where any key is valid
+    sums = probs[1, :, :, :].sum(-1)
+    assert torch.allclose(sums, torch.ones_like(sums), atol=1e-5)
+
+
+def test_rope_inverse_is_identity():
+    torch.manual_seed(4)
+    B, H, T, D = 2, 3, 5, 16
+    q = torch.randn(B, H, T, D)
+    k = torch.randn(B, H, T, D)
+    cos, sin = reference.rope_cos_sin(T, D)
+    qr, kr = ops.apply_rope(q, k, cos, sin)
+    # rotating by -theta recovers the original
+    qb, kb = ops.apply_rope(qr, kr, cos, -sin)
+    assert torch.allclose(qb, q, atol=1e-5)
+    assert torch.allclose(kb, k, atol=1e-5)
+
+
+def test_rope_partial_rotation_passthrough():
+    B, H, T, D, rot = 1, 1, 4, 16, 8
+    q = torch.randn(B, H, T, D)
+    k = torch.randn(B, H, T, D)
+    cos, sin = reference.rope_cos_sin(T, rot)
+    qr, _ = ops.apply_rope(q, k, cos, sin, rot=rot)
+    assert torch.allclose(qr[..., rot:], q[..., rot:])
+    assert not torch.allclose(qr[..., 1:2], q[..., 1:2])
+
+
+def test_sample_token_greedy_and_distribution():
+    torch.manual_seed(5)
+    logits = torch.tensor([[0.0, 5.0, 1.0], [3.0, 0.0, 0.0]])
+    assert reference.sample_token(logits, temperature=0.0).tolist() == [1, 0]
+    # statistical: sampling matches softmax
+    big = torch.tensor([[2.0, 1.0, 0.0]]).repeat(4000, 1)
+    g = torch.Generator().manual_seed(0)
+    s = reference.sample_token(big, 1.0, 0, 1.0, generator=g)
+    probs = torch.softmax(big[0], -1)
+    freq = torch.bincount(s, minlength=3).float() / len(s)
+    assert (freq - probs).abs().max() < 0.03
+
+
+def test_sample_token_top_k():
+    logits = torch.tensor([[0.0, 5.0, 4.0, -2.0]] * 50)
+    s = reference.sample_token(logits, 1.0, 2, 1.0)
+    assert set(s.tolist()) <= {1, 2}
+
+
+def test_attention_decode_reference_matches_full_attention():
+    torch.manual_seed(6)
+    B, Hq, Hkv, S, D = 2, 4, 2, 9, 8
+    q = torch.randn(B, Hq, 1, D)
+    k = torch.randn(B, Hkv, S, D)
+    v = torch.randn(B, Hkv, S, D)
+    seq_lens = torch.tensor([9, 5], dtype=torch.int32)
+    starts = torch.tensor([2, 0], dtype=torch.int32)
+    out = ops.attention_decode(q, k, v, seq_lens, 1.0 / math.sqrt(D), seq_starts=starts)
+    # manual per-row
+    for b in range(B):
+        for h in range(Hq):
+            kk = k[b, h // 2, starts[b] : seq_lens[b]].float()
+            vv = v[b, h // 2, starts[b] : seq_lens[b]].float()
+            sc = (q[b, h, 0].float() @ kk.t()) / math.sqrt(D)
+            want = torch.softmax(sc, -1) @ vv
+            assert torch.allclose(out[b, h, 0], want, atol=1e-5)
+
+
+def test_fused_adamw_cpu_matches_torch():
+    torch.manual_seed(7)
+    from trlx_amd.parallel.optim import FusedAdamW
+
+    w1 = torch.nn.Parameter(torch.randn(13, 7))
+    w2 = torch.nn.Parameter(torch.randn(5))
+    ref1 = torch.nn.Parameter(w1.detach().clone())
+    ref2 = torch.nn.Parameter(w2.detach().clone())
+    opt = FusedAdamW([w1, w2], lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.01)
+    ref_opt = torch.optim.AdamW([ref1, ref2], lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.01)
+    for step in range(5):
+        g1 = torch.randn_like(w1)
+        g2 = torch.randn_like(w2)
+        w1.grad.copy_(g1)
+        w2.grad.copy_(g2)
+        ref1.grad = g1.clone()
+        ref2.grad = g2.clone()
+        opt.step()
+        ref_opt.step()
+        opt.zero_grad()
+        ref_opt.zero_grad()
+    assert torch.allclose(w1, ref1, atol=1e-5)
+    assert torch.allclose(w2, ref2, atol=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# GPU: HIP kernels vs fp32 torch references
+# ---------------------------------------------------------------------------
+
+
+def _assert_close(got, want, atol, rtol=1e-3, name=""):
+    diff = (got.float() - want.float()).abs().max().item()
+    assert torch.allclose(got.float(), want.float(), atol=atol, rtol=rtol), f"{name}: max diff {diff}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("V", [1000, 50257])
+def test_gpu_logprobs(dtype, V):
+    torch.manual_seed(0)
+    logits = (torch.randn(64, V) * 3).to(dtype).cuda().requires_grad_(True)
+    labels = torch.randint(0, V, (64,)).cuda()
+    out = ops.logprobs_of_labels(logits, labels)
+    ref_in = logits.detach().float().cpu().requires_grad_(True)
+    want = reference.logprobs_of_labels(ref_in, labels.cpu())
+    _assert_close(out.cpu(), want, atol=5e-2 if dtype == torch.bfloat16 else 1e-4, name="logprobs fwd")
+    g = torch.randn_like(out)
+    out.backward(g)
+    want.backward(g.cpu())
+    atol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    _assert_close(logits.grad.cpu(), ref_in.grad, atol=atol, name="logprobs bwd")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("H", [768, 4096, 5000])
+def test_gpu_rmsnorm(dtype, H):
+    torch.manual_seed(1)
+    x = torch.randn(33, H).to(dtype).cuda().requires_grad_(True)
+    w = (torch.randn(H) * 0.1 + 1).to(dtype).cuda().requires_grad_(True)
+    y = ops.rmsnorm(x, w)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    yr = reference.rmsnorm(xr, wr)
+    _assert_close(y.cpu(), yr, atol=3e-2 if dtype == torch.bfloat16 else 1e-5, name="rmsnorm fwd")
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g.float().cpu())
+    atol = 8e-2 if dtype == torch.bfloat16 else 1e-3
+    _assert_close(x.grad.cpu(), xr.grad, atol=atol, name="rmsnorm dx")
+    _assert_close(w.grad.cpu(), wr.grad, atol=max(atol, 0.3 if dtype == torch.bfloat16 else 1e-2), name="rmsnorm dw")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("has_bias", [True, False])
+def test_gpu_layernorm(dtype, has_bias):
+    torch.manual_seed(2)
+    H = 768
+    x = torch.randn(57, H).to(dtype).cuda().requires_grad_(True)
+    w = (torch.randn(H) * 0.1 + 1).to(dtype).cuda().requires_grad_(True)
+    b = (torch.randn(H) * 0.1).to(dtype).cuda().requires_grad_(True) if has_bias else None
+    y = ops.layernorm(x, w, b)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    br = b.detach().float().cpu().requires_grad_(True) if has_bias else None
+    yr = reference.layernorm(xr, wr, br)
+    _assert_close(y.cpu(), yr, atol=3e-2 if dtype == torch.bfloat16 else 1e-5, name="ln fwd")
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g.float().cpu())
+    atol = 8e-2 if dtype == torch.bfloat16 else 1e-3
+    _assert_close(x.grad.cpu(), xr.grad, atol=atol, name="ln dx")
+    _assert_close(w.grad.cpu(), wr.grad, atol=max(atol, 0.3 if dtype == torch.bfloat16 else 1e-2), name="ln dw")
+    if has_bias:
+        _assert_close(b.grad.cpu(), br.grad, atol=max(atol, 0.3 if dtype == torch.bfloat16 else 1e-2), name="ln db")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("interleaved", [False, True])
+@pytest.mark.parametrize("rot_frac", [1.0, 0.25])
+def test_gpu_rope(interleaved, rot_frac):
+    torch.manual_seed(3)
+    B, H, T, D = 2, 4, 9, 64
+    rot = int(D * rot_frac)
+    q = torch.randn(B, H, T, D).bfloat16().cuda().requires_grad_(True)
+    k = torch.randn(B, H, T, D).bfloat16().cuda().requires_grad_(True)
+    cos, sin = reference.rope_cos_sin(32, rot, device="cuda")
+    positions = torch.randint(0, 32, (B, T), dtype=torch.int32).cuda()
+    qo, ko = ops.apply_rope(q, k, cos, sin, positions=positions, interleaved=interleaved, rot=rot)
+    qr = q.detach().float().cpu().requires_grad_(True)
+    kr = k.detach().float().cpu().requires_grad_(True)
+    qw, kw = ops.apply_rope(qr, kr, cos.cpu(), sin.cpu(), positions=positions.cpu(),
+                            interleaved=interleaved, rot=rot)
+    _assert_close(qo.cpu(), qw, atol=2e-2, name="rope q")
+    _assert_close(ko.cpu(), kw, atol=2e-2, name="rope k")
+    g = torch.randn_like(qo)
+    qo.backward(g)
+    qw.backward(g.float().cpu())
+    _assert_close(q.grad.cpu(), qr.grad, atol=2e-2, name="rope dq")
+
+
+@pytest.mark.gpu
+def test_gpu_gae_and_whiten():
+    torch.manual_seed(4)
+    B, T = 16, 64
+    values = torch.randn(B, T).cuda()
+    rewards = torch.randn(B, T).cuda()
+    adv, ret = ops.gae_advantages_and_returns(values, rewards, 0.99, 0.95, use_whitening=False)
+    want_adv, want_ret = reference.gae_advantages_and_returns(values.cpu(), rewards.cpu(), 0.99, 0.95,
+                                                              use_whitening=False)
+    _assert_close(adv.cpu(), want_adv, atol=1e-4, name="gae adv")
+    _assert_close(ret.cpu(), want_ret, atol=1e-4, name="gae ret")
+
+    xs = torch.randn(1000).cuda() * 2 + 3
+    w = ops.whiten(xs)
+    _assert_close(w.cpu(), reference.whiten(xs.cpu()), atol=1e-4, name="whiten")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_gpu_causal_softmax(dtype):
+    torch.manual_seed(5)
+    B, H, Tq, Tk = 2, 4, 33, 33
+    scores = (torch.randn(B, H, Tq, Tk) * 2).to(dtype).cuda().requires_grad_(True)
+    key_starts = torch.tensor([3, 0], dtype=torch.int32).cuda()
+    probs = ops.causal_softmax(scores, 0, key_starts)
+    sr = scores.detach().float().cpu().requires_grad_(True)
+    want = reference.causal_softmax(sr, 0, key_starts.cpu())
+    _assert_close(probs.cpu(), want, atol=2e-2 if dtype == torch.bfloat16 else 1e-5, name="csm fwd")
+    g = torch.randn_like(probs)
+    probs.backward(g)
+    want.backward(g.float().cpu())
+    _assert_close(scores.grad.cpu(), sr.grad, atol=2e-2 if dtype == torch.bfloat16 else 1e-4, name="csm bwd")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("D", [64, 128])
+@pytest.mark.parametrize("gqa", [1, 4])
+def test_gpu_attention_decode(D, gqa):
+    torch.manual_seed(6)
+    B, Hq, S = 4, 8, 257
+    Hkv = Hq // gqa
+    q = torch.randn(B, Hq, 1, D).bfloat16().cuda()
+    k = torch.randn(B, Hkv, S, D).bfloat16().cuda()
+    v = torch.randn(B, Hkv, S, D).bfloat16().cuda()
+    seq_lens = torch.tensor([S, 100, 31, 1], dtype=torch.int32).cuda()
+    starts = torch.tensor([0, 7, 0, 0], dtype=torch.int32).cuda()
+    out = ops.attention_decode(q, k, v, seq_lens, 1.0 / math.sqrt(D), seq_starts=starts)
+    want = ops.attention_decode(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                                seq_lens.cpu(), 1.0 / math.sqrt(D), seq_starts=starts.cpu())
+    _assert_close(out.cpu(), want, atol=3e-2, name="attn decode")
+
+
+@pytest.mark.gpu
+def test_gpu_gumbel_sampling_statistics():
+    torch.manual_seed(7)
+    V = 64
+    logits = torch.log(torch.arange(1, V + 1).float()).cuda().unsqueeze(0).repeat(8000, 1)
+    s = ops.sample_token(logits, 1.0, 0, 1.0, seed=123, offset=0)
+    probs = torch.softmax(logits[0], -1).cpu()
+    freq = torch.bincount(s.cpu(), minlength=V).float() / len(s)
+    assert (freq - probs).abs().max() < 0.02
+    # determinism: same (seed, offset) -> same tokens
+    s2 = ops.sample_token(logits, 1.0, 0, 1.0, seed=123, offset=0)
+    assert torch.equal(s, s2)
+    s3 = ops.sample_token(logits, 1.0, 0, 1.0, seed=123, offset=1)
+    assert not torch.equal(s, s3)
+
+
+@pytest.mark.gpu
+def test_gpu_gumbel_top_k():
+    logits = torch.tensor([[0.0, 5.0, 4.0, -2.0]] * 64).cuda()
+    s = ops.sample_token(logits, 1.0, 2, 1.0, seed=5, offset=0)
+    assert set(s.cpu().tolist()) <= {1, 2}
+
+
+@pytest.mark.gpu
+def test_gpu_fused_adamw_matches_cpu():
+    torch.manual_seed(8)
+    from trlx_amd.parallel.optim import FusedAdamW
+
+    w_gpu = torch.nn.Parameter(torch.randn(4099).bfloat16().cuda())
+    w_cpu = torch.nn.Parameter(w_gpu.detach().float().cpu().clone())
+    opt_g = FusedAdamW([w_gpu], lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.01)
+    opt_c = FusedAdamW([w_cpu], lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.01)
+    for _ in range(3):
+        g = torch.randn(4099)
+        w_gpu.grad.copy_(g.bfloat16().cuda())
+        w_cpu.grad.copy_(g)
+        opt_g.step()
+        opt_c.step()
+    assert (w_gpu.detach().float().cpu() - w_cpu.detach()).abs().max() < 2e-2

@@ -117,10 +117,10 @@ class ILQLConfig(MethodConfig):
         stats = dict(
             losses=dict(
                 loss=loss.item(),
-                loss_q=float(loss_q),
-                loss_v=float(loss_v),
-                loss_cql=float(loss_cql),
-                loss_awac=float(loss_awac),
+                loss_q=float(loss_q.detach()),
+                loss_v=float(loss_v.detach()),
+                loss_cql=float(loss_cql.detach()),
+                loss_awac=float(loss_awac.detach()),
             ),
             values=get_tensor_stats(V, terminal_mask, n_nonterminal),
             qvalues={str(ix): get_tensor_stats(Q[ix], terminal_mask, n_nonterminal) for ix in range(len(Q))},

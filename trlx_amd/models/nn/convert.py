@@ -318,7 +318,8 @@ def state_dict_from_hf(cfg: TransformerConfig, hf: Dict[str, torch.Tensor]) -> D
             out[o + "mlp.down_proj.bias"] = hf[p + "mlp.dense_4h_to_h.bias"]
         out["final_norm.weight"] = hf["gpt_neox.final_layer_norm.weight"]
         out["final_norm.bias"] = hf["gpt_neox.final_layer_norm.bias"]
-        out["lm_head.weight"] = hf["embed_out.weight"]
+        # transformers >=5 uses lm_head.weight; older NeoX exports embed_out
+        out["lm_head.weight"] = hf.get("lm_head.weight", hf.get("embed_out.weight"))
     elif a == "opt":
         out["embed_tokens.weight"] = hf["model.decoder.embed_tokens.weight"]
         out["embed_positions.weight"] = hf["model.decoder.embed_positions.weight"]
@@ -438,7 +439,7 @@ def state_dict_to_hf(cfg: TransformerConfig, sd: Dict[str, torch.Tensor]) -> Dic
             out[p + "mlp.dense_4h_to_h.bias"] = sd[o + "mlp.down_proj.bias"]
         out["gpt_neox.final_layer_norm.weight"] = sd["final_norm.weight"]
         out["gpt_neox.final_layer_norm.bias"] = sd["final_norm.bias"]
-        out["embed_out.weight"] = sd["lm_head.weight"]
+        out["lm_head.weight"] = sd["lm_head.weight"]
     elif a == "opt":
         qd = cfg.num_heads * cfg.head_dim
         out["model.decoder.embed_tokens.weight"] = sd["embed_tokens.weight"]

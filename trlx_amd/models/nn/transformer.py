@@ -116,7 +116,7 @@ class Attention(nn.Module):
 
         if kv_cache is not None:
             k_full, v_full = kv_cache.update(self.layer_idx, k, v, ctx.start_pos)
-            if T == 1:
+            if T == 1 and ctx.seq_lens is not None:
                 # fused flash-decode kernel
                 out = ops.attention_decode(q, k_full, v_full, ctx.seq_lens, self.scale,
                                            seq_starts=ctx.key_starts)
