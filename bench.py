@@ -1,0 +1,169 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: PPO samples/sec, GPT-2-small "sentiments" shape.
+
+BASELINE.json metric: "PPO samples/sec (whole node), GPT-2 sentiments" with
+the reference's canonical hyperparameters as the algorithmic anchor
+(default_configs: num_rollouts=128, chunk_size=128, ppo_epochs=4, batch 32,
+seq 1024, max_new_tokens=40).  One bench *step* = one full PPO outer cycle:
+collect num_rollouts experiences (generate + reward + logprob/value/ref pass)
+and run ppo_epochs optimization epochs over them — so samples/sec is the true
+end-to-end RLHF pipeline throughput, nothing skipped.
+
+Data: synthetic prompts over a 50257-token synthetic vocabulary with perfect
+decode/encode round-trip; weights: random-init GPT-2-small; reward: cheap
+deterministic function of the sample string on rank 0 (scatter to ranks, as
+the real protocol does).
+
+Launch (the driver's contract):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N --steps K --warmup W
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def build_trainer(args):
+    from trlx_amd.data.default_configs import default_ppo_config
+    from trlx_amd.models.nn.config import preset
+    from trlx_amd.pipeline.offline_pipeline import PromptPipeline
+    from trlx_amd.utils.loading import get_trainer
+
+    config = default_ppo_config()
+    config.model.model_path = args.model
+    config.model.model_extra_configs = {"config": preset(args.model).to_dict()}
+    config.model.num_layers_unfrozen = args.num_layers_unfrozen
+    config.tokenizer.tokenizer_path = "synthetic"
+    config.train.seq_length = args.seq_len
+    config.train.batch_size = args.batch_size
+    config.train.total_steps = 10**9
+    config.train.eval_interval = 10**9
+    config.train.checkpoint_interval = 10**9
+    config.train.tracker = None
+    config.train.save_best = False
+    config.method.num_rollouts = args.num_rollouts
+    config.method.chunk_size = args.chunk_size
+    config.method.ppo_epochs = args.ppo_epochs
+    config.method.gen_kwargs = dict(
+        max_new_tokens=args.max_new_tokens, top_k=0, top_p=1.0, do_sample=True
+    )
+
+    def reward_fn(samples, prompts, outputs, **kwargs):
+        # cheap deterministic stand-in for the sentiment classifier
+        return [float((len(s) * 2654435761) % 1000) / 1000.0 - 0.5 for s in samples]
+
+    trainer = get_trainer(config.train.trainer)(config=config, reward_fn=reward_fn)
+
+    torch.manual_seed(1234 + int(os.environ.get("RANK", 0)))
+    prompt_tokens = torch.randint(3, 50257, (args.num_prompts, args.prompt_len)).tolist()
+    prompts = [" ".join(f"t{t}" for t in row) for row in prompt_tokens]
+    pipeline = PromptPipeline(prompts, args.prompt_len + 2, trainer.tokenizer)
+    trainer.add_prompt_pipeline(pipeline)
+    return trainer, config
+
+
+def run_cycle(trainer, config):
+    """One PPO outer cycle: experience collection + ppo_epochs optimization."""
+    from trlx_amd.pipeline import MiniBatchIterator
+
+    trainer.store.clear_history()
+    trainer.make_experience(config.method.num_rollouts)
+    for _ in range(config.method.ppo_epochs):
+        loader = trainer.store.create_loader(config.train.batch_size, shuffle=True)
+        for minibatch in MiniBatchIterator(loader, trainer.mb_size, trainer.num_mb):
+            for microbatch in minibatch:
+                with trainer._accumulate():
+                    loss, _ = trainer.loss(microbatch)
+                    trainer.model.train()
+                    loss.backward()
+            trainer.reducer.finalize()
+            trainer.opt.step()
+            trainer.opt.zero_grad()
+            trainer.scheduler.step()
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", type=str, default="gpt2")
+    p.add_argument("--seq-len", type=int, default=1024)
+    p.add_argument("--prompt-len", type=int, default=64)
+    p.add_argument("--max-new-tokens", type=int, default=40)
+    p.add_argument("--batch-size", type=int, default=32)
+    p.add_argument("--num-rollouts", type=int, default=128)
+    p.add_argument("--chunk-size", type=int, default=128)
+    p.add_argument("--ppo-epochs", type=int, default=4)
+    p.add_argument("--num-layers-unfrozen", type=int, default=2)
+    p.add_argument("--num-prompts", type=int, default=512)
+    args = p.parse_args()
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+
+    import trlx_amd  # noqa: F401  (also loads the HIP extension path)
+    from trlx_amd.parallel import comm
+
+    trainer, config = build_trainer(args)
+
+    def sync():
+        comm.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        run_cycle(trainer, config)
+
+    sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        run_cycle(trainer, config)
+    sync()
+    elapsed = time.time() - t0
+
+    # max over ranks -> whole-job time
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=comm.get_device())
+        torch.distributed.all_reduce(t, torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_samples = args.num_rollouts * args.steps * world
+    samples_per_sec = total_samples / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "ppo_samples_per_sec",
+            "value": round(samples_per_sec, 3),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(1000.0 * elapsed / args.steps, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if torch.cuda.is_available() else "fp32",
+            "data": "synthetic prompts (vocab 50257), random-init weights, deterministic synthetic reward",
+            "config": {
+                "model": "gpt2-small-124M",
+                "global_batch": args.batch_size * world,
+                "seq_len": args.seq_len,
+                "prompt_len": args.prompt_len,
+                "max_new_tokens": args.max_new_tokens,
+                "num_rollouts_per_gpu": args.num_rollouts,
+                "ppo_epochs": args.ppo_epochs,
+                "num_layers_unfrozen": args.num_layers_unfrozen,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

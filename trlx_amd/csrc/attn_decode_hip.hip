@@ -20,7 +20,8 @@ constexpr int NWAVES = BLOCK / WAVE;
 template <int D>
 __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* __restrict__ kc,
                                    const bf16_t* __restrict__ vc, const int* __restrict__ seq_lens,
-                                   bf16_t* __restrict__ out, int Hq, int Hkv, int S, float scale) {
+                                   const int* __restrict__ seq_starts, bf16_t* __restrict__ out,
+                                   int Hq, int Hkv, int S, float scale) {
   constexpr int G = D / 8;          // lanes cooperating on one key
   constexpr int KPW = WAVE / G;     // keys per wave per iteration
   constexpr int NPART = NWAVES * KPW;  // independent (m, s, o) partials
@@ -29,6 +30,7 @@ __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* _
   const int hq = blockIdx.x % Hq;
   const int hkv = hq / (Hq / Hkv);
   const int len = seq_lens[b];
+  const int kstart = seq_starts ? seq_starts[b] : 0;
 
   const int lane = threadIdx.x % WAVE;
   const int wid = threadIdx.x / WAVE;
@@ -50,7 +52,7 @@ __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* _
 #pragma unroll
   for (int i = 0; i < 8; ++i) o[i] = 0.f;
 
-  for (int sbase = wid * KPW; sbase < len; sbase += NWAVES * KPW) {
+  for (int sbase = kstart + wid * KPW; sbase < len; sbase += NWAVES * KPW) {
     const int key = sbase + kgrp;
     float score = -INFINITY;
     float vf[8];
@@ -116,7 +118,8 @@ __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* _
 }  // namespace
 
 at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
-                            const at::Tensor& seq_lens, double scale) {
+                            const at::Tensor& seq_lens, double scale,
+                            const c10::optional<at::Tensor>& seq_starts) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16 && q.is_contiguous());
   TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous());
   TORCH_CHECK(q.dim() == 4 && q.size(2) == 1, "attention_decode: q must be [B,Hq,1,D]");
@@ -133,15 +136,22 @@ at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at:
   auto vp = reinterpret_cast<const bf16_t*>(vc.data_ptr());
   auto op = reinterpret_cast<bf16_t*>(out.data_ptr());
   auto sl = seq_lens.data_ptr<int>();
+  const int* ss = nullptr;
+  at::Tensor ssc;
+  if (seq_starts.has_value()) {
+    ssc = seq_starts->contiguous();
+    TORCH_CHECK(ssc.numel() == B && ssc.dtype() == at::kInt);
+    ss = ssc.data_ptr<int>();
+  }
   switch (D) {
     case 64:
-     hipLaunchKernelGGL(( attn_decode_kernel<64>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, op, Hq, Hkv, S, (float)scale);
+     hipLaunchKernelGGL(( attn_decode_kernel<64>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, ss, op, Hq, Hkv, S, (float)scale);
       break;
     case 128:
-     hipLaunchKernelGGL(( attn_decode_kernel<128>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, op, Hq, Hkv, S, (float)scale);
+     hipLaunchKernelGGL(( attn_decode_kernel<128>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, ss, op, Hq, Hkv, S, (float)scale);
       break;
     case 256:
-     hipLaunchKernelGGL(( attn_decode_kernel<256>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, op, Hq, Hkv, S, (float)scale);
+     hipLaunchKernelGGL(( attn_decode_kernel<256>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, ss, op, Hq, Hkv, S, (float)scale);
       break;
     default:
       TORCH_CHECK(false, "attention_decode: head dim must be 64/128/256, got ", D);

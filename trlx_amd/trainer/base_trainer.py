@@ -128,11 +128,11 @@ class NativeRLTrainer(BaseRLTrainer):
         freeze_bottom_causal_layers(base, self.config.model.num_layers_unfrozen)
         model = model.to(self.device)
         if self.dtype != torch.float32:
-            # heads stay fp32 (they were constructed fp32 on purpose)
-            base.to(self.dtype)
-            if base.rope_tables is not None:
-                base.rope_cos = base.rope_cos.float()
-                base.rope_sin = base.rope_sin.float()
+            # heads stay fp32; trunk + frozen branch go bf16
+            if hasattr(model, "cast_compute"):
+                model.cast_compute(self.dtype)
+            else:
+                base.to(self.dtype)
         return model
 
     def setup_optimizer(self):

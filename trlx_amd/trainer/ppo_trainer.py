@@ -65,7 +65,7 @@ class PPOTrainer(NativeRLTrainer):
             self.ref_model.base_model.load_state_dict(self.model.base_model.state_dict())
             self.ref_model = self.ref_model.to(self.device)
             if self.dtype != torch.float32:
-                self.ref_model.base_model.to(self.dtype)
+                self.ref_model.cast_compute(self.dtype)
             self.ref_model.eval()
 
         if config.method.target is not None:

@@ -160,10 +160,55 @@ class ByteTokenizer:
                        "truncation_side": self.truncation_side, "vocab_size": self.vocab_size}, f)
 
 
+class SyntheticVocabTokenizer(ByteTokenizer):
+    """Benchmark tokenizer: a perfect-inverse tokenizer over a full-size LM
+    vocabulary.  Token id ``i`` decodes to the string " t{i}" and encodes back
+    to exactly ``i``, so decode->reward->re-encode round-trips preserve token
+    counts — the benchmarked PPO pipeline does the true amount of work on
+    synthetic data (no downloaded vocab needed)."""
+
+    def __init__(self, vocab_size: int = 50257, padding_side: str = "left",
+                 truncation_side: str = "right"):
+        super().__init__(padding_side, truncation_side, vocab_size)
+        self.name_or_path = "synthetic-vocab-tokenizer"
+
+    def _encode_one(self, text, max_length=None, truncation=False, add_special_tokens=False):
+        specials = {self.bos_token: self.bos_token_id, self.eos_token: self.eos_token_id,
+                    self.pad_token: self.pad_token_id}
+        ids = []
+        for piece in text.split():
+            if piece in specials:
+                ids.append(specials[piece])
+            elif piece.startswith("t") and piece[1:].isdigit():
+                ids.append(min(int(piece[1:]), self.vocab_size - 1))
+            else:
+                ids.extend((b % (self.vocab_size - 3)) + 3 for b in piece.encode("utf-8"))
+        if truncation and max_length is not None and len(ids) > max_length:
+            ids = ids[-max_length:] if self.truncation_side == "left" else ids[:max_length]
+        return ids
+
+    def decode(self, ids, skip_special_tokens: bool = True):
+        if isinstance(ids, torch.Tensor):
+            ids = ids.tolist()
+        pieces = []
+        for tid in ids:
+            tid = int(tid)
+            if tid < 3:
+                if not skip_special_tokens:
+                    pieces.append([self.bos_token, self.eos_token, self.pad_token][tid])
+            else:
+                pieces.append(f"t{tid}")
+        return " ".join(pieces)
+
+
 def get_tokenizer(path: str, padding_side: str = "left", truncation_side: str = "right", **kwargs):
     """Load a tokenizer from a local path; fall back to ByteTokenizer offline."""
     import os
 
+    if path and path.startswith("synthetic"):
+        # "synthetic" or "synthetic:VOCAB"
+        vocab = int(path.split(":")[1]) if ":" in path else 50257
+        return SyntheticVocabTokenizer(vocab, padding_side, truncation_side)
     if path and os.path.isdir(path):
         byte_cfg = os.path.join(path, "byte_tokenizer.json")
         if os.path.exists(byte_cfg):
