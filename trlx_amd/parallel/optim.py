@@ -39,35 +39,36 @@ class FusedAdamW(torch.optim.Optimizer):
         self._build_arenas()
 
     def _build_arenas(self):
-        """Repoint each param group's tensors into contiguous arenas."""
+        """Repoint each param group's tensors into contiguous arenas — one
+        arena per (device, dtype) within the group (bf16 trunk and fp32 heads
+        coexist in one group)."""
         for group in self.param_groups:
             params = [p for p in group["params"] if p.requires_grad]
-            if not params:
-                self._arenas.append(None)
-                continue
-            device = params[0].device
-            dtype = params[0].dtype
-            assert all(p.device == device and p.dtype == dtype for p in params), \
-                "FusedAdamW: one device/dtype per param group"
-            total = sum(p.numel() for p in params)
-            flat_p = torch.empty(total, device=device, dtype=dtype)
-            flat_g = torch.zeros(total, device=device, dtype=dtype)
-            offset = 0
-            offsets = []
+            group_arenas = []
+            by_kind = {}
             for p in params:
-                n = p.numel()
-                flat_p[offset : offset + n].copy_(p.data.reshape(-1))
-                p.data = flat_p[offset : offset + n].view(p.shape)
-                p.grad = flat_g[offset : offset + n].view(p.shape)
-                offsets.append((offset, n))
-                offset += n
-            master = flat_p if dtype == torch.float32 else flat_p.float()
-            m = torch.zeros(total, device=device, dtype=torch.float32)
-            v = torch.zeros(total, device=device, dtype=torch.float32)
-            self._arenas.append(dict(
-                params=params, flat_p=flat_p, flat_g=flat_g, master=master, m=m, v=v,
-                offsets=offsets,
-            ))
+                by_kind.setdefault((p.device, p.dtype), []).append(p)
+            for (device, dtype), plist in by_kind.items():
+                total = sum(p.numel() for p in plist)
+                flat_p = torch.empty(total, device=device, dtype=dtype)
+                flat_g = torch.zeros(total, device=device, dtype=dtype)
+                offset = 0
+                offsets = []
+                for p in plist:
+                    n = p.numel()
+                    flat_p[offset : offset + n].copy_(p.data.reshape(-1))
+                    p.data = flat_p[offset : offset + n].view(p.shape)
+                    p.grad = flat_g[offset : offset + n].view(p.shape)
+                    offsets.append((offset, n))
+                    offset += n
+                master = flat_p if dtype == torch.float32 else flat_p.float()
+                m = torch.zeros(total, device=device, dtype=torch.float32)
+                v = torch.zeros(total, device=device, dtype=torch.float32)
+                group_arenas.append(dict(
+                    params=plist, flat_p=flat_p, flat_g=flat_g, master=master, m=m, v=v,
+                    offsets=offsets,
+                ))
+            self._arenas.append(group_arenas)
 
     # --- optimizer protocol -------------------------------------------------
 
@@ -78,19 +79,18 @@ class FusedAdamW(torch.optim.Optimizer):
             with torch.enable_grad():
                 loss = closure()
         self._step += 1
-        for group, arena in zip(self.param_groups, self._arenas):
-            if arena is None:
-                continue
+        for group, group_arenas in zip(self.param_groups, self._arenas):
             lr = group["lr"]
             beta1, beta2 = group["betas"]
             eps = group["eps"]
             wd = group["weight_decay"]
-            if arena["flat_p"].is_cuda:
-                ext = ops._require_ext("fused_adamw")
-                ext.fused_adamw(arena["flat_p"], arena["master"], arena["flat_g"], arena["m"],
-                                arena["v"], self._step, lr, beta1, beta2, eps, wd, self.grad_scale)
-            else:
-                self._cpu_step(arena, lr, beta1, beta2, eps, wd)
+            for arena in group_arenas:
+                if arena["flat_p"].is_cuda:
+                    ext = ops._require_ext("fused_adamw")
+                    ext.fused_adamw(arena["flat_p"], arena["master"], arena["flat_g"], arena["m"],
+                                    arena["v"], self._step, lr, beta1, beta2, eps, wd, self.grad_scale)
+                else:
+                    self._cpu_step(arena, lr, beta1, beta2, eps, wd)
         return loss
 
     def _cpu_step(self, arena, lr, beta1, beta2, eps, wd):
@@ -108,8 +108,8 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def zero_grad(self, set_to_none: bool = False):
         # grads are pre-pinned arena views: zero in place, never detach
-        for arena in self._arenas:
-            if arena is not None:
+        for group_arenas in self._arenas:
+            for arena in group_arenas:
                 arena["flat_g"].zero_()
 
     # --- persistence ----------------------------------------------------------
@@ -122,8 +122,8 @@ class FusedAdamW(torch.optim.Optimizer):
                 {k: v for k, v in g.items() if k != "params"} for g in self.param_groups
             ],
             "arenas": [
-                None if a is None else {"master": a["master"], "m": a["m"], "v": a["v"]}
-                for a in self._arenas
+                [{"master": a["master"], "m": a["m"], "v": a["v"]} for a in group_arenas]
+                for group_arenas in self._arenas
             ],
         }
 
@@ -132,25 +132,23 @@ class FusedAdamW(torch.optim.Optimizer):
         self.grad_scale = state.get("grad_scale", self.grad_scale)
         for g, sg in zip(self.param_groups, state["param_groups"]):
             g.update(sg)
-        for arena, sa in zip(self._arenas, state["arenas"]):
-            if arena is None or sa is None:
-                continue
-            arena["master"].copy_(sa["master"].to(arena["master"].device))
-            arena["m"].copy_(sa["m"].to(arena["m"].device))
-            arena["v"].copy_(sa["v"].to(arena["v"].device))
-            if arena["flat_p"].dtype != torch.float32:
-                arena["flat_p"].copy_(arena["master"].to(arena["flat_p"].dtype))
+        for group_arenas, sga in zip(self._arenas, state["arenas"]):
+            for arena, sa in zip(group_arenas, sga):
+                arena["master"].copy_(sa["master"].to(arena["master"].device))
+                arena["m"].copy_(sa["m"].to(arena["m"].device))
+                arena["v"].copy_(sa["v"].to(arena["v"].device))
+                if arena["flat_p"].dtype != torch.float32:
+                    arena["flat_p"].copy_(arena["master"].to(arena["flat_p"].dtype))
 
     # --- arena access for the gradient reducer --------------------------------
 
     def grad_arenas(self) -> List[Tuple[torch.Tensor, List[Tuple[torch.nn.Parameter, int, int]]]]:
         """[(flat_grad, [(param, offset, numel), ...]), ...] for bucketing."""
         out = []
-        for arena in self._arenas:
-            if arena is None:
-                continue
-            entries = [(p, off, n) for p, (off, n) in zip(arena["params"], arena["offsets"])]
-            out.append((arena["flat_g"], entries))
+        for group_arenas in self._arenas:
+            for arena in group_arenas:
+                entries = [(p, off, n) for p, (off, n) in zip(arena["params"], arena["offsets"])]
+                out.append((arena["flat_g"], entries))
         return out
 
 
