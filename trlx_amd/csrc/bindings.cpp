@@ -21,6 +21,13 @@ at::Tensor normalize(const at::Tensor& x, const at::Tensor& mean, const at::Tens
                      bool shift_mean);
 at::Tensor gumbel_sample(const at::Tensor& logits, double temperature,
                          const c10::optional<at::Tensor>& thresholds, long seed, long offset);
+at::Tensor gumbel_sample_dev(const at::Tensor& logits, double temperature,
+                             const c10::optional<at::Tensor>& thresholds, long seed,
+                             const at::Tensor& offset);
+at::Tensor decode_prep(const at::Tensor& qkv, at::Tensor& kcache, at::Tensor& vcache,
+                       const c10::optional<at::Tensor>& cos, const c10::optional<at::Tensor>& sin,
+                       const c10::optional<at::Tensor>& key_starts, const at::Tensor& cache_idx,
+                       long num_heads, long rot, bool interleaved);
 at::Tensor causal_softmax_fwd(const at::Tensor& scores, long start_pos,
                               const c10::optional<at::Tensor>& key_starts);
 at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs);
@@ -43,6 +50,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("sum_count", &sum_count);
   mod.def("normalize", &normalize);
   mod.def("gumbel_sample", &gumbel_sample);
+  mod.def("gumbel_sample_dev", &gumbel_sample_dev);
+  mod.def("decode_prep", &decode_prep);
   mod.def("causal_softmax_fwd", &causal_softmax_fwd);
   mod.def("causal_softmax_bwd", &causal_softmax_bwd);
   mod.def("attention_decode", &attention_decode);

@@ -29,7 +29,26 @@ __global__ void causal_softmax_fwd_kernel(const T* __restrict__ scores, T* __res
     const T* xr = scores + (size_t)row * Tk;
     T* pr = probs + (size_t)row * Tk;
     MS ms{-INFINITY, 0.f};
-    for (int i = kstart + threadIdx.x; i < valid; i += BLOCK) {
+    const int T8 = Tk & ~7;
+    for (int base = threadIdx.x * 8; base < T8; base += BLOCK * 8) {
+      if (base + 8 <= kstart || base >= valid) continue;
+      float v[8];
+      load8<T>(xr + base, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int j = base + i;
+        if (j < kstart || j >= valid) continue;
+        const float xi = v[i];
+        if (xi > ms.m) {
+          ms.s = ms.s * expf(ms.m - xi) + 1.f;
+          ms.m = xi;
+        } else {
+          ms.s += expf(xi - ms.m);
+        }
+      }
+    }
+    for (int i = T8 + threadIdx.x; i < valid; i += BLOCK) {
+      if (i < kstart) continue;
       const float xi = ScalarIO<T>::load(xr + i);
       if (xi > ms.m) {
         ms.s = ms.s * expf(ms.m - xi) + 1.f;
@@ -52,7 +71,17 @@ __global__ void causal_softmax_fwd_kernel(const T* __restrict__ scores, T* __res
     __syncthreads();
     const float m = bshare[0];
     const float rs = bshare[1];
-    for (int i = threadIdx.x; i < Tk; i += BLOCK) {
+    for (int base = threadIdx.x * 8; base < T8; base += BLOCK * 8) {
+      float v[8];
+      load8<T>(xr + base, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const int j = base + i;
+        v[i] = (j >= kstart && j < valid) ? expf(v[i] - m) * rs : 0.f;
+      }
+      store8<T>(pr + base, v);
+    }
+    for (int i = T8 + threadIdx.x; i < Tk; i += BLOCK) {
       float p = 0.f;
       if (i >= kstart && i < valid) p = expf(ScalarIO<T>::load(xr + i) - m) * rs;
       ScalarIO<T>::store(pr + i, p);
@@ -71,14 +100,30 @@ __global__ void causal_softmax_bwd_kernel(const T* __restrict__ probs, const T* 
     const T* dpr = dprobs + (size_t)row * Tk;
     T* dsr = dscores + (size_t)row * Tk;
     float dot = 0.f;
-    for (int i = threadIdx.x; i < Tk; i += BLOCK) {
+    const int T8 = Tk & ~7;
+    for (int base = threadIdx.x * 8; base < T8; base += BLOCK * 8) {
+      float p[8], dp[8];
+      load8<T>(pr + base, p);
+      load8<T>(dpr + base, dp);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) dot += p[i] * dp[i];
+    }
+    for (int i = T8 + threadIdx.x; i < Tk; i += BLOCK) {
       dot += ScalarIO<T>::load(pr + i) * ScalarIO<T>::load(dpr + i);
     }
     dot = block_sum<NWAVES>(dot, rbuf);
     if (threadIdx.x == 0) bshare = dot;
     __syncthreads();
     const float d = bshare;
-    for (int i = threadIdx.x; i < Tk; i += BLOCK) {
+    for (int base = threadIdx.x * 8; base < T8; base += BLOCK * 8) {
+      float p[8], dp[8];
+      load8<T>(pr + base, p);
+      load8<T>(dpr + base, dp);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) p[i] = p[i] * (dp[i] - d);
+      store8<T>(dsr + base, p);
+    }
+    for (int i = T8 + threadIdx.x; i < Tk; i += BLOCK) {
       const float p = ScalarIO<T>::load(pr + i);
       ScalarIO<T>::store(dsr + i, p * (ScalarIO<T>::load(dpr + i) - d));
     }

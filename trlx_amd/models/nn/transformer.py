@@ -56,6 +56,9 @@ class AttentionContext:
     key_starts: Optional[torch.Tensor]  # [B] int32 (left-pad offsets)
     start_pos: int  # global position of query 0 (decode steps)
     seq_lens: Optional[torch.Tensor] = None  # [B] int32 (decode: cache fill)
+    # device scalar cache write index: enables the fused decode_prep kernel
+    # and hipGraph capture (no host-side position state)
+    cache_idx: Optional[torch.Tensor] = None
 
 
 class KVCache:
@@ -103,6 +106,21 @@ class Attention(nn.Module):
     def forward(self, x, ctx: AttentionContext, rope_tables, kv_cache: Optional[KVCache] = None):
         B, T, _ = x.shape
         qkv = self.qkv_proj(x)
+
+        # fused single-token decode path: decode_prep (split+RoPE+cache
+        # append) + flash-decode attention, fully device-side (hipGraph-safe)
+        if (T == 1 and kv_cache is not None and ctx.cache_idx is not None
+                and ctx.seq_lens is not None and x.is_cuda):
+            cos_sin = rope_tables if self.cfg.position_encoding == "rope" else (None, None)
+            q = ops.decode_prep(
+                qkv, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx], ctx.cache_idx,
+                self.num_heads, cos_sin[0], cos_sin[1], ctx.key_starts, self.rot,
+                self.cfg.rope_interleaved,
+            )
+            out = ops.attention_decode(q, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx],
+                                       ctx.seq_lens, self.scale, seq_starts=ctx.key_starts)
+            return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
+
         qd = self.num_heads * self.head_dim
         kd = self.num_kv_heads * self.head_dim
         q = qkv[..., :qd].view(B, T, self.num_heads, self.head_dim).transpose(1, 2).contiguous()
@@ -253,10 +271,13 @@ class CausalTransformer(nn.Module):
 
     def make_context(self, input_ids, attention_mask, start_pos: int,
                      seq_lens: Optional[torch.Tensor] = None,
-                     key_starts: Optional[torch.Tensor] = None) -> AttentionContext:
+                     key_starts: Optional[torch.Tensor] = None,
+                     position_ids: Optional[torch.Tensor] = None) -> AttentionContext:
         B, T = input_ids.shape[:2]
         device = input_ids.device
-        if attention_mask is not None:
+        if position_ids is not None:
+            position_ids = position_ids.to(torch.int32)
+        elif attention_mask is not None:
             mask = attention_mask.to(torch.int32)
             position_ids = (mask.cumsum(-1) - 1).clamp(min=0).to(torch.int32)
             key_starts = (T - mask.sum(-1)).to(torch.int32)  # left-pad counts
@@ -287,12 +308,13 @@ class CausalTransformer(nn.Module):
         key_starts: Optional[torch.Tensor] = None,
         hidden_at_layer: Optional[int] = None,
         return_logits: bool = True,
+        cache_idx: Optional[torch.Tensor] = None,
     ) -> TransformerOutput:
         """hidden_at_layer=k stashes the hidden state FED INTO layer k
         (negative counts from the end: -2 = input of the 2nd-to-last layer)."""
-        ctx = self.make_context(input_ids, attention_mask, start_pos, seq_lens, key_starts)
-        if position_ids is not None:
-            ctx.position_ids = position_ids.to(torch.int32)
+        ctx = self.make_context(input_ids, attention_mask, start_pos, seq_lens, key_starts,
+                                position_ids)
+        ctx.cache_idx = cache_idx
         h = self.embed_tokens(input_ids)
         if self.embed_positions is not None:
             off = self.config.extra.get("position_offset", 0)

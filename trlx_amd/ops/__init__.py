@@ -300,6 +300,26 @@ def whiten(xs: torch.Tensor, shift_mean: bool = True, distributed: bool = False,
 # --------------------------------------------------------------------------
 
 
+def decode_prep(
+    qkv: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    cache_idx: torch.Tensor,
+    num_heads: int,
+    cos: Optional[torch.Tensor] = None,
+    sin: Optional[torch.Tensor] = None,
+    key_starts: Optional[torch.Tensor] = None,
+    rot: int = 0,
+    interleaved: bool = False,
+) -> torch.Tensor:
+    """Fused decode-step QKV prep: split fused qkv [B,1,QKV], RoPE q/k at
+    position (*cache_idx - key_starts[b]), append k/v to the cache at
+    *cache_idx, return q [B,Hq,1,D].  GPU-only (the eager path covers CPU)."""
+    ext = _require_ext("decode_prep")
+    return ext.decode_prep(qkv.contiguous(), k_cache, v_cache, cos, sin, key_starts, cache_idx,
+                           num_heads, rot, interleaved)
+
+
 def sample_token(
     logits: torch.Tensor,
     temperature: float = 1.0,
@@ -328,11 +348,17 @@ def sample_token(
                 # then fused gumbel sampling on the filtered logits
                 lg = _top_p_filter(logits.float() / max(temperature, 1e-6), top_p)
                 return ext.gumbel_sample(lg.contiguous(), 1.0, None, seed, offset)
+            dev_offset = isinstance(offset, torch.Tensor)
             if top_k and 0 < top_k < logits.shape[-1]:
                 lg = (logits.float() / max(temperature, 1e-6)).contiguous()
                 thr = torch.topk(lg, top_k, dim=-1).values[:, -1].contiguous()
+                if dev_offset:
+                    return ext.gumbel_sample_dev(lg, 1.0, thr, seed, offset)
                 return ext.gumbel_sample(lg, 1.0, thr, seed, offset)
-            return ext.gumbel_sample(logits.float().contiguous(), float(temperature), None, seed, offset)
+            lg = logits.float().contiguous()
+            if dev_offset:
+                return ext.gumbel_sample_dev(lg, float(temperature), None, seed, offset)
+            return ext.gumbel_sample(lg, float(temperature), None, seed, offset)
     return reference.sample_token(logits, temperature, top_k, top_p, generator)
 
 
