@@ -144,6 +144,9 @@ at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at:
     ss = ssc.data_ptr<int>();
   }
   switch (D) {
+    case 32:
+     hipLaunchKernelGGL(( attn_decode_kernel<32>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, ss, op, Hq, Hkv, S, (float)scale);
+      break;
     case 64:
      hipLaunchKernelGGL(( attn_decode_kernel<64>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, ss, op, Hq, Hkv, S, (float)scale);
       break;
@@ -154,7 +157,7 @@ at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at:
      hipLaunchKernelGGL(( attn_decode_kernel<256>), dim3(grid), dim3(BLOCK), 0, stream, qp, kp, vp, sl, ss, op, Hq, Hkv, S, (float)scale);
       break;
     default:
-      TORCH_CHECK(false, "attention_decode: head dim must be 64/128/256, got ", D);
+      TORCH_CHECK(false, "attention_decode: head dim must be 32/64/128/256, got ", D);
   }
   HIP_CHECK_LAST();
   return out;

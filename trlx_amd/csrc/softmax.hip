@@ -130,6 +130,67 @@ __global__ void causal_softmax_bwd_kernel(const T* __restrict__ probs, const T* 
   }
 }
 
+// Wave-per-row variants for short rows (Tk <= 1024): one 64-lane wave owns a
+// row — no __syncthreads, 4 rows in flight per block, scalar loads (rows this
+// short are latency- not bandwidth-bound).
+template <typename T>
+__global__ void causal_softmax_fwd_wave(const T* __restrict__ scores, T* __restrict__ probs,
+                                        const int* __restrict__ key_starts, int HTq, int Tq,
+                                        int Tk, int start_pos, long rows) {
+  const int wpb = blockDim.x / WAVE;
+  const long row0 = (long)blockIdx.x * wpb + threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const long stride = (long)gridDim.x * wpb;
+  for (long row = row0; row < rows; row += stride) {
+    const int tq = row % Tq;
+    const int kstart = key_starts ? key_starts[row / HTq] : 0;
+    const int valid = min(Tk, start_pos + tq + 1);
+    const T* xr = scores + (size_t)row * Tk;
+    T* pr = probs + (size_t)row * Tk;
+    MS ms{-INFINITY, 0.f};
+    for (int i = kstart + lane; i < valid; i += WAVE) {
+      const float xi = ScalarIO<T>::load(xr + i);
+      if (xi > ms.m) {
+        ms.s = ms.s * expf(ms.m - xi) + 1.f;
+        ms.m = xi;
+      } else {
+        ms.s += expf(xi - ms.m);
+      }
+    }
+    ms = wave_ms(ms);
+    const float m = ms.m;
+    const float rs = (ms.s > 0.f) ? 1.0f / ms.s : 0.f;
+    for (int i = lane; i < Tk; i += WAVE) {
+      float pv = 0.f;
+      if (i >= kstart && i < valid) pv = expf(ScalarIO<T>::load(xr + i) - m) * rs;
+      ScalarIO<T>::store(pr + i, pv);
+    }
+  }
+}
+
+template <typename T>
+__global__ void causal_softmax_bwd_wave(const T* __restrict__ probs, const T* __restrict__ dprobs,
+                                        T* __restrict__ dscores, int Tk, long rows) {
+  const int wpb = blockDim.x / WAVE;
+  const long row0 = (long)blockIdx.x * wpb + threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const long stride = (long)gridDim.x * wpb;
+  for (long row = row0; row < rows; row += stride) {
+    const T* pr = probs + (size_t)row * Tk;
+    const T* dpr = dprobs + (size_t)row * Tk;
+    T* dsr = dscores + (size_t)row * Tk;
+    float dot = 0.f;
+    for (int i = lane; i < Tk; i += WAVE) {
+      dot += ScalarIO<T>::load(pr + i) * ScalarIO<T>::load(dpr + i);
+    }
+    dot = wave_sum(dot);
+    for (int i = lane; i < Tk; i += WAVE) {
+      const float p = ScalarIO<T>::load(pr + i);
+      ScalarIO<T>::store(dsr + i, p * (ScalarIO<T>::load(dpr + i) - dot));
+    }
+  }
+}
+
 }  // namespace
 
 at::Tensor causal_softmax_fwd(const at::Tensor& scores, long start_pos,
@@ -149,15 +210,26 @@ at::Tensor causal_softmax_fwd(const at::Tensor& scores, long start_pos,
     ks = ksc.data_ptr<int>();
   }
   const int HTq = (int)(H * Tq);
+  const bool short_rows = Tk <= 1024;
+  const int wave_grid = (int)std::min<long>((rows + 3) / 4, 16384);
   if (scores.dtype() == at::kBFloat16) {
-    causal_softmax_fwd_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(
-        reinterpret_cast<const bf16_t*>(scores.data_ptr()),
-        reinterpret_cast<bf16_t*>(probs.data_ptr()), ks, HTq, (int)Tq, (int)Tk, (int)start_pos,
-        rows);
+    auto sp = reinterpret_cast<const bf16_t*>(scores.data_ptr());
+    auto pp = reinterpret_cast<bf16_t*>(probs.data_ptr());
+    if (short_rows)
+      causal_softmax_fwd_wave<bf16_t><<<wave_grid, BLOCK, 0, stream>>>(sp, pp, ks, HTq, (int)Tq,
+                                                                       (int)Tk, (int)start_pos, rows);
+    else
+      causal_softmax_fwd_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(sp, pp, ks, HTq, (int)Tq,
+                                                                    (int)Tk, (int)start_pos, rows);
   } else if (scores.dtype() == at::kFloat) {
-    causal_softmax_fwd_kernel<float><<<grid, BLOCK, 0, stream>>>(
-        scores.data_ptr<float>(), probs.data_ptr<float>(), ks, HTq, (int)Tq, (int)Tk,
-        (int)start_pos, rows);
+    if (short_rows)
+      causal_softmax_fwd_wave<float><<<wave_grid, BLOCK, 0, stream>>>(
+          scores.data_ptr<float>(), probs.data_ptr<float>(), ks, HTq, (int)Tq, (int)Tk,
+          (int)start_pos, rows);
+    else
+      causal_softmax_fwd_kernel<float><<<grid, BLOCK, 0, stream>>>(
+          scores.data_ptr<float>(), probs.data_ptr<float>(), ks, HTq, (int)Tq, (int)Tk,
+          (int)start_pos, rows);
   } else {
     TORCH_CHECK(false, "causal_softmax: unsupported dtype");
   }
@@ -172,15 +244,24 @@ at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs)
   if (rows == 0) return dscores;
   const int grid = (int)std::min<long>(rows, 8192);
   auto stream = c10::hip::getCurrentHIPStream();
+  const bool short_rows = Tk <= 1024;
+  const int wave_grid = (int)std::min<long>((rows + 3) / 4, 16384);
   if (probs.dtype() == at::kBFloat16) {
-    causal_softmax_bwd_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(
-        reinterpret_cast<const bf16_t*>(probs.data_ptr()),
-        reinterpret_cast<const bf16_t*>(dprobs.data_ptr()),
-        reinterpret_cast<bf16_t*>(dscores.data_ptr()), (int)Tk, rows);
+    auto pp = reinterpret_cast<const bf16_t*>(probs.data_ptr());
+    auto dpp = reinterpret_cast<const bf16_t*>(dprobs.data_ptr());
+    auto dsp = reinterpret_cast<bf16_t*>(dscores.data_ptr());
+    if (short_rows)
+      causal_softmax_bwd_wave<bf16_t><<<wave_grid, BLOCK, 0, stream>>>(pp, dpp, dsp, (int)Tk, rows);
+    else
+      causal_softmax_bwd_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(pp, dpp, dsp, (int)Tk, rows);
   } else {
-    causal_softmax_bwd_kernel<float><<<grid, BLOCK, 0, stream>>>(
-        probs.data_ptr<float>(), dprobs.data_ptr<float>(), dscores.data_ptr<float>(), (int)Tk,
-        rows);
+    if (short_rows)
+      causal_softmax_bwd_wave<float><<<wave_grid, BLOCK, 0, stream>>>(
+          probs.data_ptr<float>(), dprobs.data_ptr<float>(), dscores.data_ptr<float>(), (int)Tk, rows);
+    else
+      causal_softmax_bwd_kernel<float><<<grid, BLOCK, 0, stream>>>(
+          probs.data_ptr<float>(), dprobs.data_ptr<float>(), dscores.data_ptr<float>(), (int)Tk,
+          rows);
   }
   HIP_CHECK_LAST();
   return dscores;
