@@ -66,12 +66,19 @@ def build_trainer(args):
     return trainer, config
 
 
-def run_cycle(trainer, config):
+def run_cycle(trainer, config, phase_times=None):
     """One PPO outer cycle: experience collection + ppo_epochs optimization."""
     from trlx_amd.pipeline import MiniBatchIterator
 
+    def tick():
+        if phase_times is not None and torch.cuda.is_available():
+            torch.cuda.synchronize()
+        return time.time()
+
+    t0 = tick()
     trainer.store.clear_history()
     trainer.make_experience(config.method.num_rollouts)
+    t1 = tick()
     for _ in range(config.method.ppo_epochs):
         loader = trainer.store.create_loader(config.train.batch_size, shuffle=True)
         for minibatch in MiniBatchIterator(loader, trainer.mb_size, trainer.num_mb):
@@ -84,6 +91,12 @@ def run_cycle(trainer, config):
             trainer.opt.step()
             trainer.opt.zero_grad()
             trainer.scheduler.step()
+    t2 = tick()
+    if phase_times is not None:
+        phase_times.setdefault("experience", 0.0)
+        phase_times.setdefault("train", 0.0)
+        phase_times["experience"] += t1 - t0
+        phase_times["train"] += t2 - t1
 
 
 def main():
@@ -101,6 +114,7 @@ def main():
     p.add_argument("--ppo-epochs", type=int, default=4)
     p.add_argument("--num-layers-unfrozen", type=int, default=2)
     p.add_argument("--num-prompts", type=int, default=512)
+    p.add_argument("--phases", action="store_true", help="print per-phase times (experience vs train)")
     args = p.parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -121,11 +135,16 @@ def main():
         run_cycle(trainer, config)
 
     sync()
+    phase_times = {} if args.phases else None
     t0 = time.time()
     for _ in range(args.steps):
-        run_cycle(trainer, config)
+        run_cycle(trainer, config, phase_times)
     sync()
     elapsed = time.time() - t0
+    if args.phases and rank == 0:
+        import sys
+        per = {k: round(1000 * v / args.steps, 1) for k, v in phase_times.items()}
+        print(f"phase ms/step: {per}", file=sys.stderr)
 
     # max over ranks -> whole-job time
     if world > 1:
