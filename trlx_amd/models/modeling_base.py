@@ -133,17 +133,15 @@ class PreTrainedModelWrapper(nn.Module):
     # --- misc ---------------------------------------------------------------
 
     def cast_compute(self, dtype):
-        """Cast the compute path (trunk + frozen branch) to ``dtype`` while
-        keeping fp32 heads and the fp32 RoPE tables intact."""
-        import torch
-
-        self.base_model.to(dtype)
+        """Cast the whole compute path (trunk, frozen branch, heads) to
+        ``dtype``; RoPE tables stay fp32 (the kernels require it).  fp32
+        master weights live in the optimizer (built after this cast).
+        Matches the reference, where heads follow the model dtype
+        (modeling_ilql.py:286-289)."""
+        self.to(dtype)
         if self.base_model.rope_tables is not None:
             self.base_model.rope_cos = self.base_model.rope_cos.float()
             self.base_model.rope_sin = self.base_model.rope_sin.float()
-        frozen = getattr(self, "frozen_head", None)
-        if frozen is not None:
-            frozen.to(dtype)
         return self
 
     @property

@@ -114,13 +114,14 @@ class ILQLConfig(MethodConfig):
 
         loss = loss_q + loss_v + self.cql_scale * loss_cql + self.awac_scale * loss_awac
 
+        # stats stay device tensors (no .item() sync in the hot loop)
         stats = dict(
             losses=dict(
-                loss=loss.item(),
-                loss_q=float(loss_q.detach()),
-                loss_v=float(loss_v.detach()),
-                loss_cql=float(loss_cql.detach()),
-                loss_awac=float(loss_awac.detach()),
+                loss=loss.detach(),
+                loss_q=loss_q.detach(),
+                loss_v=loss_v.detach(),
+                loss_cql=loss_cql.detach(),
+                loss_awac=loss_awac.detach(),
             ),
             values=get_tensor_stats(V, terminal_mask, n_nonterminal),
             qvalues={str(ix): get_tensor_stats(Q[ix], terminal_mask, n_nonterminal) for ix in range(len(Q))},
@@ -149,7 +150,7 @@ class ILQLHeads(nn.Module):
 
     def forward(self, hs: torch.Tensor, states_ixs: Optional[torch.Tensor] = None,
                 actions_ixs: Optional[torch.Tensor] = None):
-        hs = hs.float()
+        hs = hs.to(self.v_head[0].weight.dtype)
         if states_ixs is not None:
             states_hs = batched_index_select(hs, states_ixs, 1)
             actions_hs = batched_index_select(hs, actions_ixs, 1)

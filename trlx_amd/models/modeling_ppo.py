@@ -148,21 +148,23 @@ class PPOConfig(MethodConfig):
 
         loss = pg_loss + self.vf_coef * vf_loss
 
+        # stats stay DEVICE TENSORS (no .item() sync in the hot loop; the
+        # tracker converts at logging time)
         stats = dict(
             losses=dict(
-                total_loss=loss.item(),
-                policy_loss=pg_loss.item(),
-                value_loss=vf_loss.item(),
+                total_loss=loss.detach(),
+                policy_loss=pg_loss.detach(),
+                value_loss=vf_loss.detach(),
             ),
             values=dict(
-                get_tensor_stats(values, mask, n),
-                values_error=torch.sum(((values - returns) * mask) ** 2) / n,
+                get_tensor_stats(values.detach(), mask, n),
+                values_error=torch.sum(((values.detach() - returns) * mask) ** 2) / n,
                 clipfrac=vf_clipfrac,
             ),
             old_values=get_tensor_stats(old_values, mask, n),
             returns=get_tensor_stats(returns, mask, n),
-            policy=dict(approx_kl=approx_kl.item(), clipfrac=pg_clipfrac.item()),
-            ratio=(ratio * mask).sum() / n,
+            policy=dict(approx_kl=approx_kl, clipfrac=pg_clipfrac),
+            ratio=(ratio.detach() * mask).sum() / n,
             padding_percentage=1 - n / mask.numel(),
         )
         return loss, flatten_dict(stats)
@@ -200,9 +202,9 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         out = self.base_model(
             input_ids, attention_mask=attention_mask, position_ids=position_ids
         )
-        values = self.v_head(out.last_hidden_state.float()).squeeze(-1)
+        values = self.v_head(out.last_hidden_state.to(self.v_head[0].weight.dtype)).squeeze(-1)
         return CausalLMOutputWithValue(
-            logits=out.logits, values=values, last_hidden_state=out.last_hidden_state
+            logits=out.logits, values=values.float(), last_hidden_state=out.last_hidden_state
         )
 
     def generate(self, input_ids, attention_mask=None, **kwargs):
@@ -270,7 +272,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
             input_ids, attention_mask=attention_mask, position_ids=position_ids,
             hidden_at_layer=stash,
         )
-        values = self.v_head(out.last_hidden_state.float()).squeeze(-1)
+        values = self.v_head(out.last_hidden_state.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         ref_logits = None
         if return_ref_logits and self.frozen_head is not None:
             ctx = self.base_model.make_context(input_ids, attention_mask, 0)
