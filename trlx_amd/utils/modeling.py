@@ -91,7 +91,8 @@ def get_tensor_stats(xs: torch.Tensor, mask: torch.Tensor, n: int) -> Dict:
     mean = (xs * mask).sum() / n
     minimum = torch.where(mask.bool(), xs, torch.full_like(xs, float("inf"))).min()
     maximum = torch.where(mask.bool(), xs, torch.full_like(xs, float("-inf"))).max()
-    std = torch.sqrt(((xs - mean) * mask).pow(2).sum() / max(n - 1, 1))
+    denom = torch.clamp(torch.as_tensor(n, dtype=torch.float32, device=xs.device) - 1, min=1)
+    std = torch.sqrt(((xs - mean) * mask).pow(2).sum() / denom)
     return dict(mean=mean, min=minimum, max=maximum, std=std)
 
 
