@@ -220,17 +220,14 @@ class TRLConfig:
         reference's ``TRLConfig.update``.
         """
         update = {}
-        # unflatten dot-paths
+        # unflatten dot-paths (dict values are allowed at any depth and merge
+        # recursively — a superset of the reference's top-level-only dicts)
         for name, value in config.items():
-            if isinstance(value, dict):
-                update[name] = value
-            else:
-                *layers, var = name.split(".")
-                if layers:
-                    d = update.setdefault(layers[0], {})
-                    for layer in layers[1:]:
-                        d = d.setdefault(layer, {})
-                    d[var] = value
+            *layers, var = name.split(".")
+            d = update
+            for layer in layers:
+                d = d.setdefault(layer, {})
+            d[var] = value
 
         if not isinstance(baseconfig, dict):
             baseconfig = baseconfig.to_dict()
