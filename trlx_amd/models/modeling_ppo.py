@@ -197,12 +197,17 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         attention_mask: Optional[torch.Tensor] = None,
         position_ids: Optional[torch.Tensor] = None,
         return_ref_logits: bool = False,
+        logits_slice=None,
         **kwargs,
     ) -> CausalLMOutputWithValue:
         out = self.base_model(
-            input_ids, attention_mask=attention_mask, position_ids=position_ids
+            input_ids, attention_mask=attention_mask, position_ids=position_ids,
+            logits_slice=logits_slice,
         )
-        values = self.v_head(out.last_hidden_state.to(self.v_head[0].weight.dtype)).squeeze(-1)
+        hs = out.last_hidden_state
+        if logits_slice is not None:
+            hs = hs[:, logits_slice[0] : logits_slice[1]]
+        values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1)
         return CausalLMOutputWithValue(
             logits=out.logits, values=values.float(), last_hidden_state=out.last_hidden_state
         )
@@ -230,11 +235,13 @@ class FrozenBranch(nn.Module):
         for p in self.parameters():
             p.requires_grad_(False)
 
-    def forward(self, hidden: torch.Tensor, ctx, rope_tables) -> torch.Tensor:
+    def forward(self, hidden: torch.Tensor, ctx, rope_tables, logits_slice=None) -> torch.Tensor:
         with torch.no_grad():
             h = hidden
             for block in self.blocks:
                 h = block(h, ctx, rope_tables)
+            if logits_slice is not None:
+                h = h[:, logits_slice[0] : logits_slice[1]]
             return self.lm_head(self.final_norm(h))
 
 
@@ -260,6 +267,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         attention_mask: Optional[torch.Tensor] = None,
         position_ids: Optional[torch.Tensor] = None,
         return_ref_logits: bool = False,
+        logits_slice=None,
         **kwargs,
     ) -> CausalLMOutputWithValue:
         """One trunk pass; optionally also the reference logits via the frozen
@@ -270,15 +278,19 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
             stash = -self.num_layers_unfrozen
         out = self.base_model(
             input_ids, attention_mask=attention_mask, position_ids=position_ids,
-            hidden_at_layer=stash,
+            hidden_at_layer=stash, logits_slice=logits_slice,
         )
-        values = self.v_head(out.last_hidden_state.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        hs = out.last_hidden_state
+        if logits_slice is not None:
+            hs = hs[:, logits_slice[0] : logits_slice[1]]
+        values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         ref_logits = None
         if return_ref_logits and self.frozen_head is not None:
             ctx = self.base_model.make_context(input_ids, attention_mask, 0)
             if position_ids is not None:
                 ctx.position_ids = position_ids.to(torch.int32)
-            ref_logits = self.frozen_head(out.hidden_at_layer, ctx, self.base_model.rope_tables)
+            ref_logits = self.frozen_head(out.hidden_at_layer, ctx, self.base_model.rope_tables,
+                                          logits_slice=logits_slice)
         return CausalLMOutputWithValue(
             logits=out.logits, values=values, ref_logits=ref_logits,
             last_hidden_state=out.last_hidden_state,

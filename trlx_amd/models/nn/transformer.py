@@ -309,6 +309,7 @@ class CausalTransformer(nn.Module):
         hidden_at_layer: Optional[int] = None,
         return_logits: bool = True,
         cache_idx: Optional[torch.Tensor] = None,
+        logits_slice: Optional[Tuple[int, int]] = None,
     ) -> TransformerOutput:
         """hidden_at_layer=k stashes the hidden state FED INTO layer k
         (negative counts from the end: -2 = input of the 2nd-to-last layer)."""
@@ -332,7 +333,13 @@ class CausalTransformer(nn.Module):
                 hidden_at = h
             h = layer(h, ctx, self.rope_tables, kv_cache)
         h = self.final_norm(h)
-        logits = self.lm_head(h) if return_logits else None
+        logits = None
+        if return_logits:
+            # logits_slice: compute the [V]-wide projection only where the
+            # loss needs it (PPO: response positions — the lm_head GEMM and
+            # its backward are the largest single kernels in the train step)
+            hs = h if logits_slice is None else h[:, logits_slice[0] : logits_slice[1]]
+            logits = self.lm_head(hs)
         return TransformerOutput(logits=logits, last_hidden_state=h, hidden_at_layer=hidden_at)
 
     def new_kv_cache(self, batch: int, max_len: int, device=None, dtype=None) -> KVCache:
