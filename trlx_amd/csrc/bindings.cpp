@@ -28,6 +28,15 @@ at::Tensor decode_prep(const at::Tensor& qkv, at::Tensor& kcache, at::Tensor& vc
                        const c10::optional<at::Tensor>& cos, const c10::optional<at::Tensor>& sin,
                        const c10::optional<at::Tensor>& key_starts, const at::Tensor& cache_idx,
                        long num_heads, long rot, bool interleaved);
+std::vector<at::Tensor> qkv_prep_fwd(const at::Tensor& qkv, long num_heads, long num_kv_heads,
+                                     long head_dim, const c10::optional<at::Tensor>& cos,
+                                     const c10::optional<at::Tensor>& sin,
+                                     const c10::optional<at::Tensor>& pos, double qscale,
+                                     long rot, bool interleaved);
+at::Tensor qkv_prep_bwd(const at::Tensor& dq, const at::Tensor& dk, const at::Tensor& dv,
+                        const c10::optional<at::Tensor>& cos, const c10::optional<at::Tensor>& sin,
+                        const c10::optional<at::Tensor>& pos, double qscale, long rot,
+                        bool interleaved);
 at::Tensor causal_softmax_fwd(const at::Tensor& scores, long start_pos,
                               const c10::optional<at::Tensor>& key_starts);
 at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs);
@@ -52,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gumbel_sample", &gumbel_sample);
   mod.def("gumbel_sample_dev", &gumbel_sample_dev);
   mod.def("decode_prep", &decode_prep);
+  mod.def("qkv_prep_fwd", &qkv_prep_fwd);
+  mod.def("qkv_prep_bwd", &qkv_prep_bwd);
   mod.def("causal_softmax_fwd", &causal_softmax_fwd);
   mod.def("causal_softmax_bwd", &causal_softmax_bwd);
   mod.def("attention_decode", &attention_decode);

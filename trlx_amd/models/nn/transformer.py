@@ -121,18 +121,33 @@ class Attention(nn.Module):
                                        ctx.seq_lens, self.scale, seq_starts=ctx.key_starts)
             return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
 
-        qd = self.num_heads * self.head_dim
-        kd = self.num_kv_heads * self.head_dim
-        q = qkv[..., :qd].view(B, T, self.num_heads, self.head_dim).transpose(1, 2).contiguous()
-        k = qkv[..., qd : qd + kd].view(B, T, self.num_kv_heads, self.head_dim).transpose(1, 2).contiguous()
-        v = qkv[..., qd + kd :].view(B, T, self.num_kv_heads, self.head_dim).transpose(1, 2).contiguous()
+        if x.is_cuda and x.dtype == torch.bfloat16 and ops.extension_available():
+            # fused split + RoPE + q-scale (one kernel fwd, one bwd)
+            use_rope = self.cfg.position_encoding == "rope"
+            cos, sin = rope_tables if use_rope else (None, None)
+            q, k, v = ops.qkv_prep(
+                qkv, self.num_heads, self.num_kv_heads, self.head_dim, cos, sin,
+                positions=ctx.position_ids if use_rope else None, qscale=self.scale,
+                rot=self.rot, interleaved=self.cfg.rope_interleaved,
+            )
+            pre_scaled = True
+        else:
+            qd = self.num_heads * self.head_dim
+            kd = self.num_kv_heads * self.head_dim
+            q = qkv[..., :qd].view(B, T, self.num_heads, self.head_dim).transpose(1, 2).contiguous()
+            k = qkv[..., qd : qd + kd].view(B, T, self.num_kv_heads, self.head_dim).transpose(1, 2).contiguous()
+            v = qkv[..., qd + kd :].view(B, T, self.num_kv_heads, self.head_dim).transpose(1, 2).contiguous()
+            pre_scaled = False
 
-        if self.cfg.position_encoding == "rope":
-            cos, sin = rope_tables
-            q, k = ops.apply_rope(q, k, cos, sin, positions=ctx.position_ids,
-                                  interleaved=self.cfg.rope_interleaved, rot=self.rot)
+            if self.cfg.position_encoding == "rope":
+                cos, sin = rope_tables
+                q, k = ops.apply_rope(q, k, cos, sin, positions=ctx.position_ids,
+                                      interleaved=self.cfg.rope_interleaved, rot=self.rot)
 
         if kv_cache is not None:
+            if pre_scaled:
+                # cache stores unscaled k (decode kernel scales q itself)
+                pass
             k_full, v_full = kv_cache.update(self.layer_idx, k, v, ctx.start_pos)
             if T == 1 and ctx.seq_lens is not None:
                 # fused flash-decode kernel
@@ -148,7 +163,8 @@ class Attention(nn.Module):
             rep = self.num_heads // self.num_kv_heads
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
-        scores = torch.matmul(q * self.scale, k.transpose(-1, -2))
+        qs = q if pre_scaled else q * self.scale
+        scores = torch.matmul(qs, k.transpose(-1, -2))
         probs = ops.causal_softmax(scores.contiguous(), ctx.start_pos, ctx.key_starts)
         if self.attn_pdrop > 0 and self.training:
             probs = F.dropout(probs, self.attn_pdrop)

@@ -320,6 +320,51 @@ def decode_prep(
                            num_heads, rot, interleaved)
 
 
+class _QKVPrep(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, num_heads, num_kv_heads, head_dim, cos, sin, pos, qscale, rot,
+                interleaved, ext):
+        q, k, v = ext.qkv_prep_fwd(qkv, num_heads, num_kv_heads, head_dim, cos, sin, pos,
+                                   qscale, rot, interleaved)
+        ctx.save_for_backward(*( [cos, sin, pos] if cos is not None else [] ))
+        ctx.has_rope = cos is not None
+        ctx.qscale = qscale
+        ctx.rot = rot
+        ctx.interleaved = interleaved
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        ext = _require_ext("qkv_prep")
+        cos = sin = pos = None
+        if ctx.has_rope:
+            cos, sin, pos = ctx.saved_tensors
+        dqkv = ext.qkv_prep_bwd(dq.contiguous(), dk.contiguous(), dv.contiguous(), cos, sin,
+                                pos, ctx.qscale, ctx.rot, ctx.interleaved)
+        return dqkv, None, None, None, None, None, None, None, None, None, None
+
+
+def qkv_prep(
+    qkv: torch.Tensor,
+    num_heads: int,
+    num_kv_heads: int,
+    head_dim: int,
+    cos: Optional[torch.Tensor] = None,
+    sin: Optional[torch.Tensor] = None,
+    positions: Optional[torch.Tensor] = None,
+    qscale: float = 1.0,
+    rot: int = 0,
+    interleaved: bool = False,
+):
+    """Fused split + RoPE + q-scale for the T>1 path (GPU-only; autograd-
+    aware — backward applies the inverse rotation and re-packs grads)."""
+    ext = _require_ext("qkv_prep")
+    if positions is not None:
+        positions = positions.to(torch.int32).contiguous()
+    return _QKVPrep.apply(qkv.contiguous(), num_heads, num_kv_heads, head_dim, cos, sin,
+                          positions, qscale, rot, interleaved, ext)
+
+
 def sample_token(
     logits: torch.Tensor,
     temperature: float = 1.0,
