@@ -145,6 +145,9 @@ def main():
         import sys
         per = {k: round(1000 * v / args.steps, 1) for k, v in phase_times.items()}
         print(f"phase ms/step: {per}", file=sys.stderr)
+        exp = {k: round(1000 * float(v), 1) for k, v in
+               getattr(trainer, "last_experience_stats", {}).items() if k.startswith("time/")}
+        print(f"experience sub-phases ms: {exp}", file=sys.stderr)
 
     # max over ranks -> whole-job time
     if world > 1:

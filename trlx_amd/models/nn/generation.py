@@ -255,9 +255,14 @@ def _generate_eager(model, input_ids, attention_mask, gen: GenerateConfig, shapi
         start_pos = T + step
         position_ids = (start_pos - key_starts).to(torch.int32).unsqueeze(1)
         seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=device)
+        # on GPU route through the same fused decode kernels as the graph
+        # engine (identical numerics)
+        cache_idx = None
+        if device.type == "cuda" and ops.extension_available():
+            cache_idx = torch.full((1,), start_pos, dtype=torch.long, device=device)
         out = model(
             next_tok.unsqueeze(1), kv_cache=kv, start_pos=start_pos, position_ids=position_ids,
-            seq_lens=seq_lens, key_starts=key_starts, return_logits=False,
+            seq_lens=seq_lens, key_starts=key_starts, return_logits=False, cache_idx=cache_idx,
         )
         logits_last = model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0]
         hidden_last = out.last_hidden_state[:, -1]

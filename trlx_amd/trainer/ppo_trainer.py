@@ -351,6 +351,7 @@ class PPOTrainer(NativeRLTrainer):
                  for k in accumulated_stats[-1]}
         stats["kl_ctl_value"] = self.kl_ctl.value
         self.mean_kl = stats["policy/sqrt_kl"] ** 2
+        self.last_experience_stats = stats
         self.tracker.log(stats, step=iter_count)
 
         self.push_to_store(ppo_rl_elements)
