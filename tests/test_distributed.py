@@ -193,3 +193,42 @@ def test_grad_reducer_no_sync():
 
 def test_distributed_ppo_end_to_end():
     _spawn(_worker_ppo_train, 29515)
+
+
+def _worker_zero_equivalence(rank):
+    """ZeRO sharded optimizer must produce the same weights as replicated DP."""
+    from trlx_amd.parallel.ddp import GradReducer
+    from trlx_amd.parallel.optim import FusedAdamW
+
+    torch.manual_seed(11)
+    model_z = torch.nn.Sequential(torch.nn.Linear(9, 16), torch.nn.Tanh(), torch.nn.Linear(16, 3))
+    model_r = torch.nn.Sequential(torch.nn.Linear(9, 16), torch.nn.Tanh(), torch.nn.Linear(16, 3))
+    model_r.load_state_dict(model_z.state_dict())
+
+    opt_z = FusedAdamW(list(model_z.parameters()), lr=1e-2, weight_decay=0.01,
+                       grad_scale=1.0 / WORLD, zero=True)
+    red_z = GradReducer(opt_z, model_z, zero=True)
+    opt_r = FusedAdamW(list(model_r.parameters()), lr=1e-2, weight_decay=0.01,
+                       grad_scale=1.0 / WORLD)
+    red_r = GradReducer(opt_r, model_r, bucket_size_mb=1)
+
+    for step in range(3):
+        g = torch.Generator().manual_seed(50 + step)
+        x = torch.randn(4, 9, generator=g) + rank
+        y = torch.randn(4, 3, generator=g)
+        for model, opt, red in ((model_z, opt_z, red_z), (model_r, opt_r, red_r)):
+            loss = ((model(x) - y) ** 2).sum()
+            loss.backward()
+            red.finalize()
+            opt.step()
+            opt.zero_grad()
+
+    for pz, pr in zip(model_z.parameters(), model_r.parameters()):
+        assert torch.allclose(pz, pr, atol=1e-5), (pz - pr).abs().max()
+    # shard memory is 1/WORLD of the arena
+    arena = opt_z._arenas[0][0]
+    assert arena["master"].numel() * WORLD == arena["flat_p"].numel()
+
+
+def test_zero_sharded_optimizer_equivalence():
+    _spawn(_worker_zero_equivalence, 29516)

@@ -36,7 +36,7 @@ __global__ void adamw_kernel(PT* __restrict__ p, float* __restrict__ master,
       w -= lr * wd * w;
       w -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
       master[i] = w;
-      if (!std::is_same<PT, float>::value) ScalarIO<PT>::store(p + i, w);
+      ScalarIO<PT>::store(p + i, w);  // aliased (fp32, non-ZeRO) or sharded
     }
   }
 }
@@ -72,7 +72,6 @@ void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Ten
     }
   } else if (p.dtype() == at::kFloat) {
     auto pp = p.data_ptr<float>();
-    TORCH_CHECK(p.data_ptr() == master.data_ptr(), "fp32 params must alias their master arena");
     if (g.dtype() == at::kBFloat16) {
       LAUNCH_ADAMW(float, bf16_t, pp, reinterpret_cast<const bf16_t*>(g.data_ptr()));
     } else {

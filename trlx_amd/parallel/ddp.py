@@ -37,17 +37,22 @@ class GradReducer:
     """Bucketed async gradient all-reduce driven by post-accumulate hooks."""
 
     def __init__(self, optimizer, model: torch.nn.Module, bucket_size_mb: int = 128,
-                 process_group=None, average: bool = False):
+                 process_group=None, average: bool = False, zero: bool = False):
         """``average=False`` assumes the optimizer fuses the 1/N scaling
-        (FusedAdamW.grad_scale); ``average=True`` divides after reduction."""
+        (FusedAdamW.grad_scale); ``average=True`` divides after reduction.
+        ``zero=True``: instead of bucketed all-reduce, finalize() reduce-
+        scatters each grad arena into this rank's shard (the sharded optimizer
+        consumes only the shard)."""
         self.enabled = dist.is_initialized() and dist.get_world_size(process_group) > 1
         self.group = process_group
         self.average = average
+        self.zero = zero and self.enabled
+        self.optimizer = optimizer
         self.buckets: List[Bucket] = []
         self._param_bucket = {}
         self._sync = True
         self._hooks = []
-        if not self.enabled:
+        if not self.enabled or self.zero:
             return
 
         bucket_numel_budget = bucket_size_mb * 1024 * 1024 // 2  # bf16 elems
@@ -96,6 +101,22 @@ class GradReducer:
         if not self.enabled:
             return
         world = dist.get_world_size(self.group)
+        if self.zero:
+            # ZeRO: one reduce-scatter per grad arena; the sharded optimizer
+            # reads only [lo, lo+shard)
+            for group_arenas in self.optimizer._arenas:
+                for arena in group_arenas:
+                    flat_g = arena["flat_g"]
+                    lo, shard = arena["lo"], arena["shard"]
+                    if dist.get_backend(self.group) == "gloo":
+                        dist.all_reduce(flat_g, dist.ReduceOp.SUM, group=self.group)
+                    else:
+                        out = torch.empty_like(flat_g[lo : lo + shard])
+                        dist.reduce_scatter_tensor(out, flat_g, dist.ReduceOp.SUM, group=self.group)
+                        flat_g[lo : lo + shard].copy_(out)
+                    if self.average:
+                        flat_g[lo : lo + shard].div_(world)
+            return
         for bucket in self.buckets:
             if bucket.handle is not None:
                 bucket.handle.wait()

@@ -82,6 +82,7 @@ class NativeRLTrainer(BaseRLTrainer):
         self.reducer = GradReducer(
             self.opt, self.model, bucket_size_mb=config.train.bucket_size_mb,
             average=not isinstance(self.opt, FusedAdamW),
+            zero=config.train.zero_stage >= 1 and isinstance(self.opt, FusedAdamW),
         )
         self.reducer.broadcast_parameters(self.model)
 
@@ -138,7 +139,7 @@ class NativeRLTrainer(BaseRLTrainer):
     def setup_optimizer(self):
         return build_optimizer(
             self.model, self.config.optimizer.name, self.config.optimizer.kwargs,
-            world=self.world_size,
+            world=self.world_size, zero=self.config.train.zero_stage >= 1,
         )
 
     def setup_scheduler(self):
