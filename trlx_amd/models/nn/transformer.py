@@ -28,6 +28,8 @@ import torch.nn.functional as F
 
 from ... import ops
 from ...ops.reference import rope_cos_sin
+from ...parallel import topo
+from ...parallel.tp import ColumnParallelLinear, RowParallelLinear
 from .config import TransformerConfig
 
 
@@ -92,12 +94,22 @@ class Attention(nn.Module):
         super().__init__()
         self.cfg = cfg
         self.layer_idx = layer_idx
-        self.num_heads = cfg.num_heads
-        self.num_kv_heads = cfg.num_kv_heads
+        tp = topo.tp_size()
+        assert cfg.num_heads % tp == 0 and cfg.num_kv_heads % tp == 0, \
+            f"heads ({cfg.num_heads}/{cfg.num_kv_heads}) must divide tp={tp}"
+        # local (per-TP-rank) head counts; attention runs on the local shard
+        self.num_heads = cfg.num_heads // tp
+        self.num_kv_heads = cfg.num_kv_heads // tp
         self.head_dim = cfg.head_dim
         self.scale = cfg.attn_scale if cfg.attn_scale is not None else 1.0 / math.sqrt(cfg.head_dim)
-        self.qkv_proj = nn.Linear(cfg.hidden_size, cfg.qkv_out, bias=cfg.attn_bias)
-        self.o_proj = nn.Linear(cfg.num_heads * cfg.head_dim, cfg.hidden_size, bias=cfg.attn_bias)
+        if tp > 1:
+            # column-parallel qkv (head-aligned shard), row-parallel output
+            self.qkv_proj = ColumnParallelLinear(cfg.hidden_size, cfg.qkv_out, bias=cfg.attn_bias)
+            self.o_proj = RowParallelLinear(cfg.num_heads * cfg.head_dim, cfg.hidden_size,
+                                            bias=cfg.attn_bias)
+        else:
+            self.qkv_proj = nn.Linear(cfg.hidden_size, cfg.qkv_out, bias=cfg.attn_bias)
+            self.o_proj = nn.Linear(cfg.num_heads * cfg.head_dim, cfg.hidden_size, bias=cfg.attn_bias)
         self.rot = int(cfg.head_dim * cfg.rope_pct) if cfg.position_encoding == "rope" else 0
         if self.rot % 2:
             self.rot -= 1
@@ -185,14 +197,23 @@ class MLP(nn.Module):
     def __init__(self, cfg: TransformerConfig):
         super().__init__()
         self.swiglu = cfg.swiglu
+        tp = topo.tp_size()
         i = cfg.intermediate_size
-        if cfg.swiglu:
-            self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * i, bias=cfg.mlp_bias)
+        assert i % tp == 0, f"intermediate {i} must divide tp={tp}"
+        if tp > 1:
+            if cfg.swiglu:
+                self.gate_up_proj = ColumnParallelLinear(cfg.hidden_size, 2 * i, bias=cfg.mlp_bias)
+            else:
+                self.fc_in = ColumnParallelLinear(cfg.hidden_size, i, bias=cfg.mlp_bias)
+            self.down_proj = RowParallelLinear(i, cfg.hidden_size, bias=cfg.mlp_bias)
         else:
-            self.fc_in = nn.Linear(cfg.hidden_size, i, bias=cfg.mlp_bias)
-        self.down_proj = nn.Linear(i, cfg.hidden_size, bias=cfg.mlp_bias)
+            if cfg.swiglu:
+                self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * i, bias=cfg.mlp_bias)
+            else:
+                self.fc_in = nn.Linear(cfg.hidden_size, i, bias=cfg.mlp_bias)
+            self.down_proj = nn.Linear(i, cfg.hidden_size, bias=cfg.mlp_bias)
         self.act = _ACTS[cfg.activation]
-        self.isize = i
+        self.isize = i // tp
 
     def forward(self, x):
         if self.swiglu:
@@ -266,7 +287,7 @@ class CausalTransformer(nn.Module):
         self.apply(self._init_weights)
 
     def _init_weights(self, module):
-        if isinstance(module, nn.Linear):
+        if isinstance(module, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):
             module.weight.data.normal_(mean=0.0, std=0.02)
             if module.bias is not None:
                 module.bias.data.zero_()
@@ -360,7 +381,8 @@ class CausalTransformer(nn.Module):
 
     def new_kv_cache(self, batch: int, max_len: int, device=None, dtype=None) -> KVCache:
         p = next(self.parameters())
-        return KVCache(len(self.layers), batch, self.config.num_kv_heads, max_len,
+        kv_heads_local = self.config.num_kv_heads // topo.tp_size()
+        return KVCache(len(self.layers), batch, kv_heads_local, max_len,
                        self.config.head_dim, device or p.device, dtype or p.dtype)
 
     def num_parameters(self) -> int:

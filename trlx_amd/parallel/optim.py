@@ -193,7 +193,10 @@ def build_optimizer(model: torch.nn.Module, name: str, kwargs: dict, world: int 
     params = [p for p in model.parameters() if p.requires_grad]
     kwargs = dict(kwargs)
     if name in ("fused_adamw", "adamw"):
-        return FusedAdamW(params, grad_scale=1.0 / world, zero=zero, **kwargs)
+        from . import topo
+
+        return FusedAdamW(params, grad_scale=1.0 / world, zero=zero,
+                          zero_group=topo.dp_group(), **kwargs)
     if name == "adam":
         return torch.optim.Adam(params, **kwargs)
     if name == "sgd":

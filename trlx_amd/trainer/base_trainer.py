@@ -47,9 +47,22 @@ class NativeRLTrainer(BaseRLTrainer):
         super().__init__(config, **kwargs)
 
         comm.init_distributed()
+        from ..parallel import topo
+
+        topo.init_model_parallel(config.train.tensor_parallel_size,
+                                 config.train.pipeline_parallel_size)
         self.device = comm.get_device()
         self.world_size = comm.world_size()
+        self.dp_size = topo.dp_size()
+        self.dp_group = topo.dp_group()
+        self.tp_size = topo.tp_size()
         self.local_rank = comm.local_rank()
+        # re-seed with the DP rank so TP peers share RNG streams (identical
+        # rollouts within a TP group, decorrelated across DP replicas —
+        # reference modeling_nemo_ppo.py:384-393)
+        from ..utils import set_seed
+
+        set_seed(config.train.seed, rank_offset=topo.dp_rank())
         comm.barrier()
 
         self.mb_count = 0
@@ -83,6 +96,7 @@ class NativeRLTrainer(BaseRLTrainer):
             self.opt, self.model, bucket_size_mb=config.train.bucket_size_mb,
             average=not isinstance(self.opt, FusedAdamW),
             zero=config.train.zero_stage >= 1 and isinstance(self.opt, FusedAdamW),
+            process_group=self.dp_group,
         )
         self.reducer.broadcast_parameters(self.model)
 
@@ -137,9 +151,11 @@ class NativeRLTrainer(BaseRLTrainer):
         return model
 
     def setup_optimizer(self):
+        from ..parallel import topo
+
         return build_optimizer(
             self.model, self.config.optimizer.name, self.config.optimizer.kwargs,
-            world=self.world_size, zero=self.config.train.zero_stage >= 1,
+            world=topo.dp_size(), zero=self.config.train.zero_stage >= 1,
         )
 
     def setup_scheduler(self):

@@ -200,40 +200,61 @@ class PPOTrainer(NativeRLTrainer):
             prompt_tensors = batch["input_ids"].to(device)
             prompt_sizes = torch.full((len(prompt_tensors),), prompt_tensors.shape[1],
                                       device=device, dtype=torch.long)
+            metadata_local = {k: v for k, v in batch.items()
+                              if k not in ("input_ids", "attention_mask")}
 
-            padded_samples = comm.pad_across_processes(samples, 1, self.tokenizer.eos_token_id)
-            padded_prompts = comm.pad_across_processes(prompt_tensors, 1, self.tokenizer.eos_token_id)
-            gathered_samples = comm.gather(padded_samples)
-            gathered_prompts = comm.gather(padded_prompts)
-            gathered_prompt_sizes = comm.gather(prompt_sizes)
-            metadata = gather_dict({k: v for k, v in batch.items()
-                                    if k not in ("input_ids", "attention_mask")})
+            from ..parallel import topo
 
-            if comm.is_main_process():
-                all_str_samples, all_str_prompts, all_str_outputs = self.decode(
-                    gathered_prompts, gathered_samples, gathered_prompt_sizes, append_eos_token=True
-                )
+            if topo.tp_size() > 1:
+                # tensor-parallel mode: every rank scores its own samples
+                # locally (TP peers hold identical rollouts) — the NeMo-path
+                # protocol (reference nemo_ppo_trainer.py:195-197)
+                l_samples, l_prompts, l_outputs = self.decode(
+                    prompt_tensors, samples, append_eos_token=True)
                 rollout_score_time = time()
-                all_scores = self.reward_fn(
-                    samples=all_str_samples, prompts=all_str_prompts, outputs=all_str_outputs,
-                    tokenizer=self.tokenizer, **metadata,
+                local_scores = self.reward_fn(
+                    samples=l_samples, prompts=l_prompts, outputs=l_outputs,
+                    tokenizer=self.tokenizer, **metadata_local,
                 )
-                all_scores = [torch.tensor(score, dtype=torch.float, device=device).view(-1)
-                              for score in all_scores]
-                all_scores = pad_sequence(all_scores, batch_first=True, padding_value=-np.inf)
-                max_len = torch.tensor(all_scores.shape[1], dtype=torch.long, device=device)
+                local_scores = [torch.tensor(score, dtype=torch.float, device=device).view(-1)
+                                for score in local_scores]
+                scores = pad_sequence(local_scores, batch_first=True, padding_value=-np.inf)
                 stats["time/rollout_score"] = time() - rollout_score_time
-                all_scores = list(all_scores.reshape(self.world_size, -1, max_len).unbind())
             else:
-                all_scores = None
-                max_len = torch.tensor(0, dtype=torch.long, device=device)
+                # accelerate-path protocol: gather everything, score on rank 0,
+                # scatter per-rank chunks (reference accelerate_ppo_trainer.py:292-338)
+                padded_samples = comm.pad_across_processes(samples, 1, self.tokenizer.eos_token_id)
+                padded_prompts = comm.pad_across_processes(prompt_tensors, 1, self.tokenizer.eos_token_id)
+                gathered_samples = comm.gather(padded_samples)
+                gathered_prompts = comm.gather(padded_prompts)
+                gathered_prompt_sizes = comm.gather(prompt_sizes)
+                metadata = gather_dict(metadata_local)
 
-            if torch.distributed.is_initialized():
-                torch.distributed.broadcast(max_len, 0)
-                scores = torch.empty((len(samples), max_len), device=device)
-                torch.distributed.scatter(scores, all_scores)
-            else:
-                scores = all_scores[0].clone().detach()
+                if comm.is_main_process():
+                    all_str_samples, all_str_prompts, all_str_outputs = self.decode(
+                        gathered_prompts, gathered_samples, gathered_prompt_sizes, append_eos_token=True
+                    )
+                    rollout_score_time = time()
+                    all_scores = self.reward_fn(
+                        samples=all_str_samples, prompts=all_str_prompts, outputs=all_str_outputs,
+                        tokenizer=self.tokenizer, **metadata,
+                    )
+                    all_scores = [torch.tensor(score, dtype=torch.float, device=device).view(-1)
+                                  for score in all_scores]
+                    all_scores = pad_sequence(all_scores, batch_first=True, padding_value=-np.inf)
+                    max_len = torch.tensor(all_scores.shape[1], dtype=torch.long, device=device)
+                    stats["time/rollout_score"] = time() - rollout_score_time
+                    all_scores = list(all_scores.reshape(self.world_size, -1, max_len).unbind())
+                else:
+                    all_scores = None
+                    max_len = torch.tensor(0, dtype=torch.long, device=device)
+
+                if torch.distributed.is_initialized():
+                    torch.distributed.broadcast(max_len, 0)
+                    scores = torch.empty((len(samples), max_len), device=device)
+                    torch.distributed.scatter(scores, all_scores)
+                else:
+                    scores = all_scores[0].clone().detach()
             scores_mask = scores != -np.inf
 
             str_samples, str_prompts, str_outputs = self.decode(prompt_tensors, samples,

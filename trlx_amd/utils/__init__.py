@@ -11,7 +11,7 @@ import subprocess
 import time
 from enum import Enum
 from numbers import Number
-from typing import Any, Dict, Iterable, Tuple
+from typing import Any, Dict, Iterable, Optional, Tuple
 
 import numpy as np
 import torch
@@ -33,10 +33,13 @@ def significant(x: Number, ndigits=2) -> Number:
     return round(x, ndigits - int(math.floor(math.log10(abs(x)))))
 
 
-def set_seed(seed: int):
+def set_seed(seed: int, rank_offset: Optional[int] = None):
     """Seed python/numpy/torch; offsets by rank so DP rollouts decorrelate
-    (reference trlx/utils/__init__.py:44)."""
-    seed = int(seed) + int(os.environ.get("RANK", 0))
+    (reference trlx/utils/__init__.py:44).  Under tensor parallelism pass
+    ``rank_offset=dp_rank`` so TP peers share the stream."""
+    if rank_offset is None:
+        rank_offset = int(os.environ.get("RANK", 0))
+    seed = int(seed) + rank_offset
     random.seed(seed)
     np.random.seed(seed)
     torch.manual_seed(seed)
