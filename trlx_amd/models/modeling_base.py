@@ -58,6 +58,10 @@ class PreTrainedModelWrapper(nn.Module):
             config = preset(config)
         wrapped_kwargs, _ = cls._split_kwargs(kwargs)
         base = CausalTransformer(config)
+        if wrapped_kwargs.get("peft_config") is not None:
+            from .lora import apply_lora
+
+            apply_lora(base, wrapped_kwargs["peft_config"])
         return cls(base, **wrapped_kwargs)
 
     @classmethod
@@ -85,6 +89,15 @@ class PreTrainedModelWrapper(nn.Module):
             missing = [m for m in missing if not m.startswith("rope_")]
             if missing:
                 logger.warning(f"missing keys loading {pretrained_model_name_or_path}: {missing}")
+            adapter_path = os.path.join(pretrained_model_name_or_path, "adapter_model.pt")
+            if os.path.exists(adapter_path) and wrapped_kwargs.get("peft_config") is not None:
+                from .lora import apply_lora, load_lora_state_dict
+
+                apply_lora(base, wrapped_kwargs["peft_config"])
+                load_lora_state_dict(base, torch.load(adapter_path, map_location="cpu",
+                                                      weights_only=True))
+                wrapped_kwargs = dict(wrapped_kwargs)
+                wrapped_kwargs["_lora_applied"] = True
         elif pretrained_model_name_or_path in PRESETS:
             logger.warning(
                 f"'{pretrained_model_name_or_path}' is not a local directory; building a "
@@ -96,6 +109,12 @@ class PreTrainedModelWrapper(nn.Module):
                 f"'{pretrained_model_name_or_path}' is neither a local HF directory nor a known "
                 f"preset ({sorted(PRESETS)}); there is no network access to fetch it."
             )
+        if wrapped_kwargs.pop("_lora_applied", False):
+            pass
+        elif wrapped_kwargs.get("peft_config") is not None:
+            from .lora import apply_lora
+
+            apply_lora(base, wrapped_kwargs["peft_config"])
         model = cls(base, **wrapped_kwargs)
         if heads_sd:
             model.post_init(heads_sd)
@@ -122,10 +141,25 @@ class PreTrainedModelWrapper(nn.Module):
         return {k: v for k, v in full.items() if not k.startswith("base_model.")}
 
     def save_pretrained(self, save_directory: str, **kwargs):
-        """Write an HF directory for the base model + wrapper head weights."""
+        """Write an HF directory for the base model + wrapper head weights.
+        Under LoRA, the adapter is saved separately (adapter_model.pt, the
+        reference's peft behavior — modeling_base.py:328-355) and the exported
+        base weights exclude the adapter."""
+        from .lora import has_lora, lora_state_dict
+
         os.makedirs(save_directory, exist_ok=True)
-        base_sd = {k: v.cpu() for k, v in self.base_model.state_dict().items()}
+        base_sd = {}
+        for k, v in self.base_model.state_dict().items():
+            if "lora_A" in k or "lora_B" in k:
+                continue
+            base_sd[k] = v.cpu()
+        # LoRALinear nests the original module as `.base`
+        base_sd = {k.replace(".base.weight", ".weight").replace(".base.bias", ".bias"): v
+                   for k, v in base_sd.items()}
         save_hf_dir(save_directory, self.config, base_sd)
+        if has_lora(self.base_model):
+            torch.save(lora_state_dict(self.base_model),
+                       os.path.join(save_directory, "adapter_model.pt"))
         heads = {k: v.cpu() for k, v in self.heads_state_dict().items()}
         if heads:
             torch.save(heads, os.path.join(save_directory, WRAPPER_HEADS_NAME))

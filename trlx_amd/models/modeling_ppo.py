@@ -291,6 +291,16 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
                 ctx.position_ids = position_ids.to(torch.int32)
             ref_logits = self.frozen_head(out.hidden_at_layer, ctx, self.base_model.rope_tables,
                                           logits_slice=logits_slice)
+        elif return_ref_logits and self.peft_config is not None:
+            # peft hydra: the base model WITHOUT adapters is the reference
+            # (reference accelerate_ppo_trainer.py:74-77 + peft disable_adapter)
+            from .lora import lora_disabled
+
+            with torch.no_grad(), lora_disabled(self.base_model):
+                ref_logits = self.base_model(
+                    input_ids, attention_mask=attention_mask, position_ids=position_ids,
+                    logits_slice=logits_slice,
+                ).logits
         return CausalLMOutputWithValue(
             logits=out.logits, values=values, ref_logits=ref_logits,
             last_hidden_state=out.last_hidden_state,

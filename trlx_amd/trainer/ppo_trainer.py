@@ -57,8 +57,10 @@ class PPOTrainer(NativeRLTrainer):
         self.store = PPORolloutStorage(self.tokenizer.pad_token_id, self.tokenizer.padding_side)
         self.store.clear_history()
 
-        # separate full reference model only without a hydra branch
-        if hasattr(self.model, "frozen_head") and self.model.frozen_head is not None:
+        # separate full reference model only without a hydra branch or peft
+        # (reference accelerate_ppo_trainer.py:74-77)
+        if (hasattr(self.model, "frozen_head") and self.model.frozen_head is not None) or \
+                self.config.model.peft_config is not None:
             self.ref_model = None
         else:
             self.ref_model = self.get_arch(self.config)
