@@ -32,7 +32,12 @@ from ..utils import (
     significant,
 )
 from ..utils import logging
-from ..utils.modeling import flatten_dict, gather_dict, freeze_bottom_causal_layers
+from ..utils.modeling import (
+    flatten_dict,
+    freeze_bottom_causal_layers,
+    freeze_bottom_seq2seq_layers,
+    gather_dict,
+)
 from ..utils.tokenizer import get_tokenizer
 from ..utils.trackers import make_tracker
 
@@ -140,7 +145,12 @@ class NativeRLTrainer(BaseRLTrainer):
     def setup_model(self):
         model = self.get_arch(self.config)
         base = model.base_model if hasattr(model, "base_model") else model
-        freeze_bottom_causal_layers(base, self.config.model.num_layers_unfrozen)
+        if self.config.model.peft_config is not None:
+            pass  # LoRA already froze the base
+        elif self.config.model.model_arch_type == "seq2seq":
+            freeze_bottom_seq2seq_layers(base, self.config.model.num_layers_unfrozen)
+        else:
+            freeze_bottom_causal_layers(base, self.config.model.num_layers_unfrozen)
         model = model.to(self.device)
         if self.dtype != torch.float32:
             # heads stay fp32; trunk + frozen branch go bf16

@@ -102,6 +102,14 @@ class ILQLTrainer(NativeRLTrainer):
         path = config.model.model_path
         kwargs = dict(two_qs=config.method.two_qs, alpha=config.method.alpha,
                       peft_config=config.model.peft_config)
+        if config.model.model_arch_type == "seq2seq":
+            from ..models.modeling_seq2seq import AutoModelForSeq2SeqLMWithILQLHeads
+            from ..models.nn.seq2seq import Seq2SeqConfig
+
+            if isinstance(path, str) and config.model.model_extra_configs.get("config"):
+                return AutoModelForSeq2SeqLMWithILQLHeads.from_config(
+                    Seq2SeqConfig.from_dict(config.model.model_extra_configs["config"]), **kwargs)
+            return AutoModelForSeq2SeqLMWithILQLHeads.from_pretrained(path, **kwargs)
         if isinstance(path, str) and config.model.model_extra_configs.get("config"):
             from ..models.nn.config import TransformerConfig
 
@@ -116,12 +124,21 @@ class ILQLTrainer(NativeRLTrainer):
 
     def loss(self, batch: ILQLBatch):
         batch = to_device(batch, self.device)
-        out = self.model(
-            input_ids=batch.input_ids,
-            attention_mask=batch.attention_mask,
-            actions_ixs=batch.actions_ixs,
-            states_ixs=batch.states_ixs,
-        )
+        if self.config.model.model_arch_type == "seq2seq":
+            out = self.model(
+                input_ids=batch.input_ids,
+                attention_mask=batch.attention_mask,
+                decoder_input_ids=batch.decoder_input_ids,
+                actions_ixs=batch.actions_ixs,
+                states_ixs=batch.states_ixs,
+            )
+        else:
+            out = self.model(
+                input_ids=batch.input_ids,
+                attention_mask=batch.attention_mask,
+                actions_ixs=batch.actions_ixs,
+                states_ixs=batch.states_ixs,
+            )
         return self.ilql.loss((out.logits, (out.qs, out.target_qs, out.vs)), batch)
 
     def create_train_dataloader(self):
@@ -136,4 +153,46 @@ class ILQLTrainer(NativeRLTrainer):
 
     def make_experience(self, samples, rewards, max_length=2048):
         """Build the ILQL store (called by trlx.train for offline data)."""
+        if self.config.model.model_arch_type == "seq2seq":
+            return self.make_experience_seq2seq(samples, rewards, max_length)
         self.store = make_experience(samples, rewards, self.tokenizer, max_length=max_length)
+
+    def make_experience_seq2seq(self, samples, rewards, max_length=2048):
+        """(prompt, output) pairs -> encoder input + decoder output tensors
+        (reference accelerate_ilql_trainer.py:178-244)."""
+        from ..pipeline.offline_pipeline import ILQLSeq2SeqRolloutStorage
+
+        logger.info("Collecting rollouts")
+        if self.tokenizer:
+            samples = [tokenize_dialogue(s, self.tokenizer, max_length) for s in samples]
+
+        all_input_ids = []
+        all_output_ids = []
+        all_actions_ixs = []
+        all_states_ixs = []
+        all_dones = []
+        for sample in samples:
+            all_input_ids.append(torch.tensor(sample[0].tokens))
+            all_output_ids.append(torch.tensor(sample[1].tokens))
+            actions_ixs = []
+            length = 0
+            for phrase in sample:
+                if phrase.is_output:
+                    length = len(phrase.tokens)
+                    actions_ixs.append(torch.arange(0, length - 1))
+            states_ixs = torch.hstack((*actions_ixs, torch.tensor(length - 1)))
+            all_dones.append(torch.tensor([1] * (len(states_ixs) - 1) + [0], dtype=int))
+            all_actions_ixs.append(torch.hstack(actions_ixs))
+            all_states_ixs.append(states_ixs)
+
+        returns = torch.tensor(rewards, dtype=float)
+        returns = (returns - returns.mean()) / (returns.std() + torch.finfo(returns.dtype).eps)
+        rewards = [torch.zeros(len(x)) for x in all_actions_ixs]
+        for rs, ret in zip(rewards, returns):
+            rs[-1] = ret
+
+        attention_mask = [torch.ones(len(x), dtype=int) for x in all_input_ids]
+        self.store = ILQLSeq2SeqRolloutStorage(
+            all_input_ids, attention_mask, all_output_ids, rewards,
+            all_states_ixs, all_actions_ixs, all_dones,
+        )

@@ -83,7 +83,23 @@ class PPOTrainer(NativeRLTrainer):
 
     def get_arch(self, config: TRLConfig):
         if config.model.model_arch_type == "seq2seq":
-            raise NotImplementedError("seq2seq PPO lands with the T5 family")
+            from ..models.modeling_seq2seq import (
+                AutoModelForSeq2SeqLMWithHydraValueHead,
+                AutoModelForSeq2SeqLMWithValueHead,
+            )
+            from ..models.nn.seq2seq import Seq2SeqConfig
+
+            model_cls = AutoModelForSeq2SeqLMWithHydraValueHead
+            kwargs = dict(num_layers_unfrozen=config.model.num_layers_unfrozen,
+                          peft_config=config.model.peft_config)
+            if config.model.num_layers_unfrozen == -1:
+                model_cls = AutoModelForSeq2SeqLMWithValueHead
+                kwargs.pop("num_layers_unfrozen")
+            path = config.model.model_path
+            if isinstance(path, str) and not os.path.isdir(path) and config.model.model_extra_configs.get("config"):
+                return model_cls.from_config(
+                    Seq2SeqConfig.from_dict(config.model.model_extra_configs["config"]), **kwargs)
+            return model_cls.from_pretrained(path, **kwargs)
         model_cls = AutoModelForCausalLMWithHydraValueHead
         kwargs = dict(
             num_layers_unfrozen=config.model.num_layers_unfrozen,
@@ -116,6 +132,32 @@ class PPOTrainer(NativeRLTrainer):
         advantages, returns = self.config.method.get_advantages_and_returns(
             old_values, old_rewards, response_length
         )
+
+        if self.config.model.model_arch_type == "seq2seq":
+            # encoder gets the query, decoder the response
+            # (reference accelerate_ppo_trainer.py:146-174)
+            input_ids = query_tensors
+            decoder_input_ids = response_tensors
+            attention_mask = input_ids.ne(self.tokenizer.pad_token_id).long()
+            decoder_attention_mask = decoder_input_ids.ne(self.tokenizer.pad_token_id).long()
+            decoder_attention_mask[:, 0] = 1
+            outputs = self.model(
+                input_ids=input_ids, attention_mask=attention_mask,
+                decoder_input_ids=decoder_input_ids,
+                decoder_attention_mask=decoder_attention_mask,
+            )
+            logprobs = logprobs_of_labels(outputs.logits[:, :-1, :], decoder_input_ids[:, 1:])
+            mask = decoder_input_ids.ne(self.tokenizer.pad_token_id).long()
+            start, end = 0, response_length
+            logprobs, values_pred, mask = (
+                logprobs[:, start:end],
+                outputs.values[:, start:end],
+                mask[:, start + 1 : end + 1],
+            )
+            return self.config.method.loss(
+                logprobs=logprobs, values=values_pred, old_logprobs=old_logprobs,
+                old_values=old_values, advantages=advantages, returns=returns, mask=mask,
+            )
 
         tokens = torch.cat((query_tensors, response_tensors), dim=1)
         attention_mask = tokens.not_equal(self.tokenizer.pad_token_id).long().to(tokens.device)
@@ -264,6 +306,11 @@ class PPOTrainer(NativeRLTrainer):
 
             # re-tokenize outputs (stop sequences may have trimmed them)
             outputs = self.tokenizer(str_outputs).input_ids
+            if self.config.model.model_arch_type == "seq2seq":
+                # decoder sequences start with the decoder-start/pad token
+                # (reference accelerate_ppo_trainer.py:352-356)
+                dst = getattr(self.model.config, "decoder_start_token_id", self.tokenizer.pad_token_id)
+                outputs = [[dst] + o for o in outputs]
             outputs = list(map(torch.LongTensor, outputs))
             maxsize = max(max(map(len, outputs)), 1)
             outputs = [
@@ -291,49 +338,93 @@ class PPOTrainer(NativeRLTrainer):
             elif self.config.method.scale_reward == "ref":
                 scores /= self.ref_std
 
-            # one trunk pass for policy logits + values (+ ref logits via
-            # hydra); the vocab-wide lm_head/logprob math runs only on the
-            # response region [start, T-1) — prompt-position KL (a logged
-            # stat in the reference, not a training signal) is therefore
-            # measured over the response region
-            all_tokens = torch.cat((prompt_tensors, sample_outputs), dim=1)
-            attention_mask = all_tokens.not_equal(self.tokenizer.pad_token_id).long().to(device)
-            n_samples = samples.shape[0]
-            start = prompt_tensors.shape[1] - 1
-            T_all = all_tokens.shape[1]
-            with torch.no_grad():
-                outputs = self.model(all_tokens, attention_mask=attention_mask,
-                                     return_ref_logits=True, logits_slice=(start, T_all - 1))
-                logits, values = outputs.logits, outputs.values
-                if outputs.ref_logits is not None:
-                    ref_logits = outputs.ref_logits
-                elif self.ref_model is not None:
-                    ref_logits = self.ref_model(all_tokens, attention_mask=attention_mask,
-                                                logits_slice=(start, T_all - 1)).logits
-                else:
-                    # num_layers_unfrozen == -1 and no separate ref: KL vs itself
-                    ref_logits = logits
+            if self.config.model.model_arch_type == "seq2seq":
+                # seq2seq experience pass (reference accelerate_ppo_trainer.py:383-414)
+                n_samples = samples.shape[0]
+                enc_mask = batch["attention_mask"].to(device)
+                dec_mask = sample_outputs.ne(self.tokenizer.pad_token_id).long()
+                dec_mask[:, 0] = 1
+                with torch.no_grad():
+                    outputs = self.model(
+                        input_ids=prompt_tensors, attention_mask=enc_mask,
+                        decoder_input_ids=sample_outputs, decoder_attention_mask=dec_mask,
+                        return_ref_logits=True,
+                    )
+                    logits, values = outputs.logits, outputs.values
+                    if outputs.ref_logits is not None:
+                        ref_logits = outputs.ref_logits
+                    elif self.ref_model is not None:
+                        ref_logits = self.ref_model(
+                            input_ids=prompt_tensors, attention_mask=enc_mask,
+                            decoder_input_ids=sample_outputs, decoder_attention_mask=dec_mask,
+                        ).logits
+                    else:
+                        ref_logits = logits
+                    logprobs = logprobs_of_labels(logits[:, :-1, :], sample_outputs[:, 1:])
+                    ref_logprobs = logprobs_of_labels(ref_logits[:, :-1, :], sample_outputs[:, 1:])
 
-                # local index j <-> global position start + j
-                logprobs = logprobs_of_labels(logits, all_tokens[:, start + 1 :])
-                ref_logprobs = logprobs_of_labels(ref_logits, all_tokens[:, start + 1 :])
+                attn_for_kl = sample_outputs.ne(self.tokenizer.pad_token_id).long().to(device)
+                start = 0
+                log_ratio = (logprobs - ref_logprobs) * attn_for_kl[:, :-1]
+                kl = log_ratio.exp() - 1 - log_ratio
+                mean_kl_per_token = kl.mean()
+                mean_kl = kl.sum(1).mean()
 
-            log_ratio = (logprobs - ref_logprobs) * attention_mask[:, start:-1]
-            kl = log_ratio.exp() - 1 - log_ratio
-            mean_kl_per_token = kl.mean()
-            mean_kl = kl.sum(1).mean()
+                logprobs = logprobs.cpu()
+                prompt_tensors = prompt_tensors.cpu()
+                sample_outputs = sample_outputs.cpu()
+                values = values.cpu()[:, :-1]
 
-            logprobs = logprobs.cpu()
-            prompt_tensors = prompt_tensors.cpu()
-            sample_outputs = sample_outputs.cpu()
-            values = values.cpu()
+                ends = start + attn_for_kl[:, start:].sum(1).cpu() + 1
+                all_values = [values[ix, start : ends[ix]] for ix in range(n_samples)]
+                all_logprobs = [logprobs[ix, start : ends[ix]] for ix in range(n_samples)]
+                kl_penalty = self.kl_ctl.value * -log_ratio.cpu()
+                kl_penalty = [xs[start : ends[ix]] for ix, xs in enumerate(kl_penalty)]
+            else:
+                # causal experience pass — one trunk pass for policy logits +
+                # values (+ ref logits via hydra); the vocab-wide
+                # lm_head/logprob math runs only on the response region
+                # [start, T-1) — prompt-position KL (a logged stat in the
+                # reference, not a training signal) is therefore measured over
+                # the response region
+                all_tokens = torch.cat((prompt_tensors, sample_outputs), dim=1)
+                attention_mask = all_tokens.not_equal(self.tokenizer.pad_token_id).long().to(device)
+                n_samples = samples.shape[0]
+                start = prompt_tensors.shape[1] - 1
+                T_all = all_tokens.shape[1]
+                with torch.no_grad():
+                    outputs = self.model(all_tokens, attention_mask=attention_mask,
+                                         return_ref_logits=True, logits_slice=(start, T_all - 1))
+                    logits, values = outputs.logits, outputs.values
+                    if outputs.ref_logits is not None:
+                        ref_logits = outputs.ref_logits
+                    elif self.ref_model is not None:
+                        ref_logits = self.ref_model(all_tokens, attention_mask=attention_mask,
+                                                    logits_slice=(start, T_all - 1)).logits
+                    else:
+                        # num_layers_unfrozen == -1 and no separate ref: KL vs itself
+                        ref_logits = logits
 
-            ends = attention_mask[:, start:].sum(1).cpu() + 1
-            all_values = [values[ix, : ends[ix]] for ix in range(n_samples)]
-            all_logprobs = [logprobs[ix, : ends[ix]] for ix in range(n_samples)]
+                    # local index j <-> global position start + j
+                    logprobs = logprobs_of_labels(logits, all_tokens[:, start + 1 :])
+                    ref_logprobs = logprobs_of_labels(ref_logits, all_tokens[:, start + 1 :])
 
-            kl_penalty = self.kl_ctl.value * -log_ratio.cpu()
-            kl_penalty = [xs[: ends[ix]] for ix, xs in enumerate(kl_penalty)]
+                log_ratio = (logprobs - ref_logprobs) * attention_mask[:, start:-1]
+                kl = log_ratio.exp() - 1 - log_ratio
+                mean_kl_per_token = kl.mean()
+                mean_kl = kl.sum(1).mean()
+
+                logprobs = logprobs.cpu()
+                prompt_tensors = prompt_tensors.cpu()
+                sample_outputs = sample_outputs.cpu()
+                values = values.cpu()
+
+                ends = attention_mask[:, start:].sum(1).cpu() + 1
+                all_values = [values[ix, : ends[ix]] for ix in range(n_samples)]
+                all_logprobs = [logprobs[ix, : ends[ix]] for ix in range(n_samples)]
+
+                kl_penalty = self.kl_ctl.value * -log_ratio.cpu()
+                kl_penalty = [xs[: ends[ix]] for ix, xs in enumerate(kl_penalty)]
 
             rollout_count = 0
             for sample_idx in range(n_samples):

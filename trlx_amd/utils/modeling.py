@@ -151,6 +151,24 @@ def freeze_bottom_causal_layers(model, num_layers_unfrozen: int = 0):
             p.requires_grad_(False)
 
 
+def freeze_bottom_seq2seq_layers(model, num_layers_unfrozen: int = 0):
+    """Seq2seq freezing (reference trlx/utils/modeling.py
+    freeze_bottom_seq2seq_layers): freeze shared embeddings, the whole
+    encoder, and all but the top ``num_layers_unfrozen`` decoder blocks."""
+    if num_layers_unfrozen == -1:
+        return
+    model.shared.weight.requires_grad_(False)
+    for p in model.encoder_blocks.parameters():
+        p.requires_grad_(False)
+    for p in model.encoder_final_norm.parameters():
+        p.requires_grad_(False)
+    blocks = list(model.decoder_blocks)
+    frozen = blocks if num_layers_unfrozen == 0 else blocks[:-num_layers_unfrozen]
+    for block in frozen:
+        for p in block.parameters():
+            p.requires_grad_(False)
+
+
 def union_dicts(*dicts) -> Dict:
     out = {}
     for d in dicts:
