@@ -150,3 +150,18 @@ def test_seq2seq_ilql_end_to_end(tmp_path):
         config=cfg,
     )
     assert trainer.iter_count == 2
+
+
+@pytest.mark.gpu
+def test_seq2seq_gpu_forward_and_generate():
+    torch.manual_seed(0)
+    m = Seq2SeqTransformer(tiny_s2s()).cuda().to(torch.bfloat16).eval()
+    ids = torch.randint(3, 300, (2, 9)).cuda()
+    mask = torch.ones_like(ids)
+    dec = torch.randint(3, 300, (2, 5)).cuda()
+    dec[:, 0] = 2
+    with torch.no_grad():
+        out = m(ids, mask, decoder_input_ids=dec)
+    assert out.logits.isfinite().all()
+    gen = m.generate(ids, mask, max_new_tokens=6, do_sample=True)
+    assert gen.shape[0] == 2 and gen.shape[1] <= 7
