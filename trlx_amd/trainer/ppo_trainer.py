@@ -31,7 +31,7 @@ from ..models.modeling_ppo import (
     AutoModelForCausalLMWithValueHead,
     FixedKLController,
 )
-from ..parallel import comm
+from ..parallel import comm, topo
 from ..pipeline.offline_pipeline import PromptPipeline
 from ..pipeline.ppo_pipeline import PPORolloutStorage
 from ..trainer import register_trainer
@@ -247,8 +247,6 @@ class PPOTrainer(NativeRLTrainer):
             metadata_local = {k: v for k, v in batch.items()
                               if k not in ("input_ids", "attention_mask")}
 
-            from ..parallel import topo
-
             if topo.tp_size() > 1:
                 # tensor-parallel mode: every rank scores its own samples
                 # locally (TP peers hold identical rollouts) — the NeMo-path
@@ -301,8 +299,13 @@ class PPOTrainer(NativeRLTrainer):
                     scores = all_scores[0].clone().detach()
             scores_mask = scores != -np.inf
 
-            str_samples, str_prompts, str_outputs = self.decode(prompt_tensors, samples,
-                                                                append_eos_token=True)
+            if topo.tp_size() <= 1 and self.world_size == 1:
+                # single process: the gathered batch IS the local batch
+                str_samples, str_prompts, str_outputs = (
+                    all_str_samples, all_str_prompts, all_str_outputs)
+            else:
+                str_samples, str_prompts, str_outputs = self.decode(prompt_tensors, samples,
+                                                                    append_eos_token=True)
 
             # re-tokenize outputs (stop sequences may have trimmed them)
             outputs = self.tokenizer(str_outputs).input_ids
@@ -370,15 +373,11 @@ class PPOTrainer(NativeRLTrainer):
                 mean_kl_per_token = kl.mean()
                 mean_kl = kl.sum(1).mean()
 
-                logprobs = logprobs.cpu()
-                prompt_tensors = prompt_tensors.cpu()
-                sample_outputs = sample_outputs.cpu()
-                values = values.cpu()[:, :-1]
-
-                ends = start + attn_for_kl[:, start:].sum(1).cpu() + 1
+                values = values[:, :-1]
+                ends = (start + attn_for_kl[:, start:].sum(1) + 1).tolist()
                 all_values = [values[ix, start : ends[ix]] for ix in range(n_samples)]
                 all_logprobs = [logprobs[ix, start : ends[ix]] for ix in range(n_samples)]
-                kl_penalty = self.kl_ctl.value * -log_ratio.cpu()
+                kl_penalty = self.kl_ctl.value * -log_ratio
                 kl_penalty = [xs[start : ends[ix]] for ix, xs in enumerate(kl_penalty)]
             else:
                 # causal experience pass — one trunk pass for policy logits +
@@ -414,16 +413,13 @@ class PPOTrainer(NativeRLTrainer):
                 mean_kl_per_token = kl.mean()
                 mean_kl = kl.sum(1).mean()
 
-                logprobs = logprobs.cpu()
-                prompt_tensors = prompt_tensors.cpu()
-                sample_outputs = sample_outputs.cpu()
-                values = values.cpu()
-
-                ends = attention_mask[:, start:].sum(1).cpu() + 1
+                # rollout tensors stay DEVICE-resident (no CPU round trip —
+                # the store and loss both live on the GPU)
+                ends = (attention_mask[:, start:].sum(1) + 1).tolist()
                 all_values = [values[ix, : ends[ix]] for ix in range(n_samples)]
                 all_logprobs = [logprobs[ix, : ends[ix]] for ix in range(n_samples)]
 
-                kl_penalty = self.kl_ctl.value * -log_ratio.cpu()
+                kl_penalty = self.kl_ctl.value * -log_ratio
                 kl_penalty = [xs[: ends[ix]] for ix, xs in enumerate(kl_penalty)]
 
             rollout_count = 0
@@ -431,12 +427,12 @@ class PPOTrainer(NativeRLTrainer):
                 rewards = kl_penalty[sample_idx].clone()
                 if scores.shape[1] == 1:
                     # terminal reward at the last (eos) token
-                    rewards[-1] += scores[sample_idx][0].cpu()
+                    rewards[-1] += scores[sample_idx][0]
                 else:
                     # dense per-token rewards
                     score = scores[sample_idx]
                     score_right_padding = torch.sum(scores_mask[sample_idx])
-                    score = score[:score_right_padding].cpu()
+                    score = score[:score_right_padding]
                     p_score = torch.zeros_like(rewards)
                     p_score[: score.shape[0]] += score
                     rewards += p_score
