@@ -116,8 +116,10 @@ class Attention(nn.Module):
         self.attn_pdrop = cfg.attn_pdrop
 
     def forward(self, x, ctx: AttentionContext, rope_tables, kv_cache: Optional[KVCache] = None):
-        B, T, _ = x.shape
+        B = x.shape[0]
         qkv = self.qkv_proj(x)
+        # under sequence parallelism the projection gathers the full sequence
+        T = qkv.shape[1]
 
         # fused single-token decode path: decode_prep (split+RoPE+cache
         # append) + flash-decode attention, fully device-side (hipGraph-safe)
@@ -284,7 +286,22 @@ class CausalTransformer(nn.Module):
             self.register_buffer("rope_sin", sin, persistent=False)
         else:
             self.rope_cos = self.rope_sin = None
+        self.sequence_parallel = False
         self.apply(self._init_weights)
+
+    def set_sequence_parallel(self, enabled: bool):
+        """Toggle Megatron-style sequence parallelism (activations sharded
+        along T across the TP group in the norm/residual segments; gathered at
+        column-parallel inputs, reduce-scattered at row-parallel outputs).
+        The reference flips this around generation (modeling_nemo_ppo.py:
+        820-836); here SP engages automatically only for cache-less forwards
+        with T divisible by the TP size."""
+        from ...parallel.tp import ColumnParallelLinear, RowParallelLinear
+
+        self.sequence_parallel = enabled
+        for m in self.modules():
+            if isinstance(m, (ColumnParallelLinear, RowParallelLinear)):
+                m.sequence_parallel = enabled
 
     def _init_weights(self, module):
         if isinstance(module, (nn.Linear, ColumnParallelLinear, RowParallelLinear)):
@@ -295,6 +312,21 @@ class CausalTransformer(nn.Module):
             module.weight.data.normal_(mean=0.0, std=0.02)
 
     # --- helpers -----------------------------------------------------------
+
+    def _sp_flags_on(self) -> bool:
+        from ...parallel.tp import ColumnParallelLinear
+
+        for m in self.modules():
+            if isinstance(m, ColumnParallelLinear):
+                return m.sequence_parallel
+        return False
+
+    def _set_sp_flags(self, enabled: bool):
+        from ...parallel.tp import ColumnParallelLinear, RowParallelLinear
+
+        for m in self.modules():
+            if isinstance(m, (ColumnParallelLinear, RowParallelLinear)):
+                m.sequence_parallel = enabled
 
     def embed_parameters(self):
         params = list(self.embed_tokens.parameters())
@@ -360,6 +392,17 @@ class CausalTransformer(nn.Module):
         if self.embd_pdrop > 0 and self.training:
             h = F.dropout(h, self.embd_pdrop)
 
+        # sequence parallelism: shard activations along T for the
+        # norm/residual segments (SP stays off for KV-cached generation)
+        sp_active = (self.sequence_parallel and topo.tp_size() > 1 and kv_cache is None
+                     and input_ids.shape[1] % topo.tp_size() == 0)
+        if sp_active != self._sp_flags_on():
+            self._set_sp_flags(sp_active)
+        if sp_active:
+            from ...parallel.tp import scatter_to_sp
+
+            h = scatter_to_sp(h)
+
         n = len(self.layers)
         stash_at = None
         if hidden_at_layer is not None:
@@ -370,6 +413,14 @@ class CausalTransformer(nn.Module):
                 hidden_at = h
             h = layer(h, ctx, self.rope_tables, kv_cache)
         h = self.final_norm(h)
+        if sp_active:
+            # the loss region downstream is replicated on every TP rank ->
+            # slice-backward gather (see _GatherFromSPReplicated)
+            from ...parallel.tp import gather_from_sp_replicated
+
+            h = gather_from_sp_replicated(h)
+            if hidden_at is not None:
+                hidden_at = gather_from_sp_replicated(hidden_at)
         logits = None
         if return_logits:
             # logits_slice: compute the [V]-wide projection only where the

@@ -61,8 +61,107 @@ def reduce_from_tp(x: torch.Tensor) -> torch.Tensor:
     return _ReduceFromTP.apply(x)
 
 
+# --- sequence-parallel region functions (Megatron SP; SURVEY.md §2.2 SP row) ----
+
+
+def _sp_all_gather(x: torch.Tensor) -> torch.Tensor:
+    """Concatenate sequence shards along dim 1 from the TP group."""
+    W = topo.tp_size()
+    outs = [torch.empty_like(x) for _ in range(W)]
+    dist.all_gather(outs, x.contiguous(), group=topo.tp_group())
+    return torch.cat(outs, dim=1)
+
+
+def _sp_reduce_scatter(x: torch.Tensor) -> torch.Tensor:
+    """Sum-reduce along the TP group and keep this rank's sequence shard."""
+    W = topo.tp_size()
+    x = x.contiguous()
+    dist.all_reduce(x, dist.ReduceOp.SUM, group=topo.tp_group())
+    T = x.shape[1] // W
+    r = topo.tp_rank()
+    return x[:, r * T : (r + 1) * T].contiguous()
+
+
+class _GatherFromSP(torch.autograd.Function):
+    """fwd: all-gather sequence shards; bwd: reduce-scatter gradients
+    (Megatron gather_from_sequence_parallel_region)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return _sp_all_gather(x)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _sp_reduce_scatter(grad)
+
+
+class _ReduceScatterToSP(torch.autograd.Function):
+    """fwd: reduce-scatter to sequence shards; bwd: all-gather gradients."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return _sp_reduce_scatter(x)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _sp_all_gather(grad)
+
+
+class _ScatterToSP(torch.autograd.Function):
+    """fwd: keep this rank's sequence shard; bwd: all-gather gradients."""
+
+    @staticmethod
+    def forward(ctx, x):
+        W = topo.tp_size()
+        T = x.shape[1] // W
+        r = topo.tp_rank()
+        return x[:, r * T : (r + 1) * T].contiguous()
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _sp_all_gather(grad)
+
+
+class _GatherFromSPReplicated(torch.autograd.Function):
+    """fwd: all-gather sequence shards; bwd: take THIS rank's shard of the
+    gradient.  Correct when the downstream computation (loss) is REPLICATED
+    identically on every TP rank — the framework's convention for the final
+    lm_head/loss region (reduce-scatter there would double-count the
+    replicated gradients)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        return _sp_all_gather(x)
+
+    @staticmethod
+    def backward(ctx, grad):
+        W = topo.tp_size()
+        T = grad.shape[1] // W
+        r = topo.tp_rank()
+        return grad[:, r * T : (r + 1) * T].contiguous()
+
+
+def gather_from_sp(x):
+    return _GatherFromSP.apply(x)
+
+
+def gather_from_sp_replicated(x):
+    return _GatherFromSPReplicated.apply(x)
+
+
+def reduce_scatter_to_sp(x):
+    return _ReduceScatterToSP.apply(x)
+
+
+def scatter_to_sp(x):
+    return _ScatterToSP.apply(x)
+
+
 class ColumnParallelLinear(nn.Module):
-    """y_local = x @ W_l^T + b_l with W sharded on output dim."""
+    """y_local = x @ W_l^T + b_l with W sharded on output dim.
+
+    With ``sequence_parallel`` the input arrives sequence-sharded and is
+    all-gathered here (grad reduce-scatters back) — the SP entry point."""
 
     def __init__(self, in_features: int, out_features: int, bias: bool = True):
         super().__init__()
@@ -73,14 +172,21 @@ class ColumnParallelLinear(nn.Module):
         self.out_local = out_features // tp
         self.weight = nn.Parameter(torch.empty(self.out_local, in_features))
         self.bias = nn.Parameter(torch.zeros(self.out_local)) if bias else None
+        self.sequence_parallel = False
 
     def forward(self, x):
-        x = copy_to_tp(x)
+        if self.sequence_parallel and topo.tp_size() > 1:
+            x = gather_from_sp(x)
+        else:
+            x = copy_to_tp(x)
         return torch.nn.functional.linear(x, self.weight, self.bias)
 
 
 class RowParallelLinear(nn.Module):
-    """y = all_reduce(x_l @ W_l^T) + b with W sharded on input dim."""
+    """y = all_reduce(x_l @ W_l^T) + b with W sharded on input dim.
+
+    With ``sequence_parallel`` the all-reduce becomes a reduce-scatter to
+    sequence shards — the SP exit point."""
 
     def __init__(self, in_features: int, out_features: int, bias: bool = True):
         super().__init__()
@@ -91,10 +197,14 @@ class RowParallelLinear(nn.Module):
         self.in_local = in_features // tp
         self.weight = nn.Parameter(torch.empty(out_features, self.in_local))
         self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+        self.sequence_parallel = False
 
     def forward(self, x):
         y = torch.nn.functional.linear(x, self.weight, None)
-        y = reduce_from_tp(y)
+        if self.sequence_parallel and topo.tp_size() > 1:
+            y = reduce_scatter_to_sp(y)
+        else:
+            y = reduce_from_tp(y)
         if self.bias is not None:
             y = y + self.bias
         return y

@@ -151,6 +151,9 @@ class NativeRLTrainer(BaseRLTrainer):
             freeze_bottom_seq2seq_layers(base, self.config.model.num_layers_unfrozen)
         else:
             freeze_bottom_causal_layers(base, self.config.model.num_layers_unfrozen)
+        if (self.config.train.sequence_parallel and self.tp_size > 1
+                and hasattr(base, "set_sequence_parallel")):
+            base.set_sequence_parallel(True)
         model = model.to(self.device)
         if self.dtype != torch.float32:
             # heads stay fp32; trunk + frozen branch go bf16
