@@ -43,6 +43,8 @@ at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs)
 at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
                             const at::Tensor& seq_lens, double scale,
                             const c10::optional<at::Tensor>& seq_starts);
+at::Tensor lm_logprobs(const at::Tensor& hidden, const at::Tensor& weight,
+                       const at::Tensor& labels);
 void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
                  at::Tensor& v, long step, double lr, double beta1, double beta2, double eps,
                  double weight_decay, double grad_scale);
@@ -67,4 +69,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("causal_softmax_bwd", &causal_softmax_bwd);
   mod.def("attention_decode", &attention_decode);
   mod.def("fused_adamw", &fused_adamw);
+  mod.def("lm_logprobs", &lm_logprobs);
 }
