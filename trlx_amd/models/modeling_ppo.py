@@ -218,6 +218,25 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
     def generate_eval(self, input_ids, attention_mask=None, **kwargs):
         return self.generate(input_ids, attention_mask, **kwargs)
 
+    @torch.no_grad()
+    def forward_experience(self, input_ids, attention_mask, lo: int, hi: int,
+                           labels: torch.Tensor):
+        """Fused experience path without a frozen branch (ref handled by the
+        caller); see the hydra variant for semantics."""
+        out = self.base_model(input_ids, attention_mask=attention_mask, return_logits=False)
+        h = out.last_hidden_state[:, lo:hi]
+        B, T = h.shape[:2]
+        values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        lm = self.base_model.lm_head
+        flat_labels = labels.reshape(-1)
+        if (h.is_cuda and h.dtype == torch.bfloat16 and lm.bias is None
+                and h.shape[-1] % 32 == 0):
+            logprobs = ops.lm_logprobs(h.reshape(-1, h.shape[-1]).contiguous(), lm.weight,
+                                       flat_labels).view(B, T)
+        else:
+            logprobs = ops.logprobs_of_labels(lm(h), labels)
+        return logprobs, None, values
+
 
 class FrozenBranch(nn.Module):
     """Frozen copies of the top ``num_layers_unfrozen`` blocks + final norm +
@@ -237,12 +256,19 @@ class FrozenBranch(nn.Module):
 
     def forward(self, hidden: torch.Tensor, ctx, rope_tables, logits_slice=None) -> torch.Tensor:
         with torch.no_grad():
+            h = self.forward_hidden(hidden, ctx, rope_tables, logits_slice)
+            return self.lm_head(h)
+
+    def forward_hidden(self, hidden: torch.Tensor, ctx, rope_tables, logits_slice=None) -> torch.Tensor:
+        """Final-norm hidden states of the frozen branch (pre-lm_head) — the
+        fused lm_logprobs path consumes these directly."""
+        with torch.no_grad():
             h = hidden
             for block in self.blocks:
                 h = block(h, ctx, rope_tables)
             if logits_slice is not None:
                 h = h[:, logits_slice[0] : logits_slice[1]]
-            return self.lm_head(self.final_norm(h))
+            return self.final_norm(h)
 
 
 class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
@@ -305,6 +331,61 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
             logits=out.logits, values=values, ref_logits=ref_logits,
             last_hidden_state=out.last_hidden_state,
         )
+
+    @torch.no_grad()
+    def forward_experience(self, input_ids, attention_mask, lo: int, hi: int,
+                           labels: torch.Tensor):
+        """Experience-phase fused path: per-token logprobs (policy + frozen
+        reference) and values for positions [lo, hi) in ONE trunk pass, using
+        the hand-MFMA lm_logprobs kernel — the [B, T, V] logits never exist.
+        ``labels`` = input_ids[:, lo+1 : hi+1].  Returns
+        (logprobs, ref_logprobs, values) each [B, hi-lo]; ref is None when
+        there is no frozen branch (caller falls back to a separate ref model
+        or adapter toggling)."""
+        stash = -self.num_layers_unfrozen if self.frozen_head is not None else None
+        out = self.base_model(input_ids, attention_mask=attention_mask,
+                              hidden_at_layer=stash, return_logits=False)
+        h = out.last_hidden_state[:, lo:hi]
+        B, T = h.shape[:2]
+        values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        lm = self.base_model.lm_head
+
+        def fused_ok(head):
+            return (h.is_cuda and h.dtype == torch.bfloat16 and head.bias is None
+                    and h.shape[-1] % 32 == 0)
+
+        flat_labels = labels.reshape(-1)
+        if fused_ok(lm):
+            logprobs = ops.lm_logprobs(h.reshape(-1, h.shape[-1]).contiguous(), lm.weight,
+                                       flat_labels).view(B, T)
+        else:
+            logprobs = ops.logprobs_of_labels(lm(h), labels)
+
+        ref_logprobs = None
+        if self.frozen_head is not None:
+            ctx = self.base_model.make_context(input_ids, attention_mask, 0)
+            rh = self.frozen_head.forward_hidden(out.hidden_at_layer, ctx,
+                                                 self.base_model.rope_tables,
+                                                 logits_slice=(lo, hi))
+            rlm = self.frozen_head.lm_head
+            if fused_ok(rlm):
+                ref_logprobs = ops.lm_logprobs(rh.reshape(-1, rh.shape[-1]).contiguous(),
+                                               rlm.weight, flat_labels).view(B, T)
+            else:
+                ref_logprobs = ops.logprobs_of_labels(rlm(rh), labels)
+        elif self.peft_config is not None:
+            from .lora import lora_disabled
+
+            with lora_disabled(self.base_model):
+                rout = self.base_model(input_ids, attention_mask=attention_mask,
+                                       return_logits=False)
+            rh = rout.last_hidden_state[:, lo:hi]
+            if fused_ok(lm):
+                ref_logprobs = ops.lm_logprobs(rh.reshape(-1, rh.shape[-1]).contiguous(),
+                                               lm.weight, flat_labels).view(B, T)
+            else:
+                ref_logprobs = ops.logprobs_of_labels(lm(rh), labels)
+        return logprobs, ref_logprobs, values
 
     def forward_hydra(
         self,

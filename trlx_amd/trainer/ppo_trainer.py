@@ -391,22 +391,34 @@ class PPOTrainer(NativeRLTrainer):
                 n_samples = samples.shape[0]
                 start = prompt_tensors.shape[1] - 1
                 T_all = all_tokens.shape[1]
+                labels = all_tokens[:, start + 1 :]
                 with torch.no_grad():
-                    outputs = self.model(all_tokens, attention_mask=attention_mask,
-                                         return_ref_logits=True, logits_slice=(start, T_all - 1))
-                    logits, values = outputs.logits, outputs.values
-                    if outputs.ref_logits is not None:
-                        ref_logits = outputs.ref_logits
-                    elif self.ref_model is not None:
-                        ref_logits = self.ref_model(all_tokens, attention_mask=attention_mask,
-                                                    logits_slice=(start, T_all - 1)).logits
+                    if hasattr(self.model, "forward_experience"):
+                        # fused hand-MFMA path: hidden -> logprobs directly,
+                        # the [B, T, V] logits never materialize
+                        logprobs, ref_logprobs, values = self.model.forward_experience(
+                            all_tokens, attention_mask, start, T_all - 1, labels)
+                        if ref_logprobs is None:
+                            if self.ref_model is not None:
+                                ref_logprobs, _, _ = self.ref_model.forward_experience(
+                                    all_tokens, attention_mask, start, T_all - 1, labels)
+                            else:
+                                # num_layers_unfrozen == -1, no ref: KL vs itself
+                                ref_logprobs = logprobs
                     else:
-                        # num_layers_unfrozen == -1 and no separate ref: KL vs itself
-                        ref_logits = logits
-
-                    # local index j <-> global position start + j
-                    logprobs = logprobs_of_labels(logits, all_tokens[:, start + 1 :])
-                    ref_logprobs = logprobs_of_labels(ref_logits, all_tokens[:, start + 1 :])
+                        outputs = self.model(all_tokens, attention_mask=attention_mask,
+                                             return_ref_logits=True,
+                                             logits_slice=(start, T_all - 1))
+                        logits, values = outputs.logits, outputs.values
+                        if outputs.ref_logits is not None:
+                            ref_logits = outputs.ref_logits
+                        elif self.ref_model is not None:
+                            ref_logits = self.ref_model(all_tokens, attention_mask=attention_mask,
+                                                        logits_slice=(start, T_all - 1)).logits
+                        else:
+                            ref_logits = logits
+                        logprobs = logprobs_of_labels(logits, labels)
+                        ref_logprobs = logprobs_of_labels(ref_logits, labels)
 
                 log_ratio = (logprobs - ref_logprobs) * attention_mask[:, start:-1]
                 kl = log_ratio.exp() - 1 - log_ratio
