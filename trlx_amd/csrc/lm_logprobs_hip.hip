@@ -79,8 +79,7 @@ __global__ __launch_bounds__(BLOCK) void lm_logprobs_tile_kernel(
           (const __attribute__((address_space(1))) unsigned int*)(weight + (size_t)gb_row * H + k0 + tk),
           (__attribute__((address_space(3))) unsigned int*)(b_lds + lds_off), 16, 0, 0);
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __syncthreads();
+    __syncthreads();  // barrier drains vmcnt (guide §5 m97 analysis)
 
     // fragment layout (mfma_f32_16x16x32_bf16): lane l holds rows l%16,
     // k-slice (l/16)*8 for A; columns l%16 (= W rows) for B.

@@ -229,7 +229,10 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         lm = self.base_model.lm_head
         flat_labels = labels.reshape(-1)
-        if (h.is_cuda and h.dtype == torch.bfloat16 and lm.bias is None
+        import os
+
+        if (os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") == "1"
+                and h.is_cuda and h.dtype == torch.bfloat16 and lm.bias is None
                 and h.shape[-1] % 32 == 0):
             logprobs = ops.lm_logprobs(h.reshape(-1, h.shape[-1]).contiguous(), lm.weight,
                                        flat_labels).view(B, T)
@@ -351,7 +354,14 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         lm = self.base_model.lm_head
 
         def fused_ok(head):
-            return (h.is_cuda and h.dtype == torch.bfloat16 and head.bias is None
+            # measured (profiles/r01_notes.md): the current single-buffered
+            # MFMA structure loses to hipBLASLt+logprobs (1603 vs 706 us at
+            # [5248, 50257, 768]); opt-in until the kernel moves to the
+            # 8-phase pipelined schedule
+            import os
+
+            return (os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") == "1"
+                    and h.is_cuda and h.dtype == torch.bfloat16 and head.bias is None
                     and h.shape[-1] % 32 == 0)
 
         flat_labels = labels.reshape(-1)
