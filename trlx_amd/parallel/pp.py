@@ -1,0 +1,205 @@
+"""Pipeline parallelism: stage-split transformer + fill-drain schedule.
+
+Parity target: the reference's NeMo/Apex pipeline path (SURVEY.md §2.2 PP
+row: per-stage model construction modeling_nemo_ppo.py:497-536, p2p
+activations of shape [seq, micro_batch, hidden], fwd/bwd schedule 713-731).
+
+MI355X design: activations/grads move between stages as [B, T, H] bf16
+tensors over RCCL p2p (xGMI on-node); the schedule is GPipe fill-drain —
+all microbatch forwards (stashing each stage's input/output pair), then all
+backwards stitched through autograd (grad of the stage output is received
+from downstream, the grad of the stage input is sent upstream).  1F1B
+interleaving and token-level pipelined generation are planned follow-ups.
+
+Rank layout (topo.init_model_parallel): pp is the middle axis —
+rank = (dp_idx * pp_size + pp_idx) * tp_size + tp_idx — so a PP group's
+stages sit tp_size apart (still intra-node for tp*pp <= 8).
+"""
+
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from . import topo
+from ..models.nn.config import TransformerConfig
+from ..models.nn.transformer import Block, Norm, TransformerOutput
+
+
+def split_layers(num_layers: int, pp: int, stage: int):
+    """Contiguous layer range for ``stage`` (early stages get the remainder)."""
+    base = num_layers // pp
+    extra = num_layers % pp
+    lo = stage * base + min(stage, extra)
+    hi = lo + base + (1 if stage < extra else 0)
+    return lo, hi
+
+
+class PipelineStage(nn.Module):
+    """One pipeline stage of the causal transformer.
+
+    Stage 0 owns the embeddings; the last stage owns the final norm +
+    lm_head.  Weights load from a full state dict via ``load_full_state_dict``
+    (the PP resharding of reference modeling_nemo_ppo.py:321-352).
+    """
+
+    def __init__(self, config: TransformerConfig, stage: int, num_stages: int):
+        super().__init__()
+        self.config = config
+        self.stage = stage
+        self.num_stages = num_stages
+        self.is_first = stage == 0
+        self.is_last = stage == num_stages - 1
+        self.lo, self.hi = split_layers(config.num_layers, num_stages, stage)
+        cfg = config
+
+        if self.is_first:
+            self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+            if cfg.position_encoding == "learned":
+                off = cfg.extra.get("position_offset", 0)
+                self.embed_positions = nn.Embedding(cfg.max_position_embeddings + off, cfg.hidden_size)
+            else:
+                self.embed_positions = None
+        self.layers = nn.ModuleList(Block(cfg, i) for i in range(self.lo, self.hi))
+        if self.is_last:
+            self.final_norm = Norm(cfg)
+            self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=cfg.lm_head_bias)
+            if cfg.tie_word_embeddings and self.is_first:
+                self.lm_head.weight = self.embed_tokens.weight
+        from ..ops.reference import rope_cos_sin as _rcs
+
+        if cfg.position_encoding == "rope":
+            rot = int(cfg.head_dim * cfg.rope_pct)
+            rot -= rot % 2
+            cos, sin = _rcs(cfg.max_position_embeddings, rot, cfg.rope_base)
+            self.register_buffer("rope_cos", cos, persistent=False)
+            self.register_buffer("rope_sin", sin, persistent=False)
+        else:
+            self.rope_cos = self.rope_sin = None
+
+    @property
+    def rope_tables(self):
+        return (self.rope_cos, self.rope_sin) if self.rope_cos is not None else None
+
+    def load_full_state_dict(self, sd: dict):
+        """Filter + reindex a full-model state dict onto this stage."""
+        own = {}
+        for k, v in sd.items():
+            if k.startswith("layers."):
+                idx = int(k.split(".")[1])
+                if self.lo <= idx < self.hi:
+                    own[f"layers.{idx - self.lo}." + k.split(".", 2)[2]] = v
+            elif k.startswith(("embed_tokens.", "embed_positions.")) and self.is_first:
+                own[k] = v
+            elif k.startswith(("final_norm.", "lm_head.")) and self.is_last:
+                own[k] = v
+        missing, unexpected = self.load_state_dict(own, strict=False)
+        return missing, unexpected
+
+    def make_context(self, input_ids, attention_mask):
+        from ..models.nn.transformer import AttentionContext
+
+        B, T = input_ids.shape[:2]
+        device = input_ids.device
+        key_starts = None
+        if attention_mask is not None:
+            mask = attention_mask.to(torch.int32)
+            position_ids = (mask.cumsum(-1) - 1).clamp(min=0).to(torch.int32)
+            key_starts = (T - mask.sum(-1)).to(torch.int32)
+        else:
+            position_ids = (torch.arange(T, device=device, dtype=torch.int32)
+                            .unsqueeze(0).expand(B, T).contiguous())
+        return AttentionContext(position_ids=position_ids, key_starts=key_starts, start_pos=0)
+
+    def forward(self, hidden_or_ids, ctx):
+        if self.is_first:
+            h = self.embed_tokens(hidden_or_ids)
+            if self.embed_positions is not None:
+                off = self.config.extra.get("position_offset", 0)
+                h = h + self.embed_positions(ctx.position_ids.long() + off)
+        else:
+            h = hidden_or_ids
+        for layer in self.layers:
+            h = layer(h, ctx, self.rope_tables)
+        if self.is_last:
+            h = self.final_norm(h)
+        return h
+
+    def project(self, h):
+        assert self.is_last
+        return self.lm_head(h)
+
+
+class PipelineRunner:
+    """GPipe fill-drain forward/backward over the PP group."""
+
+    def __init__(self, stage: PipelineStage, pp_group=None, pp_ranks: Optional[List[int]] = None):
+        self.stage = stage
+        self.group = pp_group
+        # global ranks of the pipeline stages in order
+        if pp_ranks is None:
+            pp_ranks = list(range(dist.get_world_size()))
+        self.pp_ranks = pp_ranks
+        self.idx = pp_ranks.index(dist.get_rank())
+        self.prev = pp_ranks[self.idx - 1] if self.idx > 0 else None
+        self.next = pp_ranks[self.idx + 1] if self.idx < len(pp_ranks) - 1 else None
+
+    def _send(self, t: torch.Tensor, dst: int):
+        dist.send(t.contiguous(), dst)
+
+    def _recv(self, shape, dtype, device, src: int) -> torch.Tensor:
+        t = torch.empty(shape, dtype=dtype, device=device)
+        dist.recv(t, src)
+        return t
+
+    def forward_backward(self, microbatches, loss_fn, hidden_dtype=torch.float32):
+        """Run fill-drain fwd+bwd.  ``microbatches``: list of dicts with
+        input_ids/attention_mask (used on stage 0 for embeddings and on every
+        stage for the attention context).  ``loss_fn(logits, mb) -> loss`` is
+        evaluated on the last stage.  Returns mean loss (last stage) or None.
+        Gradients accumulate into stage parameters; the caller averages over
+        microbatch count via the optimizer's grad scale."""
+        stage = self.stage
+        H = stage.config.hidden_size
+        device = next(stage.parameters()).device
+        stashes = []
+        losses = []
+
+        # ---- fill: forwards ------------------------------------------------
+        for mb in microbatches:
+            ids = mb["input_ids"].to(device)
+            mask = mb.get("attention_mask")
+            mask = mask.to(device) if mask is not None else None
+            ctx = stage.make_context(ids, mask)
+            if stage.is_first:
+                inp = ids
+                h_in = None
+            else:
+                B, T = ids.shape
+                h_in = self._recv((B, T, H), hidden_dtype, device, self.prev)
+                h_in.requires_grad_(True)
+                inp = h_in
+            out = stage(inp, ctx)
+            if stage.is_last:
+                logits = stage.project(out)
+                loss = loss_fn(logits, mb)
+                losses.append(loss)
+                stashes.append((h_in, loss))
+            else:
+                self._send(out.detach(), self.next)
+                stashes.append((h_in, out))
+
+        # ---- drain: backwards (reverse order) ------------------------------
+        for h_in, out_or_loss in reversed(stashes):
+            if stage.is_last:
+                out_or_loss.backward()
+            else:
+                grad = self._recv(out_or_loss.shape, hidden_dtype, device, self.next)
+                out_or_loss.backward(grad)
+            if not stage.is_first:
+                self._send(h_in.grad, self.prev)
+
+        if losses:
+            return torch.stack([l.detach() for l in losses]).mean()
+        return None
