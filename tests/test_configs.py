@@ -66,3 +66,17 @@ def test_canonical_ilql_hyperparameters():
     assert m.alpha == 0.001
     assert m.steps_for_target_q_sync == 5
     assert m.two_qs
+
+
+def test_all_shipped_configs_load():
+    """Every YAML under configs/ loads into TRLConfig
+    (parity: reference tests/test_configs.py over configs/**)."""
+    import glob
+    import os
+
+    root = os.path.join(os.path.dirname(__file__), "..", "configs")
+    paths = sorted(glob.glob(os.path.join(root, "*.yml")))
+    assert len(paths) >= 6
+    for p in paths:
+        cfg = TRLConfig.load_yaml(p)
+        assert cfg.train.seq_length > 0

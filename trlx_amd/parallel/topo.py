@@ -16,8 +16,11 @@ class ParallelState:
     pp_size: int = 1
     tp_group = None
     dp_group = None
+    pp_group = None
     tp_rank: int = 0
     dp_rank: int = 0
+    pp_rank: int = 0
+    pp_ranks = None
     dp_size: int = 1
     initialized: bool = False
 
@@ -45,19 +48,34 @@ def init_model_parallel(tp_size: int = 1, pp_size: int = 1) -> ParallelState:
 
     world = dist.get_world_size()
     rank = dist.get_rank()
-    assert world % tp_size == 0, f"world {world} not divisible by tp {tp_size}"
-    dp_size = world // tp_size
+    assert world % (tp_size * pp_size) == 0, \
+        f"world {world} not divisible by tp*pp {tp_size * pp_size}"
+    dp_size = world // (tp_size * pp_size)
+    block = tp_size * pp_size  # ranks per model replica
 
     # TP groups: contiguous rank blocks (intra-node xGMI locality)
-    for dp_idx in range(dp_size):
-        ranks = list(range(dp_idx * tp_size, (dp_idx + 1) * tp_size))
+    for b in range(world // tp_size):
+        ranks = list(range(b * tp_size, (b + 1) * tp_size))
         g = dist.new_group(ranks)
         if rank in ranks:
             s.tp_group = g
             s.tp_rank = ranks.index(rank)
-    # DP groups: same tp_idx across blocks
-    for tp_idx in range(tp_size):
-        ranks = list(range(tp_idx, world, tp_size))
+    # PP groups: stages tp_size apart within a replica block
+    s.pp_group = None
+    s.pp_rank = 0
+    s.pp_ranks = None
+    if pp_size > 1:
+        for dp_idx in range(dp_size):
+            for tp_idx in range(tp_size):
+                ranks = [dp_idx * block + p * tp_size + tp_idx for p in range(pp_size)]
+                g = dist.new_group(ranks)
+                if rank in ranks:
+                    s.pp_group = g
+                    s.pp_rank = ranks.index(rank)
+                    s.pp_ranks = ranks
+    # DP groups: same (pp_idx, tp_idx) across replica blocks
+    for off in range(block):
+        ranks = list(range(off, world, block))
         g = dist.new_group(ranks)
         if rank in ranks:
             s.dp_group = g
@@ -94,6 +112,18 @@ def dp_size() -> int:
     if _STATE.initialized:
         return _STATE.dp_size
     return dist.get_world_size() if dist.is_initialized() else 1
+
+
+def pp_size() -> int:
+    return _STATE.pp_size
+
+
+def pp_group():
+    return _STATE.pp_group
+
+
+def pp_rank() -> int:
+    return _STATE.pp_rank
 
 
 def reset():
