@@ -445,3 +445,44 @@ def _worker_pp_train(rank):
 
 def test_pipeline_parallel_equivalence():
     _spawn(_worker_pp_train, 29520)
+
+
+def _worker_tp_checkpoint(rank):
+    """TP save -> mp_rank_XX shards; single-process load merges them back."""
+    import os
+
+    from trlx_amd.models.modeling_ppo import AutoModelForCausalLMWithValueHead
+    from trlx_amd.models.nn.config import TransformerConfig
+    from trlx_amd.models.nn.transformer import CausalTransformer
+    from trlx_amd.parallel import topo
+    from trlx_amd.parallel.tp import shard_state_dict_tp
+
+    torch.manual_seed(0)
+    cfg = TransformerConfig(vocab_size=130, hidden_size=64, num_layers=2, num_heads=4,
+                            intermediate_size=128, max_position_embeddings=64,
+                            arch_name="gpt2", tie_word_embeddings=False)
+    full = CausalTransformer(cfg)
+    full_sd = {k: v for k, v in full.state_dict().items() if not k.startswith("rope_")}
+
+    out_dir = "/tmp/tp_ckpt_test"
+    topo.init_model_parallel(tp_size=WORLD)
+    try:
+        sharded = CausalTransformer(cfg)
+        sharded.load_state_dict(shard_state_dict_tp(full_sd, cfg, topo.tp_rank(), WORLD),
+                                strict=False)
+        model = AutoModelForCausalLMWithValueHead(sharded)
+        model.save_pretrained(out_dir)
+        dist.barrier()
+        assert os.path.exists(os.path.join(out_dir, f"mp_rank_{rank:02d}", "model_weights.pt"))
+    finally:
+        topo.reset()
+    dist.barrier()
+    if rank == 0:
+        loaded = AutoModelForCausalLMWithValueHead.from_pretrained(out_dir)
+        for k, v in full_sd.items():
+            got = loaded.base_model.state_dict()[k]
+            assert torch.allclose(got, v, atol=1e-6), k
+
+
+def test_tp_sharded_checkpoint_roundtrip():
+    _spawn(_worker_tp_checkpoint, 29521)

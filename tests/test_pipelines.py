@@ -151,3 +151,23 @@ def test_dialog_store_labels_mask_prompt(byte_tokenizer):
     # prompt positions are -100, output positions match input ids
     assert (labels[:2] == -100).all()
     assert (labels[labels != -100] == ids[labels != -100]).all()
+
+
+def test_rollout_json_export(tmp_path):
+    """Rollout JSON export for Algorithm Distillation
+    (reference ppo_pipeline.py:71-89)."""
+    import json
+
+    store = PPORolloutStorage(pad_token_id=0, padding_side="left")
+    store.clear_history()
+    store.push([
+        PPORLElement(torch.tensor([1, 2]), torch.tensor([3, 4]), torch.zeros(2),
+                     torch.zeros(2), torch.zeros(2))
+    ])
+    store.export_history(str(tmp_path))
+    files = list(tmp_path.glob("epoch-*.json"))
+    assert len(files) == 1
+    data = json.loads(files[0].read_text())
+    assert data[0]["query_tensor"] == [1, 2]
+    assert data[0]["response_tensor"] == [3, 4]
+    assert "logprobs" not in data[0]  # only_text=True

@@ -237,3 +237,21 @@ def test_ppo_loss_slice_matches_full_width(tmp_path):
         mask=attention_mask[:, start + 1 : end + 1],
     )
     assert torch.allclose(loss.detach(), want_loss.detach(), atol=1e-5), (loss, want_loss)
+
+
+def test_gen_kwarg_sweep_eval(tmp_path):
+    """A list-valued gen kwarg becomes an eval sweep
+    (reference accelerate_base_trainer.py:139-146, 339-360)."""
+    cfg = _tiny_model_cfg(default_ilql_config(), tmp_path)
+    cfg.train.save_best = False
+    cfg.method.gen_kwargs = dict(max_new_tokens=4, top_k=5, beta=[0, 1], temperature=1.0)
+
+    trainer = trlx_amd.train(
+        samples=["ab", "cd", "ef", "gh"],
+        rewards=[1.0, -1.0, 0.5, -0.5],
+        eval_prompts=["ab", "cd"] * 2,
+        metric_fn=lambda samples, **kw: {"len": [float(len(s)) for s in samples]},
+        config=cfg,
+    )
+    assert trainer.generate_sweep_kwarg is not None
+    assert trainer.iter_count == 2

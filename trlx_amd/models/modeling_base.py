@@ -78,6 +78,40 @@ class PreTrainedModelWrapper(nn.Module):
             base = pretrained_model_name_or_path
         elif isinstance(pretrained_model_name_or_path, PreTrainedModelWrapper):
             base = pretrained_model_name_or_path.base_model
+        elif os.path.isdir(pretrained_model_name_or_path) and os.path.exists(
+                os.path.join(pretrained_model_name_or_path, "mp_rank_00")):
+            # TP-sharded checkpoint: load this rank's shard (same TP size) or
+            # merge all shards into a full model (TP off)
+            from ..parallel import topo
+
+            with open(os.path.join(pretrained_model_name_or_path, "config.json")) as f:
+                hf_cfg = json.load(f)
+            from .nn.convert import config_from_hf
+
+            cfg = config_from_hf(hf_cfg)
+            saved_tp = int(hf_cfg.get("trlx_amd_tp_size", 1))
+            base = CausalTransformer(cfg)
+            model = cls(base, **wrapped_kwargs)
+            if topo.tp_size() == saved_tp and saved_tp > 1:
+                shard = torch.load(
+                    os.path.join(pretrained_model_name_or_path,
+                                 f"mp_rank_{topo.tp_rank():02d}", "model_weights.pt"),
+                    map_location="cpu", weights_only=True)
+                model.load_state_dict(shard, strict=False)
+            elif topo.tp_size() == 1:
+                from ..parallel.tp import merge_state_dicts_tp
+
+                shards = [
+                    torch.load(os.path.join(pretrained_model_name_or_path,
+                                            f"mp_rank_{r:02d}", "model_weights.pt"),
+                               map_location="cpu", weights_only=True)
+                    for r in range(saved_tp)
+                ]
+                model.load_state_dict(merge_state_dicts_tp(shards, cfg, saved_tp), strict=False)
+            else:
+                raise ValueError(
+                    f"checkpoint TP size {saved_tp} != current TP {topo.tp_size()}")
+            return model
         elif os.path.isdir(pretrained_model_name_or_path):
             cfg, sd = load_hf_dir(pretrained_model_name_or_path)
             # wrapper head weights (if this directory was saved by a wrapper)
@@ -144,8 +178,24 @@ class PreTrainedModelWrapper(nn.Module):
         """Write an HF directory for the base model + wrapper head weights.
         Under LoRA, the adapter is saved separately (adapter_model.pt, the
         reference's peft behavior — modeling_base.py:328-355) and the exported
-        base weights exclude the adapter."""
+        base weights exclude the adapter.  Under tensor parallelism each TP
+        rank writes its shard to mp_rank_XX/ (the NeMo format,
+        modeling_nemo_ppo.py:445-467); from_pretrained reassembles."""
         from .lora import has_lora, lora_state_dict
+        from ..parallel import topo
+
+        if topo.tp_size() > 1:
+            os.makedirs(save_directory, exist_ok=True)
+            shard_dir = os.path.join(save_directory, f"mp_rank_{topo.tp_rank():02d}")
+            os.makedirs(shard_dir, exist_ok=True)
+            torch.save({k: v.cpu() for k, v in self.state_dict().items()
+                        if not ("rope_cos" in k or "rope_sin" in k)},
+                       os.path.join(shard_dir, "model_weights.pt"))
+            if topo.tp_rank() == 0:
+                with open(os.path.join(save_directory, "config.json"), "w") as f:
+                    json.dump({**config_to_hf(self.config),
+                               "trlx_amd_tp_size": topo.tp_size()}, f, indent=2)
+            return
 
         os.makedirs(save_directory, exist_ok=True)
         base_sd = {}

@@ -267,3 +267,32 @@ def shard_state_dict_tp(sd: dict, cfg, tp_rank: int, tp: int) -> dict:
         else:
             out[k] = v
     return out
+
+
+def merge_state_dicts_tp(shards: list, cfg, tp: int) -> dict:
+    """Reassemble a full state dict from per-TP-rank shards (inverse of
+    shard_state_dict_tp; accepts wrapper-prefixed keys)."""
+    import torch
+
+    out = {}
+    for k in shards[0]:
+        vs = [sd[k] for sd in shards]
+        if ".attn.qkv_proj." in k:
+            qd_l = cfg.num_heads // tp * cfg.head_dim
+            kd_l = cfg.num_kv_heads // tp * cfg.head_dim
+            qs = [v[:qd_l] for v in vs]
+            ks = [v[qd_l : qd_l + kd_l] for v in vs]
+            vvs = [v[qd_l + kd_l :] for v in vs]
+            out[k] = torch.cat(qs + ks + vvs, dim=0)
+        elif ".mlp.gate_up_proj." in k:
+            i_l = cfg.intermediate_size // tp
+            gates = [v[:i_l] for v in vs]
+            ups = [v[i_l:] for v in vs]
+            out[k] = torch.cat(gates + ups, dim=0)
+        elif ".attn.o_proj.weight" in k or ".mlp.down_proj.weight" in k:
+            out[k] = torch.cat(vs, dim=1)
+        elif ".mlp.fc_in." in k:
+            out[k] = torch.cat(vs, dim=0)
+        else:
+            out[k] = vs[0]
+    return out
