@@ -1,6 +1,11 @@
 #!/usr/bin/env python3
 """Flagship benchmark: PPO samples/sec, GPT-2-small "sentiments" shape.
 
+``--method ilql`` instead measures offline ILQL training samples/sec
+(BASELINE.json config #2: GPT-2-small ILQL sentiments, bf16, 1 GPU): one
+step = one optimizer step over a batch of reward-labeled sequences +
+periodic target-Q Polyak sync.
+
 BASELINE.json metric: "PPO samples/sec (whole node), GPT-2 sentiments" with
 the reference's canonical hyperparameters as the algorithmic anchor
 (default_configs: num_rollouts=128, chunk_size=128, ppo_epochs=4, batch 32,
@@ -115,6 +120,7 @@ def main():
     p.add_argument("--num-layers-unfrozen", type=int, default=2)
     p.add_argument("--num-prompts", type=int, default=512)
     p.add_argument("--phases", action="store_true", help="print per-phase times (experience vs train)")
+    p.add_argument("--method", choices=["ppo", "ilql"], default="ppo")
     args = p.parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
@@ -123,6 +129,9 @@ def main():
 
     import trlx_amd  # noqa: F401  (also loads the HIP extension path)
     from trlx_amd.parallel import comm
+
+    if args.method == "ilql":
+        return run_ilql(args)
 
     trainer, config = build_trainer(args)
 
@@ -185,6 +194,85 @@ def main():
             },
         }
         print(json.dumps(result))
+
+
+def run_ilql(args):
+    import trlx_amd
+    from trlx_amd.data.default_configs import default_ilql_config
+    from trlx_amd.models.nn.config import preset
+    from trlx_amd.parallel import comm
+    from trlx_amd.utils.loading import get_trainer
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    config = default_ilql_config()
+    config.model.model_path = args.model
+    config.model.model_extra_configs = {"config": preset(args.model).to_dict()}
+    config.tokenizer.tokenizer_path = "synthetic"
+    config.train.seq_length = 64
+    config.train.batch_size = 128
+    config.train.tracker = None
+    config.train.total_steps = 10**9
+    trainer = get_trainer("ILQLTrainer")(config=config)
+
+    torch.manual_seed(1234 + rank)
+    # synthetic reward-labeled samples at the canonical ILQL shape (seq 64)
+    toks = torch.randint(3, 50257, (512, 60)).tolist()
+    samples = [" ".join(f"t{t}" for t in row) for row in toks]
+    rewards = [((i * 2654435761) % 1000) / 1000.0 - 0.5 for i in range(len(samples))]
+    trainer.make_experience(samples, rewards, config.train.seq_length)
+    loader = trainer.store.create_loader(config.train.batch_size)
+
+    def sync():
+        comm.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    def run_steps(n):
+        it = iter(loader)
+        for i in range(n):
+            try:
+                batch = next(it)
+            except StopIteration:
+                it = iter(loader)
+                batch = next(it)
+            loss, _ = trainer.loss(batch)
+            trainer.model.train()
+            loss.backward()
+            trainer.reducer.finalize()
+            trainer.opt.step()
+            trainer.opt.zero_grad()
+            if (i + 1) % config.method.steps_for_target_q_sync == 0:
+                trainer.unwrapped_model.sync_target_q_heads()
+
+    run_steps(args.warmup * 4)
+    sync()
+    t0 = time.time()
+    run_steps(args.steps * 4)
+    sync()
+    elapsed = time.time() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=comm.get_device())
+        torch.distributed.all_reduce(t, torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+    total = config.train.batch_size * args.steps * 4 * world
+    if rank == 0:
+        print(json.dumps({
+            "metric": "ilql_samples_per_sec",
+            "value": round(total / elapsed, 3),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps * 4,
+            "warmup": args.warmup * 4,
+            "ms_per_step": round(1000.0 * elapsed / (args.steps * 4), 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if torch.cuda.is_available() else "fp32",
+            "data": "synthetic reward-labeled sequences (vocab 50257), random-init weights",
+            "config": {"model": "gpt2-small-124M", "global_batch": config.train.batch_size * world,
+                       "seq_len": 64, "two_qs": True, "parallelism": f"dp{world}"},
+        }))
 
 
 if __name__ == "__main__":

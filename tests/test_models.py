@@ -276,3 +276,23 @@ def test_graph_decode_sampled_runs_and_is_fresh():
     assert a.shape == (4, 14)
     # different RNG offsets across calls -> different samples
     assert not torch.equal(a[:, 6:], b[:, 6:])
+
+
+@pytest.mark.gpu
+def test_ilql_shaped_generate_uses_graph_on_gpu():
+    torch.manual_seed(0)
+    cfg = tiny_config(vocab_size=500, hidden_size=128, num_heads=2)
+    model = AutoModelForCausalLMWithILQLHeads.from_config(cfg)
+    model = model.cuda()
+    model.cast_compute(torch.bfloat16)
+    model.eval()
+    ids = torch.randint(3, 500, (4, 6)).cuda()
+    out1 = model.generate(ids, max_new_tokens=8, beta=2, top_k=10, temperature=1.0,
+                          eos_token_id=1, pad_token_id=2)
+    assert out1.shape[0] == 4 and out1.shape[1] <= 14
+    # engine captured and reused
+    assert getattr(model.base_model, "_decode_engine", None) is not None
+    assert model.base_model._decode_engine.graph is not None
+    out2 = model.generate(ids, max_new_tokens=8, beta=2, top_k=10, temperature=1.0,
+                          eos_token_id=1, pad_token_id=2)
+    assert out2.shape[0] == 4

@@ -239,6 +239,12 @@ class AutoModelForCausalLMWithILQLHeads(PreTrainedModelWrapper):
             return pi_beta + beta * adv
 
         max_new_tokens = min(max_new_tokens, max_length - input_ids.shape[1])
+        # the shaping fn is pure torch ops -> capturable into the decode graph;
+        # keep one fn per (beta, mask) so the engine is reused across calls
+        key = (float(beta), id(logit_mask))
+        if getattr(self, "_shaping_key", None) != key:
+            self._shaping_key = key
+            self._shaping_fn = shaping_fn
         gen = GenerateConfig(
             max_new_tokens=max_new_tokens,
             do_sample=temperature > 0,
@@ -246,5 +252,7 @@ class AutoModelForCausalLMWithILQLHeads(PreTrainedModelWrapper):
             top_k=top_k,
             eos_token_id=eos_token_id,
             pad_token_id=eos_token_id if eos_token_id is not None else pad_token_id,
+            graph_safe_shaping=True,
         )
-        return generate(self.base_model, input_ids, attention_mask, gen=gen, shaping_fn=shaping_fn)
+        return generate(self.base_model, input_ids, attention_mask, gen=gen,
+                        shaping_fn=self._shaping_fn)

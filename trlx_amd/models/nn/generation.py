@@ -39,6 +39,9 @@ class GenerateConfig:
     pad_token_id: Optional[int] = None
     seed: Optional[int] = None
     use_graph: bool = True
+    # shaping fns made of pure torch ops (ILQL heads, logit masks) may be
+    # captured into the decode graph
+    graph_safe_shaping: bool = False
 
     @classmethod
     def from_kwargs(cls, **kwargs) -> "GenerateConfig":
@@ -56,8 +59,9 @@ class DecodeEngine:
     """Persistent hipGraph-captured decode step."""
 
     def __init__(self, model, batch: int, cache_len: int, max_new: int, gen: GenerateConfig,
-                 device):
+                 device, shaping_fn=None):
         self.model = model
+        self.shaping_fn = shaping_fn
         self.batch = batch
         self.cache_len = cache_len
         self.max_new = max_new
@@ -107,6 +111,8 @@ class DecodeEngine:
             return_logits=False,
         )
         logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
+        if self.shaping_fn is not None:
+            logits = self.shaping_fn(logits, out.last_hidden_state[:, -1], self.cur_tok[:, 0])
         tok = self._sample(logits)
         self._advance(tok)
         self.cache_idx += 1
@@ -153,6 +159,8 @@ class DecodeEngine:
         out = self.model(input_ids, attention_mask=attention_mask, kv_cache=self.kv,
                          start_pos=0, return_logits=False)
         logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
+        if self.shaping_fn is not None:
+            logits = self.shaping_fn(logits, out.last_hidden_state[:, -1], input_ids[:, -1])
         tok = self._sample(logits)
         self._advance(tok)
 
@@ -193,12 +201,15 @@ def generate(
         attention_mask = torch.ones_like(input_ids)
 
     try:
-        if (device.type == "cuda" and shaping_fn is None and gen.use_graph
+        shaping_graphable = shaping_fn is None or gen.graph_safe_shaping
+        if (device.type == "cuda" and shaping_graphable and gen.use_graph
                 and _graphs_enabled() and gen.max_new_tokens > 1 and gen.min_new_tokens == 0):
             engine = getattr(model, "_decode_engine", None)
             needed = T + gen.max_new_tokens
-            if engine is None or not engine.matches(B, needed, gen.max_new_tokens, gen):
-                engine = DecodeEngine(model, B, needed, gen.max_new_tokens, gen, device)
+            if (engine is None or not engine.matches(B, needed, gen.max_new_tokens, gen)
+                    or engine.shaping_fn is not shaping_fn):
+                engine = DecodeEngine(model, B, needed, gen.max_new_tokens, gen, device,
+                                      shaping_fn=shaping_fn)
                 model._decode_engine = engine
             return engine.run(input_ids, attention_mask, gen.max_new_tokens)
         return _generate_eager(model, input_ids, attention_mask, gen, shaping_fn)
