@@ -182,7 +182,7 @@ def main():
             "dtype": "bf16" if torch.cuda.is_available() else "fp32",
             "data": "synthetic prompts (vocab 50257), random-init weights, deterministic synthetic reward",
             "config": {
-                "model": "gpt2-small-124M",
+                "model": args.model,
                 "global_batch": args.batch_size * world,
                 "seq_len": args.seq_len,
                 "prompt_len": args.prompt_len,
@@ -270,7 +270,7 @@ def run_ilql(args):
             "vs_baseline": None,
             "dtype": "bf16" if torch.cuda.is_available() else "fp32",
             "data": "synthetic reward-labeled sequences (vocab 50257), random-init weights",
-            "config": {"model": "gpt2-small-124M", "global_batch": config.train.batch_size * world,
+            "config": {"model": args.model, "global_batch": config.train.batch_size * world,
                        "seq_len": 64, "two_qs": True, "parallelism": f"dp{world}"},
         }))
 

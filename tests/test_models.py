@@ -177,7 +177,7 @@ def test_kl_controllers():
 # --- HF interop ------------------------------------------------------------
 
 
-@pytest.mark.parametrize("family", ["gpt2", "llama", "gptj", "gpt_neox", "opt"])
+@pytest.mark.parametrize("family", ["gpt2", "llama", "gptj", "gpt_neox", "opt", "bloom", "gpt_bigcode"])
 def test_hf_equivalence(family):
     """Native forward must match HF transformers bit-for-tolerance on the same
     random weights — the no-network substitute for real-checkpoint tests."""
@@ -210,6 +210,15 @@ def test_hf_equivalence(family):
                                         max_position_embeddings=64, dropout=0.0,
                                         word_embed_proj_dim=H)
         hf = transformers.OPTForCausalLM(hf_cfg)
+    elif family == "bloom":
+        hf_cfg = transformers.BloomConfig(vocab_size=V, hidden_size=H, n_layer=L, n_head=NH,
+                                          hidden_dropout=0.0, attention_dropout=0.0)
+        hf = transformers.BloomForCausalLM(hf_cfg)
+    elif family == "gpt_bigcode":
+        hf_cfg = transformers.GPTBigCodeConfig(vocab_size=V, n_embd=H, n_layer=L, n_head=NH,
+                                               n_positions=64, multi_query=True,
+                                               resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0)
+        hf = transformers.GPTBigCodeForCausalLM(hf_cfg)
     hf = hf.eval()
 
     cfg = config_from_hf(hf_cfg.to_dict())

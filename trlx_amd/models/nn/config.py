@@ -127,6 +127,17 @@ PRESETS = {
         rope_interleaved=False, rope_pct=0.25, activation="gelu",
         parallel_residual=True, tie_word_embeddings=False, arch_name="gpt_neox",
     ),
+    "bloom-560m": dict(
+        vocab_size=250880, hidden_size=1024, num_layers=24, num_heads=16,
+        max_position_embeddings=2048, norm="layernorm", position_encoding="alibi",
+        activation="gelu", arch_name="bloom", tie_word_embeddings=True,
+        extra={"pre_embed_norm": True},
+    ),
+    "gpt_bigcode-santacoder": dict(
+        vocab_size=49280, hidden_size=2048, num_layers=24, num_heads=16,
+        num_kv_heads=1, max_position_embeddings=2048, norm="layernorm",
+        position_encoding="learned", activation="gelu_new", arch_name="gpt_bigcode",
+    ),
     "opt-125m": dict(
         vocab_size=50272, hidden_size=768, num_layers=12, num_heads=12,
         max_position_embeddings=2048, norm="layernorm", position_encoding="learned",

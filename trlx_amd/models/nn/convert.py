@@ -126,6 +126,39 @@ def config_from_hf(hf: dict) -> TransformerConfig:
             arch_name="opt",
             extra={"position_offset": 2},
         )
+    if model_type == "bloom":
+        H = hf["n_head"]
+        return TransformerConfig(
+            vocab_size=hf["vocab_size"],
+            hidden_size=hf.get("hidden_size", hf.get("n_embd")),
+            num_layers=hf.get("n_layer", hf.get("num_hidden_layers")),
+            num_heads=H,
+            max_position_embeddings=hf.get("seq_length", 2048),
+            norm="layernorm",
+            norm_eps=hf.get("layer_norm_epsilon", 1e-5),
+            position_encoding="alibi",
+            activation="gelu_new",
+            tie_word_embeddings=True,
+            arch_name="bloom",
+            extra={"pre_embed_norm": True},
+        )
+    if model_type == "gpt_bigcode":
+        return TransformerConfig(
+            vocab_size=hf["vocab_size"],
+            hidden_size=hf["n_embd"],
+            num_layers=hf["n_layer"],
+            num_heads=hf["n_head"],
+            num_kv_heads=1 if hf.get("multi_query", True) else hf["n_head"],
+            intermediate_size=hf.get("n_inner") or 4 * hf["n_embd"],
+            max_position_embeddings=hf["n_positions"],
+            norm="layernorm",
+            norm_eps=hf.get("layer_norm_epsilon", 1e-5),
+            position_encoding="learned",
+            activation={"gelu_new": "gelu_new", "gelu": "gelu",
+                        "gelu_pytorch_tanh": "gelu_new"}.get(hf.get("activation_function", "gelu_pytorch_tanh"), "gelu_new"),
+            tie_word_embeddings=hf.get("tie_word_embeddings", True),
+            arch_name="gpt_bigcode",
+        )
     raise ValueError(f"Unsupported HF model_type for the native transformer: {model_type}")
 
 
@@ -197,6 +230,32 @@ def config_to_hf(cfg: TransformerConfig) -> dict:
             "rotary_emb_base": cfg.rope_base,
             "hidden_act": cfg.activation,
             "use_parallel_residual": cfg.parallel_residual,
+            "tie_word_embeddings": cfg.tie_word_embeddings,
+        }
+    if cfg.arch_name == "bloom":
+        return {
+            "model_type": "bloom",
+            "architectures": ["BloomForCausalLM"],
+            "vocab_size": cfg.vocab_size,
+            "hidden_size": cfg.hidden_size,
+            "n_layer": cfg.num_layers,
+            "n_head": cfg.num_heads,
+            "layer_norm_epsilon": cfg.norm_eps,
+            "tie_word_embeddings": True,
+        }
+    if cfg.arch_name == "gpt_bigcode":
+        return {
+            "model_type": "gpt_bigcode",
+            "architectures": ["GPTBigCodeForCausalLM"],
+            "vocab_size": cfg.vocab_size,
+            "n_embd": cfg.hidden_size,
+            "n_layer": cfg.num_layers,
+            "n_head": cfg.num_heads,
+            "n_inner": cfg.intermediate_size,
+            "n_positions": cfg.max_position_embeddings,
+            "layer_norm_epsilon": cfg.norm_eps,
+            "activation_function": "gelu_pytorch_tanh",
+            "multi_query": cfg.num_kv_heads == 1,
             "tie_word_embeddings": cfg.tie_word_embeddings,
         }
     if cfg.arch_name == "opt":
@@ -320,6 +379,55 @@ def state_dict_from_hf(cfg: TransformerConfig, hf: Dict[str, torch.Tensor]) -> D
         out["final_norm.bias"] = hf["gpt_neox.final_layer_norm.bias"]
         # transformers >=5 uses lm_head.weight; older NeoX exports embed_out
         out["lm_head.weight"] = hf.get("lm_head.weight", hf.get("embed_out.weight"))
+    elif a == "bloom":
+        H, D = cfg.num_heads, cfg.head_dim
+        out["embed_tokens.weight"] = hf["transformer.word_embeddings.weight"]
+        out["embed_norm.weight"] = hf["transformer.word_embeddings_layernorm.weight"]
+        out["embed_norm.bias"] = hf["transformer.word_embeddings_layernorm.bias"]
+        for i in range(L):
+            p = f"transformer.h.{i}."
+            o = f"layers.{i}."
+            out[o + "ln_1.weight"] = hf[p + "input_layernorm.weight"]
+            out[o + "ln_1.bias"] = hf[p + "input_layernorm.bias"]
+            out[o + "ln_2.weight"] = hf[p + "post_attention_layernorm.weight"]
+            out[o + "ln_2.bias"] = hf[p + "post_attention_layernorm.bias"]
+            # bloom fuses qkv per-head [H, 3, D, hidden] like NeoX
+            w = hf[p + "self_attention.query_key_value.weight"].view(H, 3, D, -1)
+            out[o + "attn.qkv_proj.weight"] = w.permute(1, 0, 2, 3).reshape(3 * H * D, -1).contiguous()
+            b = hf[p + "self_attention.query_key_value.bias"].view(H, 3, D)
+            out[o + "attn.qkv_proj.bias"] = b.permute(1, 0, 2).reshape(3 * H * D).contiguous()
+            out[o + "attn.o_proj.weight"] = hf[p + "self_attention.dense.weight"]
+            out[o + "attn.o_proj.bias"] = hf[p + "self_attention.dense.bias"]
+            out[o + "mlp.fc_in.weight"] = hf[p + "mlp.dense_h_to_4h.weight"]
+            out[o + "mlp.fc_in.bias"] = hf[p + "mlp.dense_h_to_4h.bias"]
+            out[o + "mlp.down_proj.weight"] = hf[p + "mlp.dense_4h_to_h.weight"]
+            out[o + "mlp.down_proj.bias"] = hf[p + "mlp.dense_4h_to_h.bias"]
+        out["final_norm.weight"] = hf["transformer.ln_f.weight"]
+        out["final_norm.bias"] = hf["transformer.ln_f.bias"]
+    elif a == "gpt_bigcode":
+        out["embed_tokens.weight"] = hf["transformer.wte.weight"]
+        out["embed_positions.weight"] = hf["transformer.wpe.weight"]
+        for i in range(L):
+            p = f"transformer.h.{i}."
+            o = f"layers.{i}."
+            out[o + "ln_1.weight"] = hf[p + "ln_1.weight"]
+            out[o + "ln_1.bias"] = hf[p + "ln_1.bias"]
+            out[o + "ln_2.weight"] = hf[p + "ln_2.weight"]
+            out[o + "ln_2.bias"] = hf[p + "ln_2.bias"]
+            # gpt_bigcode uses plain Linear (NOT Conv1D): no transpose; MQA
+            # layout [Hq*D + 2*D, hidden] matches the native fused qkv
+            out[o + "attn.qkv_proj.weight"] = hf[p + "attn.c_attn.weight"]
+            out[o + "attn.qkv_proj.bias"] = hf[p + "attn.c_attn.bias"]
+            out[o + "attn.o_proj.weight"] = hf[p + "attn.c_proj.weight"]
+            out[o + "attn.o_proj.bias"] = hf[p + "attn.c_proj.bias"]
+            out[o + "mlp.fc_in.weight"] = hf[p + "mlp.c_fc.weight"]
+            out[o + "mlp.fc_in.bias"] = hf[p + "mlp.c_fc.bias"]
+            out[o + "mlp.down_proj.weight"] = hf[p + "mlp.c_proj.weight"]
+            out[o + "mlp.down_proj.bias"] = hf[p + "mlp.c_proj.bias"]
+        out["final_norm.weight"] = hf["transformer.ln_f.weight"]
+        out["final_norm.bias"] = hf["transformer.ln_f.bias"]
+        if not cfg.tie_word_embeddings and "lm_head.weight" in hf:
+            out["lm_head.weight"] = hf["lm_head.weight"]
     elif a == "opt":
         out["embed_tokens.weight"] = hf["model.decoder.embed_tokens.weight"]
         out["embed_positions.weight"] = hf["model.decoder.embed_positions.weight"]
@@ -440,6 +548,52 @@ def state_dict_to_hf(cfg: TransformerConfig, sd: Dict[str, torch.Tensor]) -> Dic
         out["gpt_neox.final_layer_norm.weight"] = sd["final_norm.weight"]
         out["gpt_neox.final_layer_norm.bias"] = sd["final_norm.bias"]
         out["lm_head.weight"] = sd["lm_head.weight"]
+    elif a == "bloom":
+        H, D = cfg.num_heads, cfg.head_dim
+        out["transformer.word_embeddings.weight"] = sd["embed_tokens.weight"]
+        out["transformer.word_embeddings_layernorm.weight"] = sd["embed_norm.weight"]
+        out["transformer.word_embeddings_layernorm.bias"] = sd["embed_norm.bias"]
+        for i in range(L):
+            p = f"transformer.h.{i}."
+            o = f"layers.{i}."
+            out[p + "input_layernorm.weight"] = sd[o + "ln_1.weight"]
+            out[p + "input_layernorm.bias"] = sd[o + "ln_1.bias"]
+            out[p + "post_attention_layernorm.weight"] = sd[o + "ln_2.weight"]
+            out[p + "post_attention_layernorm.bias"] = sd[o + "ln_2.bias"]
+            w = sd[o + "attn.qkv_proj.weight"].view(3, H, D, -1)
+            out[p + "self_attention.query_key_value.weight"] = w.permute(1, 0, 2, 3).reshape(3 * H * D, -1).contiguous()
+            b = sd[o + "attn.qkv_proj.bias"].view(3, H, D)
+            out[p + "self_attention.query_key_value.bias"] = b.permute(1, 0, 2).reshape(3 * H * D).contiguous()
+            out[p + "self_attention.dense.weight"] = sd[o + "attn.o_proj.weight"]
+            out[p + "self_attention.dense.bias"] = sd[o + "attn.o_proj.bias"]
+            out[p + "mlp.dense_h_to_4h.weight"] = sd[o + "mlp.fc_in.weight"]
+            out[p + "mlp.dense_h_to_4h.bias"] = sd[o + "mlp.fc_in.bias"]
+            out[p + "mlp.dense_4h_to_h.weight"] = sd[o + "mlp.down_proj.weight"]
+            out[p + "mlp.dense_4h_to_h.bias"] = sd[o + "mlp.down_proj.bias"]
+        out["transformer.ln_f.weight"] = sd["final_norm.weight"]
+        out["transformer.ln_f.bias"] = sd["final_norm.bias"]
+        out["lm_head.weight"] = sd.get("lm_head.weight", sd["embed_tokens.weight"])
+    elif a == "gpt_bigcode":
+        out["transformer.wte.weight"] = sd["embed_tokens.weight"]
+        out["transformer.wpe.weight"] = sd["embed_positions.weight"]
+        for i in range(L):
+            p = f"transformer.h.{i}."
+            o = f"layers.{i}."
+            out[p + "ln_1.weight"] = sd[o + "ln_1.weight"]
+            out[p + "ln_1.bias"] = sd[o + "ln_1.bias"]
+            out[p + "ln_2.weight"] = sd[o + "ln_2.weight"]
+            out[p + "ln_2.bias"] = sd[o + "ln_2.bias"]
+            out[p + "attn.c_attn.weight"] = sd[o + "attn.qkv_proj.weight"]
+            out[p + "attn.c_attn.bias"] = sd[o + "attn.qkv_proj.bias"]
+            out[p + "attn.c_proj.weight"] = sd[o + "attn.o_proj.weight"]
+            out[p + "attn.c_proj.bias"] = sd[o + "attn.o_proj.bias"]
+            out[p + "mlp.c_fc.weight"] = sd[o + "mlp.fc_in.weight"]
+            out[p + "mlp.c_fc.bias"] = sd[o + "mlp.fc_in.bias"]
+            out[p + "mlp.c_proj.weight"] = sd[o + "mlp.down_proj.weight"]
+            out[p + "mlp.c_proj.bias"] = sd[o + "mlp.down_proj.bias"]
+        out["transformer.ln_f.weight"] = sd["final_norm.weight"]
+        out["transformer.ln_f.bias"] = sd["final_norm.bias"]
+        out["lm_head.weight"] = sd.get("lm_head.weight", sd["embed_tokens.weight"])
     elif a == "opt":
         qd = cfg.num_heads * cfg.head_dim
         out["model.decoder.embed_tokens.weight"] = sd["embed_tokens.weight"]

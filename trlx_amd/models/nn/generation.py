@@ -202,7 +202,8 @@ def generate(
 
     try:
         shaping_graphable = shaping_fn is None or gen.graph_safe_shaping
-        if (device.type == "cuda" and shaping_graphable and gen.use_graph
+        alibi = getattr(model.config, "position_encoding", None) == "alibi"
+        if (device.type == "cuda" and shaping_graphable and gen.use_graph and not alibi
                 and _graphs_enabled() and gen.max_new_tokens > 1 and gen.min_new_tokens == 0):
             engine = getattr(model, "_decode_engine", None)
             needed = T + gen.max_new_tokens

@@ -50,6 +50,21 @@ class Norm(nn.Module):
         return ops.layernorm(x, self.weight, self.bias, self.eps)
 
 
+def alibi_slopes(num_heads: int) -> torch.Tensor:
+    """Standard ALiBi head slopes (Bloom; closest-power-of-two scheme)."""
+
+    def pow2_slopes(n):
+        start = 2.0 ** (-(2.0 ** -(math.log2(n) - 3)))
+        return [start * (start ** i) for i in range(n)]
+
+    if math.log2(num_heads).is_integer():
+        vals = pow2_slopes(num_heads)
+    else:
+        closest = 2 ** math.floor(math.log2(num_heads))
+        vals = pow2_slopes(closest) + pow2_slopes(2 * closest)[0::2][: num_heads - closest]
+    return torch.tensor(vals, dtype=torch.float32)
+
+
 @dataclass
 class AttentionContext:
     """Per-forward positional/masking state shared by all layers."""
@@ -61,6 +76,8 @@ class AttentionContext:
     # device scalar cache write index: enables the fused decode_prep kernel
     # and hipGraph capture (no host-side position state)
     cache_idx: Optional[torch.Tensor] = None
+    # ALiBi additive bias [B, H_local, 1, T_keys] (Bloom)
+    alibi: Optional[torch.Tensor] = None
 
 
 class KVCache:
@@ -124,7 +141,7 @@ class Attention(nn.Module):
         # fused single-token decode path: decode_prep (split+RoPE+cache
         # append) + flash-decode attention, fully device-side (hipGraph-safe)
         if (T == 1 and kv_cache is not None and ctx.cache_idx is not None
-                and ctx.seq_lens is not None and x.is_cuda):
+                and ctx.seq_lens is not None and x.is_cuda and ctx.alibi is None):
             cos_sin = rope_tables if self.cfg.position_encoding == "rope" else (None, None)
             q = ops.decode_prep(
                 qkv, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx], ctx.cache_idx,
@@ -179,6 +196,8 @@ class Attention(nn.Module):
             v = v.repeat_interleave(rep, dim=1)
         qs = q if pre_scaled else q * self.scale
         scores = torch.matmul(qs, k.transpose(-1, -2))
+        if ctx.alibi is not None:
+            scores = scores + ctx.alibi[:, :, :, : scores.shape[-1]].to(scores.dtype)
         probs = ops.causal_softmax(scores.contiguous(), ctx.start_pos, ctx.key_starts)
         if self.attn_pdrop > 0 and self.training:
             probs = F.dropout(probs, self.attn_pdrop)
@@ -272,6 +291,7 @@ class CausalTransformer(nn.Module):
             self.embed_positions = nn.Embedding(cfg.max_position_embeddings + off, cfg.hidden_size)
         else:
             self.embed_positions = None
+        self.embed_norm = Norm(cfg) if cfg.extra.get("pre_embed_norm") else None
         self.layers = nn.ModuleList(Block(cfg, i) for i in range(cfg.num_layers))
         self.final_norm = Norm(cfg)
         self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=cfg.lm_head_bias)
@@ -386,11 +406,29 @@ class CausalTransformer(nn.Module):
                                 position_ids)
         ctx.cache_idx = cache_idx
         h = self.embed_tokens(input_ids)
+        if self.embed_norm is not None:
+            h = self.embed_norm(h)
         if self.embed_positions is not None:
             off = self.config.extra.get("position_offset", 0)
             h = h + self.embed_positions(ctx.position_ids.long() + off)
         if self.embd_pdrop > 0 and self.training:
             h = F.dropout(h, self.embd_pdrop)
+
+        if self.config.position_encoding == "alibi":
+            # ALiBi bias per KEY position (Bloom semantics: mask-aware
+            # positions, pads at bias 0); broadcasts over queries and covers
+            # the cached keys during decode
+            tp = topo.tp_size()
+            slopes = alibi_slopes(self.config.num_heads).to(h.device)
+            if tp > 1:
+                hl = self.config.num_heads // tp
+                slopes = slopes[topo.tp_rank() * hl : (topo.tp_rank() + 1) * hl]
+            B, T = input_ids.shape[:2]
+            Tk = start_pos + T
+            j = torch.arange(Tk, device=h.device).unsqueeze(0)
+            ks = ctx.key_starts.unsqueeze(1) if ctx.key_starts is not None else torch.zeros(B, 1, device=h.device)
+            key_pos = (j - ks).clamp(min=0).float()  # [B, Tk]
+            ctx.alibi = slopes.view(1, -1, 1, 1) * key_pos[:, None, None, :]
 
         # sequence parallelism: shard activations along T for the
         # norm/residual segments (SP stays off for KV-cached generation)
