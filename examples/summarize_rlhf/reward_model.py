@@ -1,0 +1,125 @@
+"""Pairwise-comparison reward model on the native transformer.
+
+Parity: reference examples/summarize_rlhf/reward_model/reward_model.py
+(GPTRewardModel).  Differences, MI355X-first:
+
+- the trunk is the native ``CausalTransformer`` (fused HIP kernels), not an
+  HF module;
+- the pairwise loss (reference reward_model.py:61-90 — a per-row Python loop
+  over divergence/end indices with ``.item()`` syncs) is fully vectorized:
+  divergence index, end index and the divergence..end mask are computed as
+  batched tensor ops, so one loss evaluation is a handful of kernels with no
+  host round-trips;
+- inputs are right-padded WITHOUT an attention mask: with causal attention,
+  valid positions never attend to the padded tail, so scores at positions
+  < end are exact (the reference passes a mask but reads the same positions).
+"""
+
+import json
+import os
+from typing import Optional, Union
+
+import torch
+from torch import nn
+
+from trlx_amd.models.modeling_base import load_hf_dir
+from trlx_amd.models.nn.config import TransformerConfig, preset
+from trlx_amd.models.nn.transformer import CausalTransformer
+
+
+class RewardModel(nn.Module):
+    def __init__(self, trunk: CausalTransformer, pad_token_id: int):
+        super().__init__()
+        self.transformer = trunk
+        self.config = trunk.config
+        self.pad_id = pad_token_id
+        self.v_head = nn.Linear(self.config.hidden_size, 1, bias=False)
+
+    @classmethod
+    def from_pretrained(cls, path_or_config: Union[str, TransformerConfig],
+                        pad_token_id: int) -> "RewardModel":
+        """Build from a local HF dir (the SFT checkpoint), a preset name, or a
+        TransformerConfig (random init, for tests)."""
+        if isinstance(path_or_config, TransformerConfig):
+            return cls(CausalTransformer(path_or_config), pad_token_id)
+        if os.path.isdir(path_or_config):
+            cfg, sd = load_hf_dir(path_or_config)
+            trunk = CausalTransformer(cfg)
+            trunk.load_state_dict(sd, strict=False)
+            return cls(trunk, pad_token_id)
+        return cls(CausalTransformer(preset(path_or_config)), pad_token_id)
+
+    def _rewards(self, input_ids: torch.Tensor) -> torch.Tensor:
+        out = self.transformer(input_ids, return_logits=False)
+        return self.v_head(out.last_hidden_state).squeeze(-1).float()  # [B, T]
+
+    def _end_index(self, ids: torch.Tensor) -> torch.Tensor:
+        """First pad position per row, or T if un-padded (reference
+        reward_model.py:70-73, vectorized)."""
+        T = ids.shape[1]
+        is_pad = ids == self.pad_id
+        first_pad = is_pad.float().argmax(dim=1)
+        return torch.where(is_pad.any(dim=1), first_pad, torch.full_like(first_pad, T))
+
+    def forward(self, input_ids: torch.Tensor, attention_mask=None):
+        """input_ids = [chosen; rejected] stacked along batch (right-padded).
+
+        Returns {loss, chosen_end_scores, rejected_end_scores}; with identical
+        halves (inference) only chosen_end_scores (reference
+        reward_model.py:92-104).
+        """
+        rewards = self._rewards(input_ids)
+        bs = input_ids.shape[0] // 2
+        chosen, rejected = input_ids[:bs], input_ids[bs:]
+        c_r, r_r = rewards[:bs], rewards[bs:]
+
+        c_end = self._end_index(chosen)
+        r_end = self._end_index(rejected)
+        chosen_end_scores = c_r.gather(1, (c_end - 1).clamp(min=0).unsqueeze(1)).squeeze(1)
+        rejected_end_scores = r_r.gather(1, (r_end - 1).clamp(min=0).unsqueeze(1)).squeeze(1)
+
+        neq = chosen != rejected
+        if not bool(neq.any()):
+            return {"chosen_end_scores": chosen_end_scores}
+
+        # divergence..end mask per row (rows with identical halves contribute 0)
+        T = input_ids.shape[1]
+        div = neq.float().argmax(dim=1)
+        end = torch.maximum(c_end, r_end)
+        pos = torch.arange(T, device=input_ids.device).unsqueeze(0)
+        mask = (pos >= div.unsqueeze(1)) & (pos < end.unsqueeze(1)) & neq.any(dim=1, keepdim=True)
+        per_pos = -torch.nn.functional.logsigmoid(c_r - r_r) * mask
+        loss = (per_pos.sum(dim=1) / mask.sum(dim=1).clamp(min=1)).mean()
+        return {
+            "loss": loss,
+            "chosen_end_scores": chosen_end_scores,
+            "rejected_end_scores": rejected_end_scores,
+        }
+
+    @torch.no_grad()
+    def score(self, input_ids: torch.Tensor) -> torch.Tensor:
+        """End-of-sequence scalar score per row (inference path)."""
+        rewards = self._rewards(input_ids)
+        end = self._end_index(input_ids)
+        return rewards.gather(1, (end - 1).clamp(min=0).unsqueeze(1)).squeeze(1)
+
+    # --- persistence -------------------------------------------------------
+
+    def save_checkpoint(self, directory: str):
+        os.makedirs(directory, exist_ok=True)
+        torch.save(self.state_dict(), os.path.join(directory, "rm_model.pt"))
+        with open(os.path.join(directory, "rm_config.json"), "w") as f:
+            json.dump({"config": self.config.to_dict(), "pad_token_id": self.pad_id}, f)
+
+    @classmethod
+    def load_checkpoint(cls, directory: str, device: Optional[torch.device] = None):
+        with open(os.path.join(directory, "rm_config.json")) as f:
+            meta = json.load(f)
+        model = cls(CausalTransformer(TransformerConfig.from_dict(meta["config"])),
+                    meta["pad_token_id"])
+        sd = torch.load(os.path.join(directory, "rm_model.pt"), map_location="cpu",
+                        weights_only=True)
+        model.load_state_dict(sd)
+        if device is not None:
+            model.to(device)
+        return model

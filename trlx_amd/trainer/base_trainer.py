@@ -337,9 +337,11 @@ class NativeRLTrainer(BaseRLTrainer):
                         samples=str_samples, prompts=str_prompts, outputs=str_outputs,
                         tokenizer=self.tokenizer, **metadata,
                     )
-                    if rewards and isinstance(rewards[0], torch.Tensor):
+                    if isinstance(rewards, torch.Tensor):
+                        rewards = rewards.detach().float().cpu().view(len(str_samples), -1).sum(-1)
+                    elif len(rewards) and isinstance(rewards[0], torch.Tensor):
                         rewards = torch.tensor([r.sum().item() for r in rewards], dtype=torch.float)
-                    elif rewards and isinstance(rewards[0], list):
+                    elif len(rewards) and isinstance(rewards[0], list):
                         rewards = torch.tensor([sum(r) for r in rewards], dtype=torch.float)
                     else:
                         rewards = torch.tensor(rewards, dtype=torch.float)
