@@ -231,6 +231,54 @@ def test_gpu_layernorm(dtype, has_bias):
 
 
 @pytest.mark.gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("kind", ["rmsnorm", "layernorm"])
+def test_gpu_norm_add_fused(dtype, kind):
+    """Fused residual+norm (y, s) vs plain add + norm, fwd and bwd with a
+    downstream use of s (the residual-stream continuation)."""
+    torch.manual_seed(4)
+    H = 768
+    x = torch.randn(41, H).to(dtype).cuda().requires_grad_(True)
+    r = torch.randn(41, H).to(dtype).cuda().requires_grad_(True)
+    w = (torch.randn(H) * 0.1 + 1).to(dtype).cuda().requires_grad_(True)
+    b = (torch.randn(H) * 0.1).to(dtype).cuda().requires_grad_(True)
+    if kind == "rmsnorm":
+        y, s = ops.rmsnorm_add(x, r, w)
+    else:
+        y, s = ops.layernorm_add(x, r, w, b)
+    xr = x.detach().float().cpu().requires_grad_(True)
+    rr = r.detach().float().cpu().requires_grad_(True)
+    wr = w.detach().float().cpu().requires_grad_(True)
+    br = b.detach().float().cpu().requires_grad_(True)
+    sr = xr + rr
+    yr = reference.rmsnorm(sr, wr) if kind == "rmsnorm" else reference.layernorm(sr, wr, br)
+    fatol = 3e-2 if dtype == torch.bfloat16 else 1e-5
+    _assert_close(y.cpu(), yr, atol=fatol, name=f"{kind}_add fwd y")
+    _assert_close(s.cpu(), sr, atol=fatol, name=f"{kind}_add fwd s")
+    gy = torch.randn_like(y)
+    gs = torch.randn_like(s)
+    (y * gy + s * gs).sum().backward()
+    (yr * gy.float().cpu() + sr * gs.float().cpu()).sum().backward()
+    atol = 8e-2 if dtype == torch.bfloat16 else 1e-3
+    _assert_close(x.grad.cpu(), xr.grad, atol=atol, name=f"{kind}_add dx")
+    _assert_close(r.grad.cpu(), rr.grad, atol=atol, name=f"{kind}_add dres")
+    _assert_close(w.grad.cpu(), wr.grad, atol=max(atol, 0.3 if dtype == torch.bfloat16 else 1e-2),
+                  name=f"{kind}_add dw")
+
+
+def test_cpu_norm_add_matches_unfused():
+    torch.manual_seed(5)
+    x, r = torch.randn(7, 64), torch.randn(7, 64)
+    w = torch.randn(64) * 0.1 + 1
+    y, s = ops.rmsnorm_add(x, r, w)
+    assert torch.allclose(s, x + r)
+    assert torch.allclose(y, ops.rmsnorm(x + r, w))
+    b = torch.randn(64)
+    y2, s2 = ops.layernorm_add(x, r, w, b)
+    assert torch.allclose(y2, ops.layernorm(x + r, w, b))
+
+
+@pytest.mark.gpu
 @pytest.mark.parametrize("interleaved", [False, True])
 @pytest.mark.parametrize("rot_frac", [1.0, 0.25])
 def test_gpu_rope(interleaved, rot_frac):

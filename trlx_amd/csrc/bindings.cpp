@@ -4,14 +4,18 @@
 std::vector<at::Tensor> logprobs_fwd(const at::Tensor& logits, const at::Tensor& labels);
 at::Tensor logprobs_bwd(const at::Tensor& logits, const at::Tensor& labels, const at::Tensor& lse,
                         const at::Tensor& gout);
-std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps);
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps,
+                                    const c10::optional<at::Tensor>& res);
 std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
-                                    const at::Tensor& invr, const at::Tensor& dy);
+                                    const at::Tensor& invr, const at::Tensor& dy,
+                                    const c10::optional<at::Tensor>& dres);
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& w,
-                                      const c10::optional<at::Tensor>& b, double eps);
+                                      const c10::optional<at::Tensor>& b, double eps,
+                                      const c10::optional<at::Tensor>& res);
 std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x, const at::Tensor& w,
                                       const at::Tensor& mean, const at::Tensor& invstd,
-                                      const at::Tensor& dy);
+                                      const at::Tensor& dy,
+                                      const c10::optional<at::Tensor>& dres);
 at::Tensor rope_fwd(const at::Tensor& x, const at::Tensor& cos, const at::Tensor& sin,
                     const at::Tensor& pos, bool interleaved, bool inverse, long rot);
 std::vector<at::Tensor> gae(const at::Tensor& values, const at::Tensor& rewards, double gamma,

@@ -15,48 +15,70 @@ namespace {
 constexpr int BLOCK = 256;
 constexpr int NWAVES = BLOCK / WAVE;
 
-template <typename T>
-__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
-                                   T* __restrict__ y, float* __restrict__ invr, int H, float eps,
-                                   long N) {
+// HAS_RES: s = x + res is written to sout and normalized (residual-add
+// fused into the norm — saves the separate elementwise-add kernel and one
+// full stream read per residual connection; profile r01: adds were 5.6% of
+// cycle kernels).
+template <typename T, bool HAS_RES>
+__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                                   const T* __restrict__ w, T* __restrict__ y,
+                                   T* __restrict__ sout, float* __restrict__ invr, int H,
+                                   float eps, long N) {
   __shared__ float rbuf[NWAVES];
   const int H8 = H & ~7;
   for (long row = blockIdx.x; row < N; row += gridDim.x) {
     const T* xr = x + (size_t)row * H;
+    const T* rr = HAS_RES ? res + (size_t)row * H : nullptr;
+    T* sr = HAS_RES ? sout + (size_t)row * H : nullptr;
     T* yr = y + (size_t)row * H;
     float ss = 0.f;
     for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
       float v[8];
       load8<T>(xr + base, v);
+      if (HAS_RES) {
+        float rv[8];
+        load8<T>(rr + base, rv);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] += rv[i];
+        store8<T>(sr + base, v);
+      }
 #pragma unroll
       for (int i = 0; i < 8; ++i) ss += v[i] * v[i];
     }
     for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
       float xi = ScalarIO<T>::load(xr + i);
+      if (HAS_RES) {
+        xi += ScalarIO<T>::load(rr + i);
+        ScalarIO<T>::store(sr + i, xi);
+      }
       ss += xi * xi;
     }
     ss = block_sum<NWAVES>(ss, rbuf);
     const float r = rsqrtf(ss / H + eps);
     if (threadIdx.x == 0) invr[row] = r;
+    const T* src = HAS_RES ? sr : xr;
     for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
       float v[8], wv[8];
-      load8<T>(xr + base, v);
+      load8<T>(src + base, v);
       load8<T>(w + base, wv);
 #pragma unroll
       for (int i = 0; i < 8; ++i) v[i] = v[i] * r * wv[i];
       store8<T>(yr + base, v);
     }
     for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
-      ScalarIO<T>::store(yr + i, ScalarIO<T>::load(xr + i) * r * ScalarIO<T>::load(w + i));
+      ScalarIO<T>::store(yr + i, ScalarIO<T>::load(src + i) * r * ScalarIO<T>::load(w + i));
     }
     __syncthreads();
   }
 }
 
-template <typename T>
+// HAS_DRES: dres (the residual-stream gradient from downstream) is added
+// into dx — the backward half of the fused residual+norm.
+template <typename T, bool HAS_DRES>
 __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                                    const float* __restrict__ invr, const T* __restrict__ dy,
-                                   T* __restrict__ dx, float* __restrict__ dw, int H, long N) {
+                                   const T* __restrict__ dres, T* __restrict__ dx,
+                                   float* __restrict__ dw, int H, long N) {
   extern __shared__ float dwacc[];  // H floats
   __shared__ float rbuf[NWAVES];
   for (int i = threadIdx.x; i < H; i += BLOCK) dwacc[i] = 0.f;
@@ -65,6 +87,7 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
   for (long row = blockIdx.x; row < N; row += gridDim.x) {
     const T* xr = x + (size_t)row * H;
     const T* dyr = dy + (size_t)row * H;
+    const T* drr = HAS_DRES ? dres + (size_t)row * H : nullptr;
     T* dxr = dx + (size_t)row * H;
     const float r = invr[row];
     float c = 0.f;
@@ -85,9 +108,12 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
       load8<T>(xr + base, xv);
       load8<T>(dyr + base, dv);
       load8<T>(w + base, wv);
+      float ev[8];
+      if (HAS_DRES) load8<T>(drr + base, ev);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         o[i] = r * wv[i] * dv[i] - k * xv[i];
+        if (HAS_DRES) o[i] += ev[i];
         dwacc[base + i] += dv[i] * xv[i] * r;
       }
       store8<T>(dxr + base, o);
@@ -95,7 +121,9 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
     for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
       float xv = ScalarIO<T>::load(xr + i);
       float dv = ScalarIO<T>::load(dyr + i);
-      ScalarIO<T>::store(dxr + i, r * ScalarIO<T>::load(w + i) * dv - k * xv);
+      float o = r * ScalarIO<T>::load(w + i) * dv - k * xv;
+      if (HAS_DRES) o += ScalarIO<T>::load(drr + i);
+      ScalarIO<T>::store(dxr + i, o);
       dwacc[i] += dv * xv * r;
     }
     __syncthreads();
@@ -103,20 +131,30 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
   for (int i = threadIdx.x; i < H; i += BLOCK) atomicAdd(&dw[i], dwacc[i]);
 }
 
-template <typename T, bool HAS_BIAS>
-__global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
-                                     const T* __restrict__ b, T* __restrict__ y,
+template <typename T, bool HAS_BIAS, bool HAS_RES>
+__global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
+                                     const T* __restrict__ w, const T* __restrict__ b,
+                                     T* __restrict__ y, T* __restrict__ sout,
                                      float* __restrict__ mean, float* __restrict__ invstd, int H,
                                      float eps, long N) {
   __shared__ float rbuf[NWAVES];
   const int H8 = H & ~7;
   for (long row = blockIdx.x; row < N; row += gridDim.x) {
     const T* xr = x + (size_t)row * H;
+    const T* rr = HAS_RES ? res + (size_t)row * H : nullptr;
+    T* sr = HAS_RES ? sout + (size_t)row * H : nullptr;
     T* yr = y + (size_t)row * H;
     float s = 0.f, ss = 0.f;
     for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
       float v[8];
       load8<T>(xr + base, v);
+      if (HAS_RES) {
+        float rv[8];
+        load8<T>(rr + base, rv);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] += rv[i];
+        store8<T>(sr + base, v);
+      }
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         s += v[i];
@@ -125,6 +163,10 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restric
     }
     for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
       float xi = ScalarIO<T>::load(xr + i);
+      if (HAS_RES) {
+        xi += ScalarIO<T>::load(rr + i);
+        ScalarIO<T>::store(sr + i, xi);
+      }
       s += xi;
       ss += xi * xi;
     }
@@ -137,9 +179,10 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restric
       mean[row] = mu;
       invstd[row] = istd;
     }
+    const T* src = HAS_RES ? sr : xr;
     for (int base = threadIdx.x * 8; base < H8; base += BLOCK * 8) {
       float v[8], wv[8], bv[8];
-      load8<T>(xr + base, v);
+      load8<T>(src + base, v);
       load8<T>(w + base, wv);
       if (HAS_BIAS) load8<T>(b + base, bv);
 #pragma unroll
@@ -151,7 +194,7 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restric
       store8<T>(yr + base, v);
     }
     for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
-      float o = (ScalarIO<T>::load(xr + i) - mu) * istd * ScalarIO<T>::load(w + i);
+      float o = (ScalarIO<T>::load(src + i) - mu) * istd * ScalarIO<T>::load(w + i);
       if (HAS_BIAS) o += ScalarIO<T>::load(b + i);
       ScalarIO<T>::store(yr + i, o);
     }
@@ -159,12 +202,13 @@ __global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restric
   }
 }
 
-template <typename T>
+template <typename T, bool HAS_DRES>
 __global__ void layernorm_bwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd, const T* __restrict__ dy,
-                                     T* __restrict__ dx, float* __restrict__ dw,
-                                     float* __restrict__ db, int H, long N) {
+                                     const T* __restrict__ dres, T* __restrict__ dx,
+                                     float* __restrict__ dw, float* __restrict__ db, int H,
+                                     long N) {
   extern __shared__ float acc[];  // 2*H floats: [dw | db]
   __shared__ float rbuf[NWAVES];
   float* dwacc = acc;
@@ -178,6 +222,7 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ x, const T* __restric
   for (long row = blockIdx.x; row < N; row += gridDim.x) {
     const T* xr = x + (size_t)row * H;
     const T* dyr = dy + (size_t)row * H;
+    const T* drr = HAS_DRES ? dres + (size_t)row * H : nullptr;
     T* dxr = dx + (size_t)row * H;
     const float mu = mean[row];
     const float istd = invstd[row];
@@ -206,10 +251,13 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ x, const T* __restric
       load8<T>(xr + base, xv);
       load8<T>(dyr + base, dv);
       load8<T>(w + base, wv);
+      float ev[8];
+      if (HAS_DRES) load8<T>(drr + base, ev);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         float xhat = (xv[i] - mu) * istd;
         o[i] = istd * (dv[i] * wv[i] - s1 - xhat * s2);
+        if (HAS_DRES) o[i] += ev[i];
         dwacc[base + i] += dv[i] * xhat;
         dbacc[base + i] += dv[i];
       }
@@ -218,7 +266,9 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ x, const T* __restric
     for (int i = H8 + threadIdx.x; i < H; i += BLOCK) {
       float dv = ScalarIO<T>::load(dyr + i);
       float xhat = (ScalarIO<T>::load(xr + i) - mu) * istd;
-      ScalarIO<T>::store(dxr + i, istd * (dv * ScalarIO<T>::load(w + i) - s1 - xhat * s2));
+      float o = istd * (dv * ScalarIO<T>::load(w + i) - s1 - xhat * s2);
+      if (HAS_DRES) o += ScalarIO<T>::load(drr + i);
+      ScalarIO<T>::store(dxr + i, o);
       dwacc[i] += dv * xhat;
       dbacc[i] += dv;
     }
@@ -234,52 +284,90 @@ int pick_grid(long n) { return (int)std::min<long>(n, 2048); }
 
 }  // namespace
 
-std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps) {
+std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, double eps,
+                                    const c10::optional<at::Tensor>& res) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
   TORCH_CHECK(w.dtype() == x.dtype());
   const long N = x.size(0);
   const int H = x.size(1);
+  const bool has_res = res.has_value();
   auto y = at::empty_like(x);
   auto invr = at::empty({N}, x.options().dtype(at::kFloat));
-  if (N == 0) return {y, invr};
+  at::Tensor sout, rc;
+  if (has_res) {
+    rc = res->contiguous();
+    TORCH_CHECK(rc.sizes() == x.sizes() && rc.dtype() == x.dtype());
+    sout = at::empty_like(x);
+  }
+  if (N == 0) return has_res ? std::vector<at::Tensor>{y, invr, sout}
+                             : std::vector<at::Tensor>{y, invr};
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
   const int grid = pick_grid(N);
   if (x.dtype() == at::kBFloat16) {
-    rmsnorm_fwd_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(
-        reinterpret_cast<const bf16_t*>(x.data_ptr()), reinterpret_cast<const bf16_t*>(wc.data_ptr()),
-        reinterpret_cast<bf16_t*>(y.data_ptr()), invr.data_ptr<float>(), H, (float)eps, N);
+    auto xp = reinterpret_cast<const bf16_t*>(x.data_ptr());
+    auto wp = reinterpret_cast<const bf16_t*>(wc.data_ptr());
+    auto yp = reinterpret_cast<bf16_t*>(y.data_ptr());
+    if (has_res)
+      rmsnorm_fwd_kernel<bf16_t, true><<<grid, BLOCK, 0, stream>>>(
+          xp, reinterpret_cast<const bf16_t*>(rc.data_ptr()), wp, yp,
+          reinterpret_cast<bf16_t*>(sout.data_ptr()), invr.data_ptr<float>(), H, (float)eps, N);
+    else
+      rmsnorm_fwd_kernel<bf16_t, false><<<grid, BLOCK, 0, stream>>>(
+          xp, nullptr, wp, yp, nullptr, invr.data_ptr<float>(), H, (float)eps, N);
   } else {
-    rmsnorm_fwd_kernel<float><<<grid, BLOCK, 0, stream>>>(x.data_ptr<float>(), wc.data_ptr<float>(),
-                                                          y.data_ptr<float>(),
-                                                          invr.data_ptr<float>(), H, (float)eps, N);
+    if (has_res)
+      rmsnorm_fwd_kernel<float, true><<<grid, BLOCK, 0, stream>>>(
+          x.data_ptr<float>(), rc.data_ptr<float>(), wc.data_ptr<float>(), y.data_ptr<float>(),
+          sout.data_ptr<float>(), invr.data_ptr<float>(), H, (float)eps, N);
+    else
+      rmsnorm_fwd_kernel<float, false><<<grid, BLOCK, 0, stream>>>(
+          x.data_ptr<float>(), nullptr, wc.data_ptr<float>(), y.data_ptr<float>(), nullptr,
+          invr.data_ptr<float>(), H, (float)eps, N);
   }
   HIP_CHECK_LAST();
-  return {y, invr};
+  return has_res ? std::vector<at::Tensor>{y, invr, sout} : std::vector<at::Tensor>{y, invr};
 }
 
 std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
-                                    const at::Tensor& invr, const at::Tensor& dy) {
+                                    const at::Tensor& invr, const at::Tensor& dy,
+                                    const c10::optional<at::Tensor>& dres) {
   const long N = x.size(0);
   const int H = x.size(1);
   TORCH_CHECK(H <= 16384, "rmsnorm_bwd: H too large for LDS accumulation");
+  const bool has_dres = dres.has_value();
+  at::Tensor drc;
+  if (has_dres) drc = dres->contiguous();
   auto dx = at::empty_like(x);
   auto dwf = at::zeros({H}, x.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
+  auto dyc = dy.contiguous();
   const int grid = pick_grid(N);
   const size_t lds = (size_t)H * sizeof(float);
   if (N > 0) {
     if (x.dtype() == at::kBFloat16) {
-      rmsnorm_bwd_kernel<bf16_t><<<grid, BLOCK, lds, stream>>>(
-          reinterpret_cast<const bf16_t*>(x.data_ptr()),
-          reinterpret_cast<const bf16_t*>(wc.data_ptr()), invr.data_ptr<float>(),
-          reinterpret_cast<const bf16_t*>(dy.data_ptr()), reinterpret_cast<bf16_t*>(dx.data_ptr()),
-          dwf.data_ptr<float>(), H, N);
+      auto xp = reinterpret_cast<const bf16_t*>(x.data_ptr());
+      auto wp = reinterpret_cast<const bf16_t*>(wc.data_ptr());
+      auto dyp = reinterpret_cast<const bf16_t*>(dyc.data_ptr());
+      auto dxp = reinterpret_cast<bf16_t*>(dx.data_ptr());
+      if (has_dres)
+        rmsnorm_bwd_kernel<bf16_t, true><<<grid, BLOCK, lds, stream>>>(
+            xp, wp, invr.data_ptr<float>(), dyp,
+            reinterpret_cast<const bf16_t*>(drc.data_ptr()), dxp, dwf.data_ptr<float>(), H, N);
+      else
+        rmsnorm_bwd_kernel<bf16_t, false><<<grid, BLOCK, lds, stream>>>(
+            xp, wp, invr.data_ptr<float>(), dyp, nullptr, dxp, dwf.data_ptr<float>(), H, N);
     } else {
-      rmsnorm_bwd_kernel<float><<<grid, BLOCK, lds, stream>>>(
-          x.data_ptr<float>(), wc.data_ptr<float>(), invr.data_ptr<float>(), dy.data_ptr<float>(),
-          dx.data_ptr<float>(), dwf.data_ptr<float>(), H, N);
+      if (has_dres)
+        rmsnorm_bwd_kernel<float, true><<<grid, BLOCK, lds, stream>>>(
+            x.data_ptr<float>(), wc.data_ptr<float>(), invr.data_ptr<float>(),
+            dyc.data_ptr<float>(), drc.data_ptr<float>(), dx.data_ptr<float>(),
+            dwf.data_ptr<float>(), H, N);
+      else
+        rmsnorm_bwd_kernel<float, false><<<grid, BLOCK, lds, stream>>>(
+            x.data_ptr<float>(), wc.data_ptr<float>(), invr.data_ptr<float>(),
+            dyc.data_ptr<float>(), nullptr, dx.data_ptr<float>(), dwf.data_ptr<float>(), H, N);
     }
     HIP_CHECK_LAST();
   }
@@ -287,71 +375,106 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
 }
 
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& w,
-                                      const c10::optional<at::Tensor>& b, double eps) {
+                                      const c10::optional<at::Tensor>& b, double eps,
+                                      const c10::optional<at::Tensor>& res) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
   const long N = x.size(0);
   const int H = x.size(1);
+  const bool has_res = res.has_value();
   auto y = at::empty_like(x);
   auto mean = at::empty({N}, x.options().dtype(at::kFloat));
   auto invstd = at::empty({N}, x.options().dtype(at::kFloat));
-  if (N == 0) return {y, mean, invstd};
+  at::Tensor sout, rc;
+  if (has_res) {
+    rc = res->contiguous();
+    TORCH_CHECK(rc.sizes() == x.sizes() && rc.dtype() == x.dtype());
+    sout = at::empty_like(x);
+  }
+  if (N == 0) return has_res ? std::vector<at::Tensor>{y, mean, invstd, sout}
+                             : std::vector<at::Tensor>{y, mean, invstd};
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
   const bool has_bias = b.has_value();
   at::Tensor bc;
   if (has_bias) bc = b->contiguous();
   const int grid = pick_grid(N);
-#define LAUNCH_LN_FWD(T, HB, XP, WP, BP, YP)                                                   \
-  layernorm_fwd_kernel<T, HB><<<grid, BLOCK, 0, stream>>>(XP, WP, BP, YP, mean.data_ptr<float>(), \
-                                                          invstd.data_ptr<float>(), H, (float)eps, N)
+#define LAUNCH_LN_FWD(T, HB, HR, XP, RP, WP, BP, YP, SP)                                    \
+  layernorm_fwd_kernel<T, HB, HR><<<grid, BLOCK, 0, stream>>>(                              \
+      XP, RP, WP, BP, YP, SP, mean.data_ptr<float>(), invstd.data_ptr<float>(), H,          \
+      (float)eps, N)
   if (x.dtype() == at::kBFloat16) {
     auto xp = reinterpret_cast<const bf16_t*>(x.data_ptr());
     auto wp = reinterpret_cast<const bf16_t*>(wc.data_ptr());
     auto yp = reinterpret_cast<bf16_t*>(y.data_ptr());
-    if (has_bias)
-      LAUNCH_LN_FWD(bf16_t, true, xp, wp, reinterpret_cast<const bf16_t*>(bc.data_ptr()), yp);
-    else
-      LAUNCH_LN_FWD(bf16_t, false, xp, wp, nullptr, yp);
+    auto bp = has_bias ? reinterpret_cast<const bf16_t*>(bc.data_ptr()) : nullptr;
+    auto rp = has_res ? reinterpret_cast<const bf16_t*>(rc.data_ptr()) : nullptr;
+    auto sp = has_res ? reinterpret_cast<bf16_t*>(sout.data_ptr()) : nullptr;
+    if (has_bias && has_res) LAUNCH_LN_FWD(bf16_t, true, true, xp, rp, wp, bp, yp, sp);
+    else if (has_bias) LAUNCH_LN_FWD(bf16_t, true, false, xp, nullptr, wp, bp, yp, nullptr);
+    else if (has_res) LAUNCH_LN_FWD(bf16_t, false, true, xp, rp, wp, nullptr, yp, sp);
+    else LAUNCH_LN_FWD(bf16_t, false, false, xp, nullptr, wp, nullptr, yp, nullptr);
   } else {
     auto xp = x.data_ptr<float>();
     auto wp = wc.data_ptr<float>();
     auto yp = y.data_ptr<float>();
-    if (has_bias)
-      LAUNCH_LN_FWD(float, true, xp, wp, bc.data_ptr<float>(), yp);
-    else
-      LAUNCH_LN_FWD(float, false, xp, wp, nullptr, yp);
+    auto bp = has_bias ? bc.data_ptr<float>() : nullptr;
+    auto rp = has_res ? rc.data_ptr<float>() : nullptr;
+    auto sp = has_res ? sout.data_ptr<float>() : nullptr;
+    if (has_bias && has_res) LAUNCH_LN_FWD(float, true, true, xp, rp, wp, bp, yp, sp);
+    else if (has_bias) LAUNCH_LN_FWD(float, true, false, xp, nullptr, wp, bp, yp, nullptr);
+    else if (has_res) LAUNCH_LN_FWD(float, false, true, xp, rp, wp, nullptr, yp, sp);
+    else LAUNCH_LN_FWD(float, false, false, xp, nullptr, wp, nullptr, yp, nullptr);
   }
 #undef LAUNCH_LN_FWD
   HIP_CHECK_LAST();
-  return {y, mean, invstd};
+  return has_res ? std::vector<at::Tensor>{y, mean, invstd, sout}
+                 : std::vector<at::Tensor>{y, mean, invstd};
 }
 
 std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x, const at::Tensor& w,
                                       const at::Tensor& mean, const at::Tensor& invstd,
-                                      const at::Tensor& dy) {
+                                      const at::Tensor& dy,
+                                      const c10::optional<at::Tensor>& dres) {
   const long N = x.size(0);
   const int H = x.size(1);
   TORCH_CHECK(H <= 16384, "layernorm_bwd: H too large for LDS accumulation");
+  const bool has_dres = dres.has_value();
+  at::Tensor drc;
+  if (has_dres) drc = dres->contiguous();
   auto dx = at::empty_like(x);
   auto dwf = at::zeros({H}, x.options().dtype(at::kFloat));
   auto dbf = at::zeros({H}, x.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
+  auto dyc = dy.contiguous();
   const int grid = pick_grid(N);
   const size_t lds = 2 * (size_t)H * sizeof(float);
   if (N > 0) {
     if (x.dtype() == at::kBFloat16) {
-      layernorm_bwd_kernel<bf16_t><<<grid, BLOCK, lds, stream>>>(
-          reinterpret_cast<const bf16_t*>(x.data_ptr()),
-          reinterpret_cast<const bf16_t*>(wc.data_ptr()), mean.data_ptr<float>(),
-          invstd.data_ptr<float>(), reinterpret_cast<const bf16_t*>(dy.data_ptr()),
-          reinterpret_cast<bf16_t*>(dx.data_ptr()), dwf.data_ptr<float>(), dbf.data_ptr<float>(),
-          H, N);
+      auto xp = reinterpret_cast<const bf16_t*>(x.data_ptr());
+      auto wp = reinterpret_cast<const bf16_t*>(wc.data_ptr());
+      auto dyp = reinterpret_cast<const bf16_t*>(dyc.data_ptr());
+      auto dxp = reinterpret_cast<bf16_t*>(dx.data_ptr());
+      if (has_dres)
+        layernorm_bwd_kernel<bf16_t, true><<<grid, BLOCK, lds, stream>>>(
+            xp, wp, mean.data_ptr<float>(), invstd.data_ptr<float>(), dyp,
+            reinterpret_cast<const bf16_t*>(drc.data_ptr()), dxp, dwf.data_ptr<float>(),
+            dbf.data_ptr<float>(), H, N);
+      else
+        layernorm_bwd_kernel<bf16_t, false><<<grid, BLOCK, lds, stream>>>(
+            xp, wp, mean.data_ptr<float>(), invstd.data_ptr<float>(), dyp, nullptr, dxp,
+            dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
     } else {
-      layernorm_bwd_kernel<float><<<grid, BLOCK, lds, stream>>>(
-          x.data_ptr<float>(), wc.data_ptr<float>(), mean.data_ptr<float>(),
-          invstd.data_ptr<float>(), dy.data_ptr<float>(), dx.data_ptr<float>(),
-          dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
+      if (has_dres)
+        layernorm_bwd_kernel<float, true><<<grid, BLOCK, lds, stream>>>(
+            x.data_ptr<float>(), wc.data_ptr<float>(), mean.data_ptr<float>(),
+            invstd.data_ptr<float>(), dyc.data_ptr<float>(), drc.data_ptr<float>(),
+            dx.data_ptr<float>(), dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
+      else
+        layernorm_bwd_kernel<float, false><<<grid, BLOCK, lds, stream>>>(
+            x.data_ptr<float>(), wc.data_ptr<float>(), mean.data_ptr<float>(),
+            invstd.data_ptr<float>(), dyc.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
+            dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
     }
     HIP_CHECK_LAST();
   }

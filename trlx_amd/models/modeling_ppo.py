@@ -266,12 +266,16 @@ class FrozenBranch(nn.Module):
         """Final-norm hidden states of the frozen branch (pre-lm_head) — the
         fused lm_logprobs path consumes these directly."""
         with torch.no_grad():
-            h = hidden
+            h, res = hidden, None
             for block in self.blocks:
-                h = block(h, ctx, rope_tables)
+                h, res = block(h, ctx, rope_tables, res=res)
             if logits_slice is not None:
                 h = h[:, logits_slice[0] : logits_slice[1]]
-            return self.final_norm(h)
+                if res is not None:
+                    res = res[:, logits_slice[0] : logits_slice[1]]
+            if res is None:
+                return self.final_norm(h)
+            return self.final_norm.forward_add(h, res)[0]
 
 
 class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):

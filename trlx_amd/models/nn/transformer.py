@@ -49,6 +49,14 @@ class Norm(nn.Module):
             return ops.rmsnorm(x, self.weight, self.eps)
         return ops.layernorm(x, self.weight, self.bias, self.eps)
 
+    def forward_add(self, x, res):
+        """Fused residual + norm: returns (norm(x + res), x + res) in one
+        kernel (the elementwise residual adds were 5.6% of cycle kernel time,
+        profile r01)."""
+        if self.kind == "rmsnorm":
+            return ops.rmsnorm_add(x, res, self.weight, self.eps)
+        return ops.layernorm_add(x, res, self.weight, self.bias, self.eps)
+
 
 def alibi_slopes(num_heads: int) -> torch.Tensor:
     """Standard ALiBi head slopes (Bloom; closest-power-of-two scheme)."""
@@ -261,14 +269,30 @@ class Block(nn.Module):
             return F.dropout(x, self.resid_pdrop)
         return x
 
-    def forward(self, x, ctx, rope_tables, kv_cache=None):
+    def forward(self, x, ctx, rope_tables, kv_cache=None, res=None):
+        """Deferred-residual protocol: the true stream at block input is
+        ``x + res`` (res=None means x IS the stream).  The pending add is
+        fused into this block's first norm; the block returns its own
+        pending pair (branch_out, stream) for the next block.
+        """
         if self.cfg.parallel_residual:
+            if res is None:
+                s = x
+                h1 = self.ln_1(x)
+            else:
+                h1, s = self.ln_1.forward_add(x, res)
+            h2 = h1 if self.shared_parallel_norm else self.ln_2(s)
+            out = self._drop(self.attn(h1, ctx, rope_tables, kv_cache)) + self._drop(self.mlp(h2))
+            return out, s
+        if res is None:
+            s = x
             h1 = self.ln_1(x)
-            h2 = h1 if self.shared_parallel_norm else self.ln_2(x)
-            return x + self._drop(self.attn(h1, ctx, rope_tables, kv_cache)) + self._drop(self.mlp(h2))
-        x = x + self._drop(self.attn(self.ln_1(x), ctx, rope_tables, kv_cache))
-        x = x + self._drop(self.mlp(self.ln_2(x)))
-        return x
+        else:
+            h1, s = self.ln_1.forward_add(x, res)
+        a = self._drop(self.attn(h1, ctx, rope_tables, kv_cache))
+        h2, s2 = self.ln_2.forward_add(a, s)
+        m = self._drop(self.mlp(h2))
+        return m, s2
 
 
 @dataclass
@@ -380,9 +404,10 @@ class CausalTransformer(nn.Module):
 
     def run_layers(self, h, ctx, kv_cache=None, from_layer: int = 0, to_layer: Optional[int] = None):
         to_layer = len(self.layers) if to_layer is None else to_layer
+        res = None
         for i in range(from_layer, to_layer):
-            h = self.layers[i](h, ctx, self.rope_tables, kv_cache)
-        return h
+            h, res = self.layers[i](h, ctx, self.rope_tables, kv_cache, res=res)
+        return h if res is None else h + res
 
     # --- main entry --------------------------------------------------------
 
@@ -446,11 +471,17 @@ class CausalTransformer(nn.Module):
         if hidden_at_layer is not None:
             stash_at = hidden_at_layer % n
         hidden_at = None
+        res = None
         for i, layer in enumerate(self.layers):
             if stash_at is not None and i == stash_at:
-                hidden_at = h
-            h = layer(h, ctx, self.rope_tables, kv_cache)
-        h = self.final_norm(h)
+                # materialize the true stream at this layer's input (the
+                # pending residual is otherwise deferred into its norm)
+                hidden_at = h if res is None else h + res
+            h, res = layer(h, ctx, self.rope_tables, kv_cache, res=res)
+        if res is None:
+            h = self.final_norm(h)
+        else:
+            h, _ = self.final_norm.forward_add(h, res)
         if sp_active:
             # the loss region downstream is replicated on every TP rank ->
             # slice-backward gather (see _GatherFromSPReplicated)

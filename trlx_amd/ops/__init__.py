@@ -119,7 +119,7 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps, ext):
         x2d = x.reshape(-1, x.shape[-1]).contiguous()
-        out, inv_rms = ext.rmsnorm_fwd(x2d, weight, eps)
+        out, inv_rms = ext.rmsnorm_fwd(x2d, weight, eps, None)
         ctx.save_for_backward(x2d, weight, inv_rms)
         ctx.x_shape = x.shape
         return out.view(x.shape)
@@ -128,7 +128,8 @@ class _RMSNorm(torch.autograd.Function):
     def backward(ctx, grad_out):
         x2d, weight, inv_rms = ctx.saved_tensors
         ext = _require_ext("rmsnorm")
-        dx, dw = ext.rmsnorm_bwd(x2d, weight, inv_rms, grad_out.reshape(x2d.shape).contiguous())
+        dx, dw = ext.rmsnorm_bwd(x2d, weight, inv_rms,
+                                 grad_out.reshape(x2d.shape).contiguous(), None)
         return dx.view(ctx.x_shape), dw, None, None
 
 
@@ -140,11 +141,49 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6) -> torch.T
     return reference.rmsnorm(x, weight, eps)
 
 
+class _RMSNormAdd(torch.autograd.Function):
+    """Residual-add fused into RMSNorm: (y, s) = (rmsnorm(x + res), x + res).
+
+    The backward adds the residual-stream gradient ``ds`` into dx inside the
+    norm-backward kernel, so the whole residual connection costs zero extra
+    elementwise kernels.  dx and dres are the SAME tensor (aliasing is safe:
+    autograd accumulates grads out-of-place)."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, eps, ext):
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        r2d = res.reshape(x2d.shape).contiguous()
+        out, inv_rms, s = ext.rmsnorm_fwd(x2d, weight, eps, r2d)
+        ctx.save_for_backward(s, weight, inv_rms)
+        ctx.x_shape = x.shape
+        return out.view(x.shape), s.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, grad_out, grad_s):
+        s2d, weight, inv_rms = ctx.saved_tensors
+        ext = _require_ext("rmsnorm")
+        gs = grad_s.reshape(s2d.shape).contiguous() if grad_s is not None else None
+        dx, dw = ext.rmsnorm_bwd(s2d, weight, inv_rms,
+                                 grad_out.reshape(s2d.shape).contiguous(), gs)
+        dx = dx.view(ctx.x_shape)
+        return dx, dx, dw, None, None
+
+
+def rmsnorm_add(x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor, eps: float = 1e-6):
+    """Returns (rmsnorm(x + res), x + res)."""
+    if x.is_cuda:
+        ext = _require_ext("rmsnorm")
+        if ext is not None:
+            return _RMSNormAdd.apply(x, res, weight, eps, ext)
+    s = x + res
+    return reference.rmsnorm(s, weight, eps), s
+
+
 class _LayerNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps, ext):
         x2d = x.reshape(-1, x.shape[-1]).contiguous()
-        out, mean, invstd = ext.layernorm_fwd(x2d, weight, bias, eps)
+        out, mean, invstd = ext.layernorm_fwd(x2d, weight, bias, eps, None)
         ctx.save_for_backward(x2d, weight, mean, invstd)
         ctx.x_shape = x.shape
         ctx.has_bias = bias is not None
@@ -154,7 +193,8 @@ class _LayerNorm(torch.autograd.Function):
     def backward(ctx, grad_out):
         x2d, weight, mean, invstd = ctx.saved_tensors
         ext = _require_ext("layernorm")
-        dx, dw, db = ext.layernorm_bwd(x2d, weight, mean, invstd, grad_out.reshape(x2d.shape).contiguous())
+        dx, dw, db = ext.layernorm_bwd(x2d, weight, mean, invstd,
+                                       grad_out.reshape(x2d.shape).contiguous(), None)
         return dx.view(ctx.x_shape), dw, (db if ctx.has_bias else None), None, None
 
 
@@ -166,6 +206,41 @@ def layernorm(
         if ext is not None:
             return _LayerNorm.apply(x, weight, bias, eps, ext)
     return reference.layernorm(x, weight, bias, eps)
+
+
+class _LayerNormAdd(torch.autograd.Function):
+    """Residual-add fused into LayerNorm (see _RMSNormAdd)."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, eps, ext):
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        r2d = res.reshape(x2d.shape).contiguous()
+        out, mean, invstd, s = ext.layernorm_fwd(x2d, weight, bias, eps, r2d)
+        ctx.save_for_backward(s, weight, mean, invstd)
+        ctx.x_shape = x.shape
+        ctx.has_bias = bias is not None
+        return out.view(x.shape), s.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, grad_out, grad_s):
+        s2d, weight, mean, invstd = ctx.saved_tensors
+        ext = _require_ext("layernorm")
+        gs = grad_s.reshape(s2d.shape).contiguous() if grad_s is not None else None
+        dx, dw, db = ext.layernorm_bwd(s2d, weight, mean, invstd,
+                                       grad_out.reshape(s2d.shape).contiguous(), gs)
+        dx = dx.view(ctx.x_shape)
+        return dx, dx, dw, (db if ctx.has_bias else None), None, None
+
+
+def layernorm_add(x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor,
+                  bias: Optional[torch.Tensor], eps: float = 1e-5):
+    """Returns (layernorm(x + res), x + res)."""
+    if x.is_cuda:
+        ext = _require_ext("layernorm")
+        if ext is not None:
+            return _LayerNormAdd.apply(x, res, weight, bias, eps, ext)
+    s = x + res
+    return reference.layernorm(s, weight, bias, eps), s
 
 
 # --------------------------------------------------------------------------

@@ -120,8 +120,12 @@ class PipelineStage(nn.Module):
                 h = h + self.embed_positions(ctx.position_ids.long() + off)
         else:
             h = hidden_or_ids
+        res = None
         for layer in self.layers:
-            h = layer(h, ctx, self.rope_tables)
+            h, res = layer(h, ctx, self.rope_tables, res=res)
+        if res is not None:
+            # stage boundaries exchange the materialized stream
+            h = h + res
         if self.is_last:
             h = self.final_norm(h)
         return h
