@@ -47,6 +47,8 @@ at::Tensor causal_softmax_bwd(const at::Tensor& probs, const at::Tensor& dprobs)
 at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at::Tensor& vc,
                             const at::Tensor& seq_lens, double scale,
                             const c10::optional<at::Tensor>& seq_starts);
+at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
+                          const at::Tensor& labels);
 at::Tensor lm_logprobs(const at::Tensor& hidden, const at::Tensor& weight,
                        const at::Tensor& labels);
 void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
@@ -74,4 +76,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("attention_decode", &attention_decode);
   mod.def("fused_adamw", &fused_adamw);
   mod.def("lm_logprobs", &lm_logprobs);
+  mod.def("lm_logprobs_v2", &lm_logprobs_v2);
 }

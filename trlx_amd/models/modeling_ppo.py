@@ -231,7 +231,7 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         flat_labels = labels.reshape(-1)
         import os
 
-        if (os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") == "1"
+        if (os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") != "0"
                 and h.is_cuda and h.dtype == torch.bfloat16 and lm.bias is None
                 and h.shape[-1] % 32 == 0):
             logprobs = ops.lm_logprobs(h.reshape(-1, h.shape[-1]).contiguous(), lm.weight,
@@ -358,13 +358,14 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         lm = self.base_model.lm_head
 
         def fused_ok(head):
-            # measured (profiles/r01_notes.md): the current single-buffered
-            # MFMA structure loses to hipBLASLt+logprobs (1603 vs 706 us at
-            # [5248, 50257, 768]); opt-in until the kernel moves to the
-            # 8-phase pipelined schedule
+            # default ON: the pipelined lm_logprobs_v2 kernel beats the
+            # hipBLASLt GEMM + logprob-gather path 1.34-1.41x on the
+            # experience shapes ([1312-5248, 50257, 768]: 157/273/512 us vs
+            # 221/385/685 us, exact numerics).  Set
+            # TRLX_AMD_FUSED_LM_LOGPROBS=0 to fall back.
             import os
 
-            return (os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") == "1"
+            return (os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") != "0"
                     and h.is_cuda and h.dtype == torch.bfloat16 and head.bias is None
                     and h.shape[-1] % 32 == 0)
 

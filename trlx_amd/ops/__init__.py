@@ -105,7 +105,12 @@ def lm_logprobs(hidden: torch.Tensor, weight: torch.Tensor, labels: torch.Tensor
     if hidden.is_cuda:
         ext = _require_ext("lm_logprobs")
         if ext is not None and hasattr(ext, "lm_logprobs"):
-            return ext.lm_logprobs(hidden.contiguous(), weight.contiguous(), labels.contiguous())
+            h = hidden.contiguous()
+            if h.shape[1] % 64 == 0 and hasattr(ext, "lm_logprobs_v2") and \
+                    os.environ.get("TRLX_AMD_LM_LOGPROBS_V1") != "1":
+                # 8-phase pipelined 256x256 MFMA kernel
+                return ext.lm_logprobs_v2(h, weight.contiguous(), labels.contiguous())
+            return ext.lm_logprobs(h, weight.contiguous(), labels.contiguous())
     logits = hidden.float() @ weight.float().t()
     return reference.logprobs_of_labels(logits, labels)
 
