@@ -274,9 +274,11 @@ class PPOTrainer(NativeRLTrainer):
                 metadata = gather_dict(metadata_local)
 
                 if comm.is_main_process():
+                    rollout_decode_time = time()
                     all_str_samples, all_str_prompts, all_str_outputs = self.decode(
                         gathered_prompts, gathered_samples, gathered_prompt_sizes, append_eos_token=True
                     )
+                    stats["time/rollout_decode"] = time() - rollout_decode_time
                     rollout_score_time = time()
                     all_scores = self.reward_fn(
                         samples=all_str_samples, prompts=all_str_prompts, outputs=all_str_outputs,
@@ -309,6 +311,7 @@ class PPOTrainer(NativeRLTrainer):
                                                                     append_eos_token=True)
 
             # re-tokenize outputs (stop sequences may have trimmed them)
+            retok_time = time()
             outputs = self.tokenizer(str_outputs).input_ids
             if self.config.model.model_arch_type == "seq2seq":
                 # decoder sequences start with the decoder-start/pad token
@@ -322,6 +325,7 @@ class PPOTrainer(NativeRLTrainer):
                 for output in outputs
             ]
             sample_outputs = torch.vstack(outputs).to(device)
+            stats["time/rollout_retokenize"] = time() - retok_time
 
             if self.config.method.cliprange_reward:
                 scores = torch.clip(scores, -self.config.method.cliprange_reward,
@@ -342,6 +346,7 @@ class PPOTrainer(NativeRLTrainer):
             elif self.config.method.scale_reward == "ref":
                 scores /= self.ref_std
 
+            rollout_fwd_time = time()
             if self.config.model.model_arch_type == "seq2seq":
                 # seq2seq experience pass (reference accelerate_ppo_trainer.py:383-414)
                 n_samples = samples.shape[0]
@@ -428,6 +433,9 @@ class PPOTrainer(NativeRLTrainer):
                 ends_t = attention_mask[:, start:].sum(1) + 1
                 resp_logprobs, resp_values, resp_log_ratio = logprobs, values, log_ratio
 
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            stats["time/rollout_forward"] = time() - rollout_fwd_time
             # rollout tensors stay DEVICE-resident as whole padded chunk
             # batches — no CPU round trip and no per-element slicing (the
             # collate path cost ~2.6k tiny device copies per cycle,
