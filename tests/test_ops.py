@@ -480,3 +480,27 @@ def test_gpu_lm_logprobs_fused_mfma(N, V, H):
     want = reference.logprobs_of_labels(logits, labels)
     diff = (got - want).abs().max().item()
     assert diff < 0.08, diff  # bf16 GEMM vs fp32 reference
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("act", [0, 1, 2, 4])
+@pytest.mark.parametrize("shape", [(128, 2304, 768), (128, 768, 3072), (96, 300, 768)])
+def test_gpu_skinny_gemm(act, shape):
+    """Decode-path streaming GEMM vs F.linear + activation (fp32 reference)."""
+    M, N, K = shape
+    torch.manual_seed(6)
+    x = (torch.randn(M, K) * 0.5).bfloat16().cuda()
+    w = (torch.randn(N, K) * 0.02).bfloat16().cuda()
+    b = (torch.randn(N) * 0.1).bfloat16().cuda()
+    with torch.no_grad():
+        y = ops.skinny_linear(x, w, b, act)
+    ref = torch.nn.functional.linear(x.float().cpu(), w.float().cpu(), b.float().cpu())
+    ref = ops._ACT_FNS[act](ref)
+    _assert_close(y.float().cpu(), ref, atol=5e-2, name=f"skinny_gemm act={act}")
+
+
+def test_cpu_skinny_linear_fallback():
+    x, w, b = torch.randn(4, 64), torch.randn(32, 64), torch.randn(32)
+    y = ops.skinny_linear(x, w, b, act=2)
+    want = torch.nn.functional.gelu(torch.nn.functional.linear(x, w, b), approximate="tanh")
+    assert torch.allclose(y, want, atol=1e-6)

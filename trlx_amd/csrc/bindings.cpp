@@ -49,6 +49,8 @@ at::Tensor attention_decode(const at::Tensor& q, const at::Tensor& kc, const at:
                             const c10::optional<at::Tensor>& seq_starts);
 at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
                           const at::Tensor& labels);
+at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
+                       const c10::optional<at::Tensor>& bias, long act);
 at::Tensor lm_logprobs(const at::Tensor& hidden, const at::Tensor& weight,
                        const at::Tensor& labels);
 void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
@@ -77,4 +79,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("fused_adamw", &fused_adamw);
   mod.def("lm_logprobs", &lm_logprobs);
   mod.def("lm_logprobs_v2", &lm_logprobs_v2);
+  mod.def("skinny_gemm", &skinny_gemm);
 }

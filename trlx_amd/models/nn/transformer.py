@@ -142,7 +142,7 @@ class Attention(nn.Module):
 
     def forward(self, x, ctx: AttentionContext, rope_tables, kv_cache: Optional[KVCache] = None):
         B = x.shape[0]
-        qkv = self.qkv_proj(x)
+        qkv = dense(self.qkv_proj, x)
         # under sequence parallelism the projection gathers the full sequence
         T = qkv.shape[1]
 
@@ -158,7 +158,7 @@ class Attention(nn.Module):
             )
             out = ops.attention_decode(q, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx],
                                        ctx.seq_lens, self.scale, seq_starts=ctx.key_starts)
-            return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
+            return dense(self.o_proj, out.transpose(1, 2).reshape(B, T, -1))
 
         if x.is_cuda and x.dtype == torch.bfloat16 and ops.extension_available():
             # fused split + RoPE + q-scale (one kernel fwd, one bwd)
@@ -193,7 +193,7 @@ class Attention(nn.Module):
                 out = ops.attention_decode(q, k_full, v_full, ctx.seq_lens, self.scale,
                                            seq_starts=ctx.key_starts)
                 out = out.transpose(1, 2).reshape(B, T, -1)
-                return self.o_proj(out)
+                return dense(self.o_proj, out)
             k = k_full[:, :, : ctx.start_pos + T]
             v = v_full[:, :, : ctx.start_pos + T]
 
@@ -211,7 +211,18 @@ class Attention(nn.Module):
             probs = F.dropout(probs, self.attn_pdrop)
         out = torch.matmul(probs, v)
         out = out.transpose(1, 2).reshape(B, T, -1)
-        return self.o_proj(out)
+        return dense(self.o_proj, out)
+
+
+def dense(lin, x, act_code=0, act_fn=None):
+    """Projection dispatch: plain nn.Linear goes through the fused skinny
+    streaming GEMM during decode (ops.skinny_linear handles eligibility);
+    parallel linears keep their collective path, with the activation applied
+    after."""
+    if type(lin) is nn.Linear:
+        return ops.skinny_linear(x, lin.weight, lin.bias, act_code)
+    y = lin(x)
+    return act_fn(y) if act_fn is not None else y
 
 
 _ACTS = {
@@ -242,13 +253,16 @@ class MLP(nn.Module):
                 self.fc_in = nn.Linear(cfg.hidden_size, i, bias=cfg.mlp_bias)
             self.down_proj = nn.Linear(i, cfg.hidden_size, bias=cfg.mlp_bias)
         self.act = _ACTS[cfg.activation]
+        self.act_code = ops.ACT_CODES.get(cfg.activation)
         self.isize = i // tp
 
     def forward(self, x):
         if self.swiglu:
-            gu = self.gate_up_proj(x)
-            return self.down_proj(self.act(gu[..., : self.isize]) * gu[..., self.isize :])
-        return self.down_proj(self.act(self.fc_in(x)))
+            gu = dense(self.gate_up_proj, x)
+            return dense(self.down_proj, self.act(gu[..., : self.isize]) * gu[..., self.isize :])
+        if self.act_code is None:
+            return dense(self.down_proj, self.act(dense(self.fc_in, x)))
+        return dense(self.down_proj, dense(self.fc_in, x, self.act_code, self.act))
 
 
 class Block(nn.Module):

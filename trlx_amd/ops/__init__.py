@@ -115,6 +115,34 @@ def lm_logprobs(hidden: torch.Tensor, weight: torch.Tensor, labels: torch.Tensor
     return reference.logprobs_of_labels(logits, labels)
 
 
+ACT_CODES = {"none": 0, "gelu": 1, "gelu_new": 2, "relu": 3, "silu": 4}
+_ACT_FNS = {
+    0: lambda y: y,
+    1: torch.nn.functional.gelu,
+    2: lambda y: torch.nn.functional.gelu(y, approximate="tanh"),
+    3: torch.nn.functional.relu,
+    4: torch.nn.functional.silu,
+}
+
+
+def skinny_linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Tensor] = None,
+                  act: int = 0) -> torch.Tensor:
+    """Inference linear for skinny-M shapes (KV-cached decode): one 16x16
+    MFMA tile per wave streaming the weight at line rate, bias + activation
+    fused (csrc/skinny_gemm.hip).  Falls back to F.linear (+ activation)
+    when autograd is on, on CPU, or for large M."""
+    K = x.shape[-1]
+    M = x.numel() // K
+    if (x.is_cuda and not torch.is_grad_enabled() and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16 and M <= 256 and K % 32 == 0):
+        ext = _require_ext("skinny_gemm")
+        if ext is not None and hasattr(ext, "skinny_gemm"):
+            y = ext.skinny_gemm(x.reshape(M, K).contiguous(), weight.contiguous(), bias, act)
+            return y.view(*x.shape[:-1], weight.shape[0])
+    y = torch.nn.functional.linear(x, weight, bias)
+    return _ACT_FNS[act](y)
+
+
 # --------------------------------------------------------------------------
 # norms
 # --------------------------------------------------------------------------
