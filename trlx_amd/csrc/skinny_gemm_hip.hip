@@ -4,20 +4,25 @@
 //
 // During KV-cached decode every projection is a [B, hidden] x [out, hidden]
 // GEMM with B ~ 128: hipBLASLt's tiles run it at ~0.4 TB/s effective weight
-// bandwidth (profile r01: 9.3 us for a 3.5 MB weight read — 15x off the
-// streaming roofline; the four per-layer projections ARE the decode time).
-// Weight-streaming shape:
+// bandwidth (profile r01: ~9.3 us in-graph for a 3.5 MB weight read; the
+// four per-layer projections ARE the decode time).  Weight-streaming shape:
 //
-//   - ONE 16x16 output tile per WAVE via a single mfma_16x16x32_bf16 chain:
-//     lane l holds A row (l&15) / W row (l&15) at k-slice (l>>4)*8.  No LDS,
-//     no barriers, ~32 VGPRs -> max occupancy, latency hidden by waves.
-//   - Tile order nt-major within a block: the 4 waves of a block compute 4
-//     consecutive M-tiles of the SAME W rows, so a weight line is fetched
-//     once per block and served from L1/L2 to its neighbors; A (tiny) stays
-//     L2-resident for all N-tiles.
-//   - bias + activation (gelu-erf / gelu-tanh / relu / silu) fused into the
-//     epilogue: kills the separate elementwise kernel per projection
-//     (profile r01: gelu was 3.4% of cycle kernels, all in decode).
+//   - ONE 16x16 output tile per WAVE via an mfma_16x16x32_bf16 chain:
+//     lane l holds A row (l&15) / W row (l&15) at k-slice (l>>4)*8.  No
+//     LDS, no barriers, ~40 VGPRs -> high occupancy.
+//   - 4-deep software-pipelined register loads: the first cut issued 2
+//     loads per 32-K step back-to-back with their MFMA (measured 268 GB/s —
+//     latency-bound at ~4 waves/CU); keeping 8 independent loads in flight
+//     amortizes the ~600-cycle memory latency.
+//   - split-K when the tile count is too small to fill 256 CUs (e.g.
+//     down_proj [128, 768, 3072]: 384 waves): S K-slices write fp32 slabs
+//     [S, M, N], a tiny epilogue kernel sums + bias + activation + casts.
+//   - nt-major within a block: the 4 waves of a block compute 4 consecutive
+//     M-tiles of the SAME W panel (one weight fetch per block); A (tiny)
+//     stays L2-resident.
+//   - bias + activation (gelu-erf / gelu-tanh / relu / silu) fused: kills
+//     the separate per-projection elementwise kernel (gelu was 3.4% of
+//     cycle kernels, all in decode).
 //
 // Inference-only (decode runs under no_grad; training uses hipBLASLt).
 #include <ATen/ATen.h>
@@ -49,15 +54,22 @@ __device__ __forceinline__ float sk_act(float x, int act) {
   }
 }
 
+// SPLIT: C is fp32 slabs [S, M, N] (epilogue applied separately);
+// !SPLIT: C is bf16 [M, N] with bias+act fused.
+template <bool SPLIT>
 __global__ __launch_bounds__(SK_BLOCK) void skinny_gemm_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ W,
-    const bf16_t* __restrict__ bias, bf16_t* __restrict__ C, int M, int N, int K, int act) {
+    const bf16_t* __restrict__ bias, void* __restrict__ Cout, int M, int N, int K, int act,
+    int nsplit) {
   const int lane = threadIdx.x % WAVE;
   const int wid = threadIdx.x / WAVE;
   const int nM = (M + 15) >> 4;
   const int nN = (N + 15) >> 4;
-  const int tile = blockIdx.x * 4 + wid;
-  if (tile >= nM * nN) return;
+  int tile = blockIdx.x * 4 + wid;
+  const int ntile2 = nM * nN;
+  if (tile >= ntile2 * nsplit) return;
+  const int ks = tile / ntile2;  // K-split index
+  tile -= ks * ntile2;
   // nt-major: the block's 4 waves walk M-tiles of one W panel
   const int nt = tile / nM;
   const int mt = tile % nM;
@@ -65,20 +77,45 @@ __global__ __launch_bounds__(SK_BLOCK) void skinny_gemm_kernel(
   const int arow = min(mt * 16 + (lane & 15), M - 1);
   const int wrow = min(nt * 16 + (lane & 15), N - 1);
   const int k8 = (lane >> 4) * 8;
+  // this split's K range (multiples of 32; last split takes the remainder)
+  const int kchunk = ((K / 32) / nsplit) * 32;
+  const int k0 = ks * kchunk;
+  const int k1 = (ks == nsplit - 1) ? K : k0 + kchunk;
   const bf16_t* ap = A + (size_t)arow * K + k8;
   const bf16_t* wp = W + (size_t)wrow * K + k8;
 
   f32x4_sk acc = {0.f, 0.f, 0.f, 0.f};
-  int k = 0;
-  for (; k + 64 <= K; k += 64) {
+  int k = k0;
+  // 4-deep register pipeline (8 loads in flight)
+  if (k + 128 <= k1) {
     bf16x8_sk a0 = *reinterpret_cast<const bf16x8_sk*>(ap + k);
     bf16x8_sk b0 = *reinterpret_cast<const bf16x8_sk*>(wp + k);
     bf16x8_sk a1 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 32);
     bf16x8_sk b1 = *reinterpret_cast<const bf16x8_sk*>(wp + k + 32);
+    bf16x8_sk a2 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 64);
+    bf16x8_sk b2 = *reinterpret_cast<const bf16x8_sk*>(wp + k + 64);
+    bf16x8_sk a3 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 96);
+    bf16x8_sk b3 = *reinterpret_cast<const bf16x8_sk*>(wp + k + 96);
+    for (k += 128; k + 128 <= k1; k += 128) {
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
+      a0 = *reinterpret_cast<const bf16x8_sk*>(ap + k);
+      b0 = *reinterpret_cast<const bf16x8_sk*>(wp + k);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
+      a1 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 32);
+      b1 = *reinterpret_cast<const bf16x8_sk*>(wp + k + 32);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc, 0, 0, 0);
+      a2 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 64);
+      b2 = *reinterpret_cast<const bf16x8_sk*>(wp + k + 64);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc, 0, 0, 0);
+      a3 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 96);
+      b3 = *reinterpret_cast<const bf16x8_sk*>(wp + k + 96);
+    }
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc, 0, 0, 0);
   }
-  for (; k < K; k += 32) {
+  for (; k < k1; k += 32) {
     bf16x8_sk a = *reinterpret_cast<const bf16x8_sk*>(ap + k);
     bf16x8_sk b = *reinterpret_cast<const bf16x8_sk*>(wp + k);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
@@ -87,14 +124,35 @@ __global__ __launch_bounds__(SK_BLOCK) void skinny_gemm_kernel(
   // D layout: row = (lane>>4)*4 + r (M dim), col = lane&15 (N dim)
   const int ccol = nt * 16 + (lane & 15);
   if (ccol >= N) return;
-  float bv = bias ? ScalarIO<bf16_t>::load(bias + ccol) : 0.f;
+  if (SPLIT) {
+    float* c = reinterpret_cast<float*>(Cout) + (size_t)ks * M * N;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int crow = mt * 16 + (lane >> 4) * 4 + r;
-    if (crow < M) {
-      ScalarIO<bf16_t>::store(C + (size_t)crow * N + ccol, sk_act(acc[r] + bv, act));
+    for (int r = 0; r < 4; ++r) {
+      const int crow = mt * 16 + (lane >> 4) * 4 + r;
+      if (crow < M) c[(size_t)crow * N + ccol] = acc[r];
+    }
+  } else {
+    bf16_t* c = reinterpret_cast<bf16_t*>(Cout);
+    float bv = bias ? ScalarIO<bf16_t>::load(bias + ccol) : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int crow = mt * 16 + (lane >> 4) * 4 + r;
+      if (crow < M) {
+        ScalarIO<bf16_t>::store(c + (size_t)crow * N + ccol, sk_act(acc[r] + bv, act));
+      }
     }
   }
+}
+
+__global__ void skinny_epilogue_kernel(const float* __restrict__ ws,
+                                       const bf16_t* __restrict__ bias, bf16_t* __restrict__ C,
+                                       long MN, int N, int act, int nsplit) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= MN) return;
+  float v = 0.f;
+  for (int s = 0; s < nsplit; ++s) v += ws[(size_t)s * MN + i];
+  if (bias) v += ScalarIO<bf16_t>::load(bias + (i % N));
+  ScalarIO<bf16_t>::store(C + i, sk_act(v, act));
 }
 
 }  // namespace
@@ -117,11 +175,26 @@ at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
     bp = reinterpret_cast<const bf16_t*>(bc.data_ptr());
   }
   const int ntiles = ((M + 15) / 16) * ((N + 15) / 16);
-  const int grid = (ntiles + 3) / 4;
+  // fill the chip: target >= 1024 waves (256 CUs x 4); cap splits at K/64
+  int nsplit = 1;
+  while (nsplit < 8 && ntiles * nsplit < 1024 && (K / 32) / (nsplit * 2) >= 2) nsplit *= 2;
   auto stream = c10::hip::getCurrentHIPStream();
- hipLaunchKernelGGL(( skinny_gemm_kernel), dim3(grid), dim3(SK_BLOCK), 0, stream, 
-      reinterpret_cast<const bf16_t*>(a.data_ptr()), reinterpret_cast<const bf16_t*>(w.data_ptr()),
-      bp, reinterpret_cast<bf16_t*>(c.data_ptr()), M, N, K, (int)act);
+  const auto ap = reinterpret_cast<const bf16_t*>(a.data_ptr());
+  const auto wp = reinterpret_cast<const bf16_t*>(w.data_ptr());
+  if (nsplit == 1) {
+    const int grid = (ntiles + 3) / 4;
+   hipLaunchKernelGGL(( skinny_gemm_kernel<false>), dim3(grid), dim3(SK_BLOCK), 0, stream, 
+        ap, wp, bp, c.data_ptr(), M, N, K, (int)act, 1);
+  } else {
+    auto ws = at::empty({nsplit, (long)M, (long)N}, a.options().dtype(at::kFloat));
+    const int grid = (ntiles * nsplit + 3) / 4;
+   hipLaunchKernelGGL(( skinny_gemm_kernel<true>), dim3(grid), dim3(SK_BLOCK), 0, stream, 
+        ap, wp, nullptr, ws.data_ptr(), M, N, K, (int)act, nsplit);
+    const long MN = (long)M * N;
+   hipLaunchKernelGGL(( skinny_epilogue_kernel), dim3((int)((MN + 255) / 256)), dim3(256), 0, stream, 
+        ws.data_ptr<float>(), bp, reinterpret_cast<bf16_t*>(c.data_ptr()), MN, N, (int)act,
+        nsplit);
+  }
   HIP_CHECK_LAST();
   return c;
 }

@@ -188,6 +188,20 @@ class NativeRLTrainer(BaseRLTrainer):
         if prompt_sizes is None:
             prompt_sizes = [len(prompts[0])] * len(prompts)
 
+        # ONE device->host transfer per batch: decoding row-by-row from GPU
+        # tensors costs a ~50us sync per row (measured: 15 ms of a 190 ms
+        # PPO cycle at chunk=128)
+        def to_lists(x):
+            if torch.is_tensor(x):
+                return x.cpu().tolist()
+            if isinstance(x, (list, tuple)) and len(x) and torch.is_tensor(x[0]):
+                return [t.cpu().tolist() for t in x]
+            return x
+
+        prompts = to_lists(prompts)
+        samples = to_lists(samples)
+        prompt_sizes = to_lists(prompt_sizes)
+
         str_samples, str_prompts, str_outputs = [], [], []
         for prompt, sample, prompt_size in zip(prompts, samples, prompt_sizes):
             if self.config.model.model_arch_type == "seq2seq":
