@@ -206,7 +206,7 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         )
         hs = out.last_hidden_state
         if logits_slice is not None:
-            hs = hs[:, logits_slice[0] : logits_slice[1]]
+            hs = hs[:, logits_slice[0] : logits_slice[1]].contiguous()
         values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1)
         return CausalLMOutputWithValue(
             logits=out.logits, values=values.float(), last_hidden_state=out.last_hidden_state
@@ -224,7 +224,7 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         """Fused experience path without a frozen branch (ref handled by the
         caller); see the hydra variant for semantics."""
         out = self.base_model(input_ids, attention_mask=attention_mask, return_logits=False)
-        h = out.last_hidden_state[:, lo:hi]
+        h = out.last_hidden_state[:, lo:hi].contiguous()
         B, T = h.shape[:2]
         values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         lm = self.base_model.lm_head
@@ -315,7 +315,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         )
         hs = out.last_hidden_state
         if logits_slice is not None:
-            hs = hs[:, logits_slice[0] : logits_slice[1]]
+            hs = hs[:, logits_slice[0] : logits_slice[1]].contiguous()
         values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         ref_logits = None
         if return_ref_logits and self.frozen_head is not None:
@@ -352,7 +352,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         stash = -self.num_layers_unfrozen if self.frozen_head is not None else None
         out = self.base_model(input_ids, attention_mask=attention_mask,
                               hidden_at_layer=stash, return_logits=False)
-        h = out.last_hidden_state[:, lo:hi]
+        h = out.last_hidden_state[:, lo:hi].contiguous()
         B, T = h.shape[:2]
         values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         lm = self.base_model.lm_head
@@ -394,7 +394,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
             with lora_disabled(self.base_model):
                 rout = self.base_model(input_ids, attention_mask=attention_mask,
                                        return_logits=False)
-            rh = rout.last_hidden_state[:, lo:hi]
+            rh = rout.last_hidden_state[:, lo:hi].contiguous()
             if fused_ok(lm):
                 ref_logprobs = ops.lm_logprobs(rh.reshape(-1, rh.shape[-1]).contiguous(),
                                                lm.weight, flat_labels).view(B, T)

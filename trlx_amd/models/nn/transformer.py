@@ -509,7 +509,10 @@ class CausalTransformer(nn.Module):
             # logits_slice: compute the [V]-wide projection only where the
             # loss needs it (PPO: response positions — the lm_head GEMM and
             # its backward are the largest single kernels in the train step)
-            hs = h if logits_slice is None else h[:, logits_slice[0] : logits_slice[1]]
+            # .contiguous(): a T-slice view makes the lm_head GEMM run as a
+            # B-batched strided GEMM with M=resp_len per batch (measured
+            # 480 us vs 152 us for the flat [B*resp, V] GEMM)
+            hs = h if logits_slice is None else h[:, logits_slice[0] : logits_slice[1]].contiguous()
             logits = self.lm_head(hs)
         return TransformerOutput(logits=logits, last_hidden_state=h, hidden_at_layer=hidden_at)
 
