@@ -149,3 +149,38 @@ def test_reward_model_pairwise_loss_semantics():
     # inference path: identical halves -> only chosen_end_scores
     out_inf = model(torch.cat([chosen, chosen]))
     assert set(out_inf.keys()) == {"chosen_end_scores"}
+
+
+def test_hh_example_and_reward_server(tmp_path, monkeypatch):
+    """HH PPO example + the out-of-band reward server (parity: reference
+    examples/hh/ppo_hh.py + Triton channel)."""
+    monkeypatch.syspath_prepend("examples/hh")
+    import importlib
+
+    import conftest
+
+    hh_task = importlib.import_module("hh_task")
+    scores = hh_task.oracle_reward(["\n\nHuman: hi\n\nAssistant: sure I can help"])
+    assert scores[0] > 0
+
+    # reward server endpoint (in-process TestClient; oracle mode)
+    from fastapi.testclient import TestClient
+
+    server = importlib.import_module("reward_server")
+    client = TestClient(server.app)
+    assert client.get("/health").json()["ok"]
+    r = client.post("/reward", json={"samples": ["\n\nHuman: q\n\nAssistant: go away"]})
+    assert r.status_code == 200 and r.json()["scores"][0] < 0
+
+    # tiny PPO run with the in-process oracle
+    ppo_hh = importlib.import_module("ppo_hh")
+    ppo_hh.main({
+        "train.total_steps": 2, "train.epochs": 1, "train.batch_size": 4,
+        "train.eval_interval": 2, "train.checkpoint_interval": 100,
+        "train.checkpoint_dir": str(tmp_path / "ckpt"), "train.tracker": None,
+        "train.save_best": False, "train.seq_length": 48,
+        "model.model_extra_configs": {"config": conftest.tiny_config().to_dict()},
+        "model.num_layers_unfrozen": 1,
+        "method.num_rollouts": 4, "method.chunk_size": 4, "method.ppo_epochs": 1,
+        "method.gen_kwargs": dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True),
+    })

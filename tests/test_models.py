@@ -305,3 +305,28 @@ def test_ilql_shaped_generate_uses_graph_on_gpu():
     out2 = model.generate(ids, max_new_tokens=8, beta=2, top_k=10, temperature=1.0,
                           eos_token_id=1, pad_token_id=2)
     assert out2.shape[0] == 4
+
+
+def test_gradient_checkpointing_equivalence(tiny_model_factory=None):
+    """Activation checkpointing: identical loss and grads,
+    activations recomputed in backward."""
+    import conftest
+    from trlx_amd.models.nn.transformer import CausalTransformer
+
+    torch.manual_seed(11)
+    cfg = conftest.tiny_config()
+    m1 = CausalTransformer(cfg)
+    m2 = CausalTransformer(cfg)
+    m2.load_state_dict(m1.state_dict())
+    m2.gradient_checkpointing = True
+    m1.train()
+    m2.train()
+    ids = torch.randint(3, cfg.vocab_size, (2, 10))
+    l1 = m1(ids).logits.float().pow(2).mean()
+    l2 = m2(ids).logits.float().pow(2).mean()
+    assert torch.allclose(l1, l2, atol=1e-6)
+    l1.backward()
+    l2.backward()
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
+        if p1.grad is not None:
+            assert torch.allclose(p1.grad, p2.grad, atol=1e-5), n1

@@ -57,6 +57,10 @@ class ModelConfig:
     num_layers_unfrozen: int = -1
     model_extra_configs: Dict[str, Any] = field(default_factory=dict)
     peft_config: Optional[Any] = None
+    # recompute block activations in backward (parity: NeMo
+    # activations_checkpoint_method, megatron_20b.yaml:77-79) — trades ~30%
+    # step time for activation memory on long-context runs
+    gradient_checkpointing: bool = False
 
     @classmethod
     def from_dict(cls, config: Dict[str, Any]) -> "ModelConfig":

@@ -336,6 +336,7 @@ class CausalTransformer(nn.Module):
         if cfg.tie_word_embeddings:
             self.lm_head.weight = self.embed_tokens.weight
         self.embd_pdrop = cfg.embd_pdrop
+        self.gradient_checkpointing = False
         if cfg.position_encoding == "rope":
             rot = int(cfg.head_dim * cfg.rope_pct)
             rot -= rot % 2
@@ -486,12 +487,20 @@ class CausalTransformer(nn.Module):
             stash_at = hidden_at_layer % n
         hidden_at = None
         res = None
+        use_ckpt = self.gradient_checkpointing and self.training and torch.is_grad_enabled()
         for i, layer in enumerate(self.layers):
             if stash_at is not None and i == stash_at:
                 # materialize the true stream at this layer's input (the
                 # pending residual is otherwise deferred into its norm)
                 hidden_at = h if res is None else h + res
-            h, res = layer(h, ctx, self.rope_tables, kv_cache, res=res)
+            if use_ckpt:
+                # activation checkpointing: recompute the
+                # block in backward instead of saving activations
+                h, res = torch.utils.checkpoint.checkpoint(
+                    lambda hh, rr, _l=layer: _l(hh, ctx, self.rope_tables, kv_cache, res=rr),
+                    h, res, use_reentrant=False)
+            else:
+                h, res = layer(h, ctx, self.rope_tables, kv_cache, res=res)
         if res is None:
             h = self.final_norm(h)
         else:

@@ -154,6 +154,8 @@ class NativeRLTrainer(BaseRLTrainer):
         if (self.config.train.sequence_parallel and self.tp_size > 1
                 and hasattr(base, "set_sequence_parallel")):
             base.set_sequence_parallel(True)
+        if getattr(self.config.model, "gradient_checkpointing", False):
+            base.gradient_checkpointing = True
         model = model.to(self.device)
         if self.dtype != torch.float32:
             # heads stay fp32; trunk + frozen branch go bf16
