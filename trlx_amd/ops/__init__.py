@@ -134,7 +134,8 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Te
     K = x.shape[-1]
     M = x.numel() // K
     if (x.is_cuda and not torch.is_grad_enabled() and x.dtype == torch.bfloat16
-            and weight.dtype == torch.bfloat16 and M <= 256 and K % 32 == 0):
+            and weight.dtype == torch.bfloat16 and M <= 256 and K % 32 == 0
+            and os.environ.get("TRLX_AMD_NO_SKINNY") != "1"):
         ext = _require_ext("skinny_gemm")
         if ext is not None and hasattr(ext, "skinny_gemm"):
             y = ext.skinny_gemm(x.reshape(M, K).contiguous(), weight.contiguous(), bias, act)
