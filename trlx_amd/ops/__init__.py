@@ -130,12 +130,17 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Te
     """Inference linear for skinny-M shapes (KV-cached decode): one 16x16
     MFMA tile per wave streaming the weight at line rate, bias + activation
     fused (csrc/skinny_gemm.hip).  Falls back to F.linear (+ activation)
-    when autograd is on, on CPU, or for large M."""
+    when autograd is on, on CPU, or for large M.
+
+    OPT-IN (TRLX_AMD_SKINNY=1): wins its standalone microbench 1.2-2.4x,
+    but inside the captured decode graph hipBLASLt's algos already run
+    these shapes faster (same-box A/B: 749 vs 722 samples/s with the skinny
+    path on) — the standalone hipBLASLt baseline was cold-start slow."""
     K = x.shape[-1]
     M = x.numel() // K
     if (x.is_cuda and not torch.is_grad_enabled() and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and M <= 256 and K % 32 == 0
-            and os.environ.get("TRLX_AMD_NO_SKINNY") != "1"):
+            and os.environ.get("TRLX_AMD_SKINNY") == "1"):
         ext = _require_ext("skinny_gemm")
         if ext is not None and hasattr(ext, "skinny_gemm"):
             y = ext.skinny_gemm(x.reshape(M, K).contiguous(), weight.contiguous(), bias, act)
