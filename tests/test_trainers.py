@@ -255,3 +255,45 @@ def test_gen_kwarg_sweep_eval(tmp_path):
     )
     assert trainer.generate_sweep_kwarg is not None
     assert trainer.iter_count == 2
+
+
+@pytest.mark.gpu
+def test_train_graph_matches_eager(tmp_path, monkeypatch):
+    """hipGraph-captured train step matches eager: identical seeds/rollouts
+    must yield (near-)identical final weights after several optimizer steps
+    (the graph replays the same kernels in the same order)."""
+    weights = {}
+    for mode in ("graph", "eager"):
+        if mode == "eager":
+            monkeypatch.setenv("TRLX_AMD_NO_TRAIN_GRAPH", "1")
+        else:
+            monkeypatch.delenv("TRLX_AMD_NO_TRAIN_GRAPH", raising=False)
+        cfg = _tiny_model_cfg(default_ppo_config(), tmp_path / mode)
+        cfg.train.total_steps = 4
+        cfg.train.eval_interval = 100
+        cfg.train.checkpoint_interval = 100
+        cfg.train.save_best = False
+        cfg.model.num_layers_unfrozen = 1
+        cfg.method.num_rollouts = 8
+        cfg.method.chunk_size = 8
+        cfg.method.ppo_epochs = 2
+        cfg.train.seed = 1234
+        cfg.method.gen_kwargs["seed"] = 7
+
+        def reward_fn(samples, prompts, outputs, **kw):
+            return [float(len(o)) for o in outputs]
+
+        trainer = trlx_amd.train(
+            reward_fn=reward_fn,
+            prompts=["hello", "world", "foo bar", "baz"] * 2,
+            eval_prompts=["hello", "sky"],
+            config=cfg,
+        )
+        if mode == "graph":
+            assert getattr(trainer, "_train_graphs", None), "graph path was not used"
+        weights[mode] = {k: v.detach().float().cpu()
+                         for k, v in trainer.model.state_dict().items()}
+        del trainer
+        torch.cuda.empty_cache()
+    for k in weights["graph"]:
+        assert torch.allclose(weights["graph"][k], weights["eager"][k], atol=1e-3), k
