@@ -270,7 +270,8 @@ def test_train_graph_matches_eager(tmp_path, monkeypatch):
             monkeypatch.delenv("TRLX_AMD_NO_TRAIN_GRAPH", raising=False)
         cfg = _tiny_model_cfg(default_ppo_config(), tmp_path / mode)
         # decode kernel needs head_dim >= 32
-        cfg.model.model_extra_configs["config"]["num_heads"] = 2
+        cfg.model.model_extra_configs["config"].update(
+            num_heads=2, num_kv_heads=2, head_dim=32)
         cfg.train.total_steps = 4
         cfg.train.eval_interval = 100
         cfg.train.checkpoint_interval = 100

@@ -477,6 +477,8 @@ class NativeRLTrainer(BaseRLTrainer):
                     if not all(torch.is_tensor(v) for v in stats.values()):
                         torch.cuda.current_stream().wait_stream(side)
                         self._train_graphs[key] = False
+                        logger.info("train-step graph disabled: non-tensor stats %s",
+                                    [k for k, v in stats.items() if not torch.is_tensor(v)])
                         return None
                     loss.backward()
             torch.cuda.current_stream().wait_stream(side)
@@ -490,6 +492,8 @@ class NativeRLTrainer(BaseRLTrainer):
             self.opt.zero_grad()
             entry = (graph, static, loss_out, stats_out)
             self._train_graphs[key] = entry
+            logger.info("captured train-step graph #%d for %s",
+                        len([v for v in self._train_graphs.values() if v]), key[:2])
         elif entry is False:
             return None
 
