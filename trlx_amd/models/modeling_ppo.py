@@ -181,7 +181,7 @@ class CausalLMOutputWithValue:
 class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
     """Native LM + scalar value head (reference modeling_ppo.py:266-382)."""
 
-    _supported_modules = ["v_head"]
+    _supported_modules = ["v_head", "v_branch"]
     _supported_args = ["peft_config", "num_value_layers_unfrozen"]
 
     def __init__(self, base_model: CausalTransformer, peft_config=None, num_value_layers_unfrozen: int = 0):
@@ -190,6 +190,23 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         self.num_value_layers_unfrozen = num_value_layers_unfrozen
         # value head computes in fp32 regardless of trunk dtype
         self.v_head = make_head(self.config.hidden_size, 1, dtype=torch.float32)
+        # num_value_layers_unfrozen > 0: the value function gets its own
+        # trainable copies of the top layers (reference make_value_branch)
+        self.v_branch = None
+        if num_value_layers_unfrozen > 0:
+            self.v_branch = ValueBranch(base_model, num_value_layers_unfrozen)
+
+    def _value_stash(self):
+        return [-self.num_value_layers_unfrozen] if self.v_branch is not None else []
+
+    def _branch_values(self, out, input_ids, attention_mask, position_ids, logits_slice):
+        hidden = out.hidden_at_layer
+        if isinstance(hidden, dict):
+            hidden = hidden[-self.num_value_layers_unfrozen]
+        ctx = self.base_model.make_context(input_ids, attention_mask, 0)
+        if position_ids is not None:
+            ctx.position_ids = position_ids.to(torch.int32)
+        return self.v_branch(hidden, ctx, self.base_model.rope_tables, logits_slice)
 
     def forward(
         self,
@@ -202,14 +219,18 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
     ) -> CausalLMOutputWithValue:
         out = self.base_model(
             input_ids, attention_mask=attention_mask, position_ids=position_ids,
-            logits_slice=logits_slice,
+            logits_slice=logits_slice, hidden_at_layer=self._value_stash() or None,
         )
-        hs = out.last_hidden_state
-        if logits_slice is not None:
-            hs = hs[:, logits_slice[0] : logits_slice[1]].contiguous()
-        values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1)
+        if self.v_branch is not None:
+            values = self._branch_values(out, input_ids, attention_mask, position_ids,
+                                         logits_slice)
+        else:
+            hs = out.last_hidden_state
+            if logits_slice is not None:
+                hs = hs[:, logits_slice[0] : logits_slice[1]].contiguous()
+            values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         return CausalLMOutputWithValue(
-            logits=out.logits, values=values.float(), last_hidden_state=out.last_hidden_state
+            logits=out.logits, values=values, last_hidden_state=out.last_hidden_state
         )
 
     def generate(self, input_ids, attention_mask=None, **kwargs):
@@ -223,10 +244,14 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
                            labels: torch.Tensor):
         """Fused experience path without a frozen branch (ref handled by the
         caller); see the hydra variant for semantics."""
-        out = self.base_model(input_ids, attention_mask=attention_mask, return_logits=False)
+        out = self.base_model(input_ids, attention_mask=attention_mask, return_logits=False,
+                              hidden_at_layer=self._value_stash() or None)
         h = out.last_hidden_state[:, lo:hi].contiguous()
         B, T = h.shape[:2]
-        values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        if self.v_branch is not None:
+            values = self._branch_values(out, input_ids, attention_mask, None, (lo, hi))
+        else:
+            values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         lm = self.base_model.lm_head
         flat_labels = labels.reshape(-1)
         import os
@@ -239,6 +264,38 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         else:
             logprobs = ops.logprobs_of_labels(lm(h), labels)
         return logprobs, None, values
+
+
+class ValueBranch(nn.Module):
+    """Trainable value branch: copies of the top ``num_value_layers_unfrozen``
+    blocks + final norm + scalar head, fed from the stashed pre-branch hidden
+    state — the value function gets its own layers decoupled from the policy
+    (reference modeling_ppo.py:255-263 ``make_value_branch``)."""
+
+    def __init__(self, base_model: CausalTransformer, num_layers: int):
+        super().__init__()
+        self.num_layers = num_layers
+        self.blocks = nn.ModuleList(
+            copy.deepcopy(block) for block in base_model.layers[-num_layers:]
+        )
+        self.final_norm = copy.deepcopy(base_model.final_norm)
+        self.v_head = make_head(base_model.config.hidden_size, 1, dtype=torch.float32)
+        for p in self.parameters():  # copies of frozen trunk layers train here
+            p.requires_grad_(True)
+
+    def forward(self, hidden: torch.Tensor, ctx, rope_tables, logits_slice=None) -> torch.Tensor:
+        h, res = hidden, None
+        for block in self.blocks:
+            h, res = block(h, ctx, rope_tables, res=res)
+        if logits_slice is not None:
+            h = h[:, logits_slice[0] : logits_slice[1]].contiguous()
+            if res is not None:
+                res = res[:, logits_slice[0] : logits_slice[1]].contiguous()
+        if res is None:
+            h = self.final_norm(h)
+        else:
+            h = self.final_norm.forward_add(h, res)[0]
+        return self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
 
 
 class FrozenBranch(nn.Module):
@@ -306,23 +363,30 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         """One trunk pass; optionally also the reference logits via the frozen
         branch on the stashed pre-branch hidden state (MI355X fusion of the
         reference's forward + forward_hydra pair)."""
-        stash = None
+        stash = list(self._value_stash())
         if return_ref_logits and self.frozen_head is not None:
-            stash = -self.num_layers_unfrozen
+            stash.append(-self.num_layers_unfrozen)
         out = self.base_model(
             input_ids, attention_mask=attention_mask, position_ids=position_ids,
-            hidden_at_layer=stash, logits_slice=logits_slice,
+            hidden_at_layer=stash or None, logits_slice=logits_slice,
         )
-        hs = out.last_hidden_state
-        if logits_slice is not None:
-            hs = hs[:, logits_slice[0] : logits_slice[1]].contiguous()
-        values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        if self.v_branch is not None:
+            values = self._branch_values(out, input_ids, attention_mask, position_ids,
+                                         logits_slice)
+        else:
+            hs = out.last_hidden_state
+            if logits_slice is not None:
+                hs = hs[:, logits_slice[0] : logits_slice[1]].contiguous()
+            values = self.v_head(hs.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         ref_logits = None
         if return_ref_logits and self.frozen_head is not None:
             ctx = self.base_model.make_context(input_ids, attention_mask, 0)
             if position_ids is not None:
                 ctx.position_ids = position_ids.to(torch.int32)
-            ref_logits = self.frozen_head(out.hidden_at_layer, ctx, self.base_model.rope_tables,
+            ref_hidden = out.hidden_at_layer
+            if isinstance(ref_hidden, dict):
+                ref_hidden = ref_hidden[-self.num_layers_unfrozen]
+            ref_logits = self.frozen_head(ref_hidden, ctx, self.base_model.rope_tables,
                                           logits_slice=logits_slice)
         elif return_ref_logits and self.peft_config is not None:
             # peft hydra: the base model WITHOUT adapters is the reference
@@ -349,12 +413,17 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         (logprobs, ref_logprobs, values) each [B, hi-lo]; ref is None when
         there is no frozen branch (caller falls back to a separate ref model
         or adapter toggling)."""
-        stash = -self.num_layers_unfrozen if self.frozen_head is not None else None
+        stash = list(self._value_stash())
+        if self.frozen_head is not None:
+            stash.append(-self.num_layers_unfrozen)
         out = self.base_model(input_ids, attention_mask=attention_mask,
-                              hidden_at_layer=stash, return_logits=False)
+                              hidden_at_layer=stash or None, return_logits=False)
         h = out.last_hidden_state[:, lo:hi].contiguous()
         B, T = h.shape[:2]
-        values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        if self.v_branch is not None:
+            values = self._branch_values(out, input_ids, attention_mask, None, (lo, hi))
+        else:
+            values = self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         lm = self.base_model.lm_head
 
         def fused_ok(head):
@@ -379,7 +448,10 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         ref_logprobs = None
         if self.frozen_head is not None:
             ctx = self.base_model.make_context(input_ids, attention_mask, 0)
-            rh = self.frozen_head.forward_hidden(out.hidden_at_layer, ctx,
+            ref_hidden = out.hidden_at_layer
+            if isinstance(ref_hidden, dict):
+                ref_hidden = ref_hidden[-self.num_layers_unfrozen]
+            rh = self.frozen_head.forward_hidden(ref_hidden, ctx,
                                                  self.base_model.rope_tables,
                                                  logits_slice=(lo, hi))
             rlm = self.frozen_head.lm_head

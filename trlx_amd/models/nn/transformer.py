@@ -482,17 +482,29 @@ class CausalTransformer(nn.Module):
             h = scatter_to_sp(h)
 
         n = len(self.layers)
-        stash_at = None
-        if hidden_at_layer is not None:
-            stash_at = hidden_at_layer % n
-        hidden_at = None
+        # hidden_at_layer: int -> single stash (tensor result); list of ints
+        # -> multiple stashes (dict result, keyed by the REQUESTED values) —
+        # the hydra reference branch and the trainable value branch can hook
+        # different depths
+        multi = isinstance(hidden_at_layer, (list, tuple))
+        wanted = list(hidden_at_layer) if multi else (
+            [hidden_at_layer] if hidden_at_layer is not None else [])
+        stash_for = {}
+        for w in wanted:
+            stash_for.setdefault(w % n, []).append(w)
+        hidden_at = {} if multi else None
         res = None
         use_ckpt = self.gradient_checkpointing and self.training and torch.is_grad_enabled()
         for i, layer in enumerate(self.layers):
-            if stash_at is not None and i == stash_at:
+            if i in stash_for:
                 # materialize the true stream at this layer's input (the
                 # pending residual is otherwise deferred into its norm)
-                hidden_at = h if res is None else h + res
+                stream = h if res is None else h + res
+                if multi:
+                    for w in stash_for[i]:
+                        hidden_at[w] = stream
+                else:
+                    hidden_at = stream
             if use_ckpt:
                 # activation checkpointing: recompute the
                 # block in backward instead of saving activations
@@ -511,7 +523,9 @@ class CausalTransformer(nn.Module):
             from ...parallel.tp import gather_from_sp_replicated
 
             h = gather_from_sp_replicated(h)
-            if hidden_at is not None:
+            if isinstance(hidden_at, dict):
+                hidden_at = {k: gather_from_sp_replicated(v) for k, v in hidden_at.items()}
+            elif hidden_at is not None:
                 hidden_at = gather_from_sp_replicated(hidden_at)
         logits = None
         if return_logits:
