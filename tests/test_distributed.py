@@ -416,7 +416,17 @@ def _worker_pp_train(rank):
         return F.cross_entropy(logits[:, :-1].reshape(-1, 150).float(),
                                ids[:, 1:].reshape(-1))
 
-    mean_loss = runner.forward_backward(mbs, loss_fn)
+    mean_loss = runner.forward_backward(mbs, loss_fn)  # default: 1F1B
+
+    # gpipe schedule must produce identical losses and grads
+    grads_1f1b = {k: p.grad.clone() for k, p in stage.named_parameters()}
+    stage.zero_grad()
+    runner_gp = PipelineRunner(stage, pp_ranks=[0, 1], schedule="gpipe")
+    mean_loss_gp = runner_gp.forward_backward(mbs, loss_fn)
+    if stage.is_last:
+        assert torch.allclose(mean_loss, mean_loss_gp, atol=1e-6)
+    for k, p in stage.named_parameters():
+        assert torch.allclose(p.grad, grads_1f1b[k], atol=1e-5), k
 
     # reference: same microbatches through the full model
     full.zero_grad()

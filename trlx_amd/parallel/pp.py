@@ -5,11 +5,10 @@ row: per-stage model construction modeling_nemo_ppo.py:497-536, p2p
 activations of shape [seq, micro_batch, hidden], fwd/bwd schedule 713-731).
 
 MI355X design: activations/grads move between stages as [B, T, H] bf16
-tensors over RCCL p2p (xGMI on-node); the schedule is GPipe fill-drain —
-all microbatch forwards (stashing each stage's input/output pair), then all
-backwards stitched through autograd (grad of the stage output is received
-from downstream, the grad of the stage input is sent upstream).  1F1B
-interleaving and token-level pipelined generation are planned follow-ups.
+tensors over RCCL p2p (xGMI on-node).  Default schedule is 1F1B (warmup
+forwards, one-forward/one-backward steady state, drain — activation
+footprint bounded by pipeline depth); GPipe fill-drain is kept as an
+alternative.  Token-level pipelined generation is a planned follow-up.
 
 Rank layout (topo.init_model_parallel): pp is the middle axis —
 rank = (dp_idx * pp_size + pp_idx) * tp_size + tp_idx — so a PP group's
@@ -136,11 +135,23 @@ class PipelineStage(nn.Module):
 
 
 class PipelineRunner:
-    """GPipe fill-drain forward/backward over the PP group."""
+    """Pipeline forward/backward over the PP group.
 
-    def __init__(self, stage: PipelineStage, pp_group=None, pp_ranks: Optional[List[int]] = None):
+    Two schedules:
+    - ``1f1b`` (default; parity: the NeMo/Megatron schedule): each stage runs
+      ``pp_size - idx - 1`` warmup forwards, then alternates one-forward/
+      one-backward (backwards in FIFO order), then drains.  Peak live
+      activations are bounded by the pipeline depth instead of the microbatch
+      count.  Sends are non-blocking (isend) so the crossing act/grad
+      messages of the steady state can't deadlock.
+    - ``gpipe``: plain fill-drain (all forwards, then all backwards LIFO).
+    """
+
+    def __init__(self, stage: PipelineStage, pp_group=None, pp_ranks: Optional[List[int]] = None,
+                 schedule: str = "1f1b"):
         self.stage = stage
         self.group = pp_group
+        self.schedule = schedule
         # global ranks of the pipeline stages in order
         if pp_ranks is None:
             pp_ranks = list(range(dist.get_world_size()))
@@ -148,9 +159,16 @@ class PipelineRunner:
         self.idx = pp_ranks.index(dist.get_rank())
         self.prev = pp_ranks[self.idx - 1] if self.idx > 0 else None
         self.next = pp_ranks[self.idx + 1] if self.idx < len(pp_ranks) - 1 else None
+        self._pending = []  # (work, tensor) — tensor kept alive until waited
 
     def _send(self, t: torch.Tensor, dst: int):
-        dist.send(t.contiguous(), dst)
+        t = t.contiguous()
+        self._pending.append((dist.isend(t, dst), t))
+
+    def _drain_sends(self):
+        for work, _ in self._pending:
+            work.wait()
+        self._pending.clear()
 
     def _recv(self, shape, dtype, device, src: int) -> torch.Tensor:
         t = torch.empty(shape, dtype=dtype, device=device)
@@ -158,12 +176,76 @@ class PipelineRunner:
         return t
 
     def forward_backward(self, microbatches, loss_fn, hidden_dtype=torch.float32):
-        """Run fill-drain fwd+bwd.  ``microbatches``: list of dicts with
+        """Run pipelined fwd+bwd.  ``microbatches``: list of dicts with
         input_ids/attention_mask (used on stage 0 for embeddings and on every
         stage for the attention context).  ``loss_fn(logits, mb) -> loss`` is
         evaluated on the last stage.  Returns mean loss (last stage) or None.
         Gradients accumulate into stage parameters; the caller averages over
         microbatch count via the optimizer's grad scale."""
+        if self.schedule == "1f1b":
+            return self._forward_backward_1f1b(microbatches, loss_fn, hidden_dtype)
+        return self._forward_backward_gpipe(microbatches, loss_fn, hidden_dtype)
+
+    def _forward_backward_1f1b(self, microbatches, loss_fn, hidden_dtype=torch.float32):
+        stage = self.stage
+        H = stage.config.hidden_size
+        device = next(stage.parameters()).device
+        from collections import deque
+
+        stashes = deque()
+        losses = []
+        n = len(microbatches)
+        fwd_i = 0
+
+        def do_forward():
+            nonlocal fwd_i
+            mb = microbatches[fwd_i]
+            fwd_i += 1
+            ids = mb["input_ids"].to(device)
+            mask = mb.get("attention_mask")
+            mask = mask.to(device) if mask is not None else None
+            ctx = stage.make_context(ids, mask)
+            if stage.is_first:
+                h_in = None
+                inp = ids
+            else:
+                B, T = ids.shape
+                h_in = self._recv((B, T, H), hidden_dtype, device, self.prev)
+                h_in.requires_grad_(True)
+                inp = h_in
+            out = stage(inp, ctx)
+            if stage.is_last:
+                loss = loss_fn(stage.project(out), mb)
+                losses.append(loss.detach())
+                stashes.append((h_in, loss))
+            else:
+                self._send(out.detach(), self.next)
+                stashes.append((h_in, out))
+
+        def do_backward():
+            h_in, out_or_loss = stashes.popleft()  # FIFO: 1F1B backs up in order
+            if stage.is_last:
+                out_or_loss.backward()
+            else:
+                grad = self._recv(out_or_loss.shape, hidden_dtype, device, self.next)
+                out_or_loss.backward(grad)
+            if not stage.is_first:
+                self._send(h_in.grad, self.prev)
+
+        warmup = min(len(self.pp_ranks) - self.idx - 1, n)
+        for _ in range(warmup):
+            do_forward()
+        for _ in range(n - warmup):
+            do_forward()
+            do_backward()
+        for _ in range(warmup):
+            do_backward()
+        self._drain_sends()
+        if losses:
+            return torch.stack(losses).mean()
+        return None
+
+    def _forward_backward_gpipe(self, microbatches, loss_fn, hidden_dtype=torch.float32):
         stage = self.stage
         H = stage.config.hidden_size
         device = next(stage.parameters()).device
@@ -204,6 +286,7 @@ class PipelineRunner:
             if not stage.is_first:
                 self._send(h_in.grad, self.prev)
 
+        self._drain_sends()
         if losses:
             return torch.stack([l.detach() for l in losses]).mean()
         return None
