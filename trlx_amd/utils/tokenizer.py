@@ -206,10 +206,13 @@ class SyntheticVocabTokenizer(ByteTokenizer):
         if isinstance(ids, torch.Tensor):
             ids = ids.tolist()
         tab = self._pieces()
+        n = len(tab)
         if skip_special_tokens:
-            return " ".join([tab[t] for t in ids if t >= 3])
+            # ids beyond the table (model vocab > tokenizer vocab) still decode
+            return " ".join([tab[t] if t < n else "t%d" % t for t in ids if t >= 3])
         specials = [self.bos_token, self.eos_token, self.pad_token]
-        return " ".join([tab[t] if t >= 3 else specials[t] for t in ids])
+        return " ".join([(tab[t] if t < n else "t%d" % t) if t >= 3 else specials[t]
+                         for t in ids])
 
 
 def get_tokenizer(path: str, padding_side: str = "left", truncation_side: str = "right", **kwargs):
