@@ -51,6 +51,11 @@ at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
                           const at::Tensor& labels);
 at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
                        const c10::optional<at::Tensor>& bias, long act);
+std::vector<at::Tensor> lm_logprobs_v2_with_lse(const at::Tensor& hidden,
+                                                const at::Tensor& weight,
+                                                const at::Tensor& labels, bool want_lse);
+at::Tensor ce_dlogits(const at::Tensor& hidden, const at::Tensor& weight,
+                      const at::Tensor& labels, const at::Tensor& lse, const at::Tensor& dlp);
 at::Tensor lm_logprobs(const at::Tensor& hidden, const at::Tensor& weight,
                        const at::Tensor& labels);
 void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
@@ -80,4 +85,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("lm_logprobs", &lm_logprobs);
   mod.def("lm_logprobs_v2", &lm_logprobs_v2);
   mod.def("skinny_gemm", &skinny_gemm);
+  mod.def("lm_logprobs_v2_with_lse", &lm_logprobs_v2_with_lse);
+  mod.def("ce_dlogits", &ce_dlogits);
 }

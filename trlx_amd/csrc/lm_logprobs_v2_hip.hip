@@ -317,7 +317,8 @@ __global__ __launch_bounds__(BLOCK2) void lm_logprobs_v2_kernel(
 
 __global__ void lm_logprobs_v2_reduce(const float* __restrict__ partials,
                                       const float* __restrict__ label_logit,
-                                      float* __restrict__ out, int N, int nV) {
+                                      float* __restrict__ out, float* __restrict__ lse_out,
+                                      int N, int nV) {
   const int wpb = blockDim.x / WAVE;
   const int n = blockIdx.x * wpb + threadIdx.x / WAVE;
   if (n >= N) return;
@@ -345,15 +346,29 @@ __global__ void lm_logprobs_v2_reduce(const float* __restrict__ partials,
       ms.s += o.s * __expf(o.m - ms.m);
     }
   }
-  if (lane == 0) out[n] = label_logit[n] - (ms.m + logf(ms.s));
+  if (lane == 0) {
+    const float lse = ms.m + logf(ms.s);
+    out[n] = label_logit[n] - lse;
+    if (lse_out) lse_out[n] = lse;
+  }
 }
 
 constexpr int V2_LDS = 4 * TILE_BYTES + 8192 + BM2 * (int)sizeof(long);  // 139264
 
 }  // namespace
 
+std::vector<at::Tensor> lm_logprobs_v2_with_lse(const at::Tensor& hidden,
+                                                const at::Tensor& weight,
+                                                const at::Tensor& labels, bool want_lse);
+
 at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
                           const at::Tensor& labels) {
+  return lm_logprobs_v2_with_lse(hidden, weight, labels, false)[0];
+}
+
+std::vector<at::Tensor> lm_logprobs_v2_with_lse(const at::Tensor& hidden,
+                                                const at::Tensor& weight,
+                                                const at::Tensor& labels, bool want_lse) {
   const int mode = [] {
     const char* e = getenv("TRLX_AMD_LMLP_MODE");
     return e ? atoi(e) : 0;
@@ -368,7 +383,9 @@ at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
   TORCH_CHECK(weight.size(1) == H && labels.numel() == N);
   TORCH_CHECK(H % BK2 == 0, "lm_logprobs_v2: hidden size must be a multiple of 64");
   auto out = at::empty({N}, hidden.options().dtype(at::kFloat));
-  if (N == 0) return out;
+  auto lse = want_lse ? at::empty({N}, hidden.options().dtype(at::kFloat))
+                      : at::empty({0}, hidden.options().dtype(at::kFloat));
+  if (N == 0) return {out, lse};
   const int nV = (V + BN2 - 1) / BN2;
   const int nM = (N + BM2 - 1) / BM2;
   auto partials = at::empty({nV, (long)N, 2}, hidden.options().dtype(at::kFloat));
@@ -399,7 +416,8 @@ at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
         label_logit.data_ptr<float>(), labels.data_ptr<long>(), N, H, V, nV, nM, mode);
   const int wpb = 256 / WAVE;
  hipLaunchKernelGGL(( lm_logprobs_v2_reduce), dim3((N + wpb - 1) / wpb), dim3(256), 0, stream, 
-      partials.data_ptr<float>(), label_logit.data_ptr<float>(), out.data_ptr<float>(), N, nV);
+      partials.data_ptr<float>(), label_logit.data_ptr<float>(), out.data_ptr<float>(),
+      want_lse ? lse.data_ptr<float>() : nullptr, N, nV);
   HIP_CHECK_LAST();
-  return out;
+  return {out, lse};
 }

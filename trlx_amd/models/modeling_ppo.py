@@ -215,11 +215,13 @@ class AutoModelForCausalLMWithValueHead(PreTrainedModelWrapper):
         position_ids: Optional[torch.Tensor] = None,
         return_ref_logits: bool = False,
         logits_slice=None,
+        return_logits: bool = True,
         **kwargs,
     ) -> CausalLMOutputWithValue:
         out = self.base_model(
             input_ids, attention_mask=attention_mask, position_ids=position_ids,
             logits_slice=logits_slice, hidden_at_layer=self._value_stash() or None,
+            return_logits=return_logits,
         )
         if self.v_branch is not None:
             values = self._branch_values(out, input_ids, attention_mask, position_ids,
@@ -358,6 +360,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         position_ids: Optional[torch.Tensor] = None,
         return_ref_logits: bool = False,
         logits_slice=None,
+        return_logits: bool = True,
         **kwargs,
     ) -> CausalLMOutputWithValue:
         """One trunk pass; optionally also the reference logits via the frozen
@@ -369,6 +372,7 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         out = self.base_model(
             input_ids, attention_mask=attention_mask, position_ids=position_ids,
             hidden_at_layer=stash or None, logits_slice=logits_slice,
+            return_logits=return_logits,
         )
         if self.v_branch is not None:
             values = self._branch_values(out, input_ids, attention_mask, position_ids,

@@ -115,6 +115,48 @@ def lm_logprobs(hidden: torch.Tensor, weight: torch.Tensor, labels: torch.Tensor
     return reference.logprobs_of_labels(logits, labels)
 
 
+class _LMLogprobsTrain(torch.autograd.Function):
+    """Training-path fused lm_head+logprobs: forward via the pipelined MFMA
+    kernel (logits never materialize, saves only the fp32 logsumexp);
+    backward recomputes the logits tile-by-tile and emits dlogits directly
+    (csrc/ce_backward.hip), then contracts with W / hidden on hipBLASLt for
+    dh and dW."""
+
+    @staticmethod
+    def forward(ctx, hidden, weight, labels, ext):
+        out, lse = ext.lm_logprobs_v2_with_lse(hidden, weight, labels, True)
+        ctx.save_for_backward(hidden, weight, labels, lse)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        hidden, weight, labels, lse = ctx.saved_tensors
+        ext = _require_ext("lm_logprobs_train")
+        dlogits = ext.ce_dlogits(hidden, weight, labels, lse,
+                                 grad_out.contiguous().float())
+        dh = dlogits @ weight
+        dw = dlogits.t() @ hidden if ctx.needs_input_grad[1] else None
+        return dh, dw, None, None
+
+
+def lm_logprobs_train(hidden: torch.Tensor, weight: torch.Tensor,
+                      labels: torch.Tensor) -> torch.Tensor:
+    """Differentiable log p(labels) from final hidden states + lm_head weight.
+
+    GPU bf16 (H % 64 == 0): fused kernels, the [N, V] logits exist only as
+    the bf16 dlogits of the backward.  Otherwise: plain logits path with
+    the same math."""
+    if (hidden.is_cuda and hidden.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16 and hidden.shape[-1] % 64 == 0
+            and os.environ.get("TRLX_AMD_FUSED_LM_LOGPROBS") != "0"):
+        ext = _require_ext("lm_logprobs_train")
+        if ext is not None and hasattr(ext, "ce_dlogits"):
+            return _LMLogprobsTrain.apply(hidden.contiguous(), weight.contiguous(),
+                                          labels.contiguous(), ext)
+    logits = torch.nn.functional.linear(hidden, weight)
+    return logprobs_of_labels(logits, labels)
+
+
 ACT_CODES = {"none": 0, "gelu": 1, "gelu_new": 2, "relu": 3, "silu": 4}
 _ACT_FNS = {
     0: lambda y: y,

@@ -25,6 +25,7 @@ from torch.nn.utils.rnn import pad_sequence
 
 from ..data.configs import TRLConfig
 from ..data.ppo_types import PPORLBatch, PPORLElement
+from .. import ops
 from ..models.modeling_ppo import (
     AdaptiveKLController,
     AutoModelForCausalLMWithHydraValueHead,
@@ -166,8 +167,21 @@ class PPOTrainer(NativeRLTrainer):
         # + slice (accelerate_ppo_trainer.py:178-192), ~2.5x less head work
         start = query_tensors.shape[1] - 1
         end = start + response_length
-        outputs = self.model(tokens, attention_mask, logits_slice=(start, end))
-        logprobs = logprobs_of_labels(outputs.logits, tokens[:, start + 1 : end + 1])
+        labels_sl = tokens[:, start + 1 : end + 1]
+        lm = getattr(self.model.base_model, "lm_head", None)
+        if (lm is not None and lm.bias is None and tokens.is_cuda
+                and lm.weight.dtype == torch.bfloat16 and lm.weight.shape[1] % 64 == 0):
+            # fused train path: logprobs straight from hidden states; the
+            # [N, V] logits exist only as the bf16 dlogits of the backward
+            outputs = self.model(tokens, attention_mask, logits_slice=(start, end),
+                                 return_logits=False)
+            h = outputs.last_hidden_state[:, start:end].contiguous()
+            logprobs = ops.lm_logprobs_train(
+                h.reshape(-1, h.shape[-1]), lm.weight, labels_sl.reshape(-1)
+            ).view(labels_sl.shape)
+        else:
+            outputs = self.model(tokens, attention_mask, logits_slice=(start, end))
+            logprobs = logprobs_of_labels(outputs.logits, labels_sl)
         values_pred = outputs.values
         mask = attention_mask[:, start + 1 : end + 1]
 
