@@ -169,10 +169,18 @@ class PPOTrainer(NativeRLTrainer):
         end = start + response_length
         labels_sl = tokens[:, start + 1 : end + 1]
         lm = getattr(self.model.base_model, "lm_head", None)
-        if (lm is not None and lm.bias is None and tokens.is_cuda
+        if (os.environ.get("TRLX_AMD_FUSED_CE") == "1"
+                and lm is not None and lm.bias is None and tokens.is_cuda
                 and lm.weight.dtype == torch.bfloat16 and lm.weight.shape[1] % 64 == 0):
-            # fused train path: logprobs straight from hidden states; the
-            # [N, V] logits exist only as the bf16 dlogits of the backward
+            # OPT-IN fused train path: logprobs straight from hidden states;
+            # the [N, V] logits exist only as the bf16 dlogits of the
+            # backward.  Numerically verified, but the backward RECOMPUTES
+            # the logits GEMM where the stock path just re-reads the saved
+            # bf16 logits (264 MB round trip ~= 42 us vs ~180 us recompute at
+            # N=1312): same-box A/B measured 715 vs 730 samples/s, so the
+            # stock path stays the default at GPT-2 scale.  The fused path
+            # wins when activation memory is the constraint (logits for a
+            # [B, T, 50k+] slice can dominate at long response lengths).
             outputs = self.model(tokens, attention_mask, logits_slice=(start, end),
                                  return_logits=False)
             h = outputs.last_hidden_state[:, start:end].contiguous()
