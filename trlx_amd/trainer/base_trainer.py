@@ -588,8 +588,12 @@ class NativeRLTrainer(BaseRLTrainer):
                                     self.save(directory)
                                 self.save_pretrained(os.path.join(directory, "hf_model"))
 
-                    desc = " | ".join(f"{k}: {v:.2f}" for k, v in stats.items() if k.startswith("loss"))
-                    tbar.set_description(f"[{desc}]")
+                    if self.iter_count % 10 == 1 or self.iter_count >= self.total_steps:
+                        # formatting loss stats forces a device sync; do it
+                        # every few steps, not every step
+                        desc = " | ".join(f"{k}: {v:.2f}" for k, v in stats.items()
+                                          if k.startswith("loss"))
+                        tbar.set_description(f"[{desc}]")
                     tbar.update()
                     self.tracker.log(stats, step=self.iter_count)
 
