@@ -54,10 +54,8 @@ __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* _
 
   for (int sbase = kstart + wid * KPW; sbase < len; sbase += NWAVES * KPW) {
     const int key = sbase + kgrp;
-    float score = -INFINITY;
-    float vf[8];
     if (key < len) {
-      float kf[8];
+      float kf[8], vf[8];
       load8<bf16_t>(kbase + (size_t)key * D + d0, kf);
       load8<bf16_t>(vbase + (size_t)key * D + d0, vf);
       float partial = 0.f;
@@ -66,20 +64,15 @@ __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* _
       // reduce across the G lanes of this key group (contiguous lanes)
 #pragma unroll
       for (int off = G / 2; off > 0; off >>= 1) partial += __shfl_xor(partial, off);
-      score = partial;
-    }
-    // online softmax update for this group's accumulator
-    if (score > m) {
-      const float corr = expf(m - score);
-      s = s * corr + 1.f;
+      const float score = partial;
+      // branchless online update (first valid key: m=-inf -> corr=0)
+      const float mnew = fmaxf(m, score);
+      const float corr = (m > -INFINITY) ? __expf(m - mnew) : 0.f;
+      const float pw = __expf(score - mnew);
+      s = s * corr + pw;
 #pragma unroll
-      for (int i = 0; i < 8; ++i) o[i] = o[i] * corr + vf[i];
-      m = score;
-    } else if (score != -INFINITY) {
-      const float p = expf(score - m);
-      s += p;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) o[i] += p * vf[i];
+      for (int i = 0; i < 8; ++i) o[i] = o[i] * corr + pw * vf[i];
+      m = mnew;
     }
   }
 
@@ -105,7 +98,7 @@ __global__ void attn_decode_kernel(const bf16_t* __restrict__ q, const bf16_t* _
     for (int p = 0; p < NPART; ++p) {
       const float mp = ms_buf[p][0];
       if (mp == -INFINITY) continue;
-      const float w = expf(mp - mstar);
+      const float w = __expf(mp - mstar);
       sstar += ms_buf[p][1] * w;
       acc += o_buf[p][threadIdx.x] * w;
     }
