@@ -57,3 +57,14 @@ except Exception as e:
 wq = (torch.randn(2304, HID, device=dev) * 0.02).bfloat16()
 f = lambda: torch.nn.functional.linear(x, wq)
 print(f"qkv blaslt     standalone {t(f):6.2f}us  graphed {tg(f):6.2f}us (in-bench-graph ~9.3)")
+
+# fc_in + gelu: skinny(fused) vs blaslt+gelu, both GRAPHED (in-graph algos differ)
+wf = (torch.randn(3072, HID, device=dev) * 0.02).bfloat16()
+bf = (torch.randn(3072, device=dev) * 0.1).bfloat16()
+f_blaslt = lambda: torch.nn.functional.gelu(torch.nn.functional.linear(x, wf, bf), approximate="tanh")
+f_skinny = lambda: ext.skinny_gemm(x, wf, bf, 2)
+print(f"fc_in+gelu blaslt graphed {tg(f_blaslt):6.2f}us | skinny graphed {tg(f_skinny):6.2f}us")
+wo = (torch.randn(HID, HID, device=dev) * 0.02).bfloat16()
+f_blaslt2 = lambda: torch.nn.functional.linear(x, wo)
+f_skinny2 = lambda: ext.skinny_gemm(x, wo, None, 0)
+print(f"o_proj blaslt graphed {tg(f_blaslt2):6.2f}us | skinny graphed {tg(f_skinny2):6.2f}us")

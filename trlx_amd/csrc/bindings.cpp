@@ -56,6 +56,10 @@ std::vector<at::Tensor> lm_logprobs_v2_with_lse(const at::Tensor& hidden,
                                                 const at::Tensor& labels, bool want_lse);
 at::Tensor ce_dlogits(const at::Tensor& hidden, const at::Tensor& weight,
                       const at::Tensor& labels, const at::Tensor& lse, const at::Tensor& dlp);
+void decode_advance(const at::Tensor& tok, at::Tensor& out_tokens, at::Tensor& cur_tok,
+                    at::Tensor& finished, at::Tensor& rng_offset, at::Tensor& step_col,
+                    at::Tensor& cache_idx, at::Tensor& seq_lens, at::Tensor& pos_ids,
+                    const c10::optional<at::Tensor>& key_starts, long eos, long pad);
 at::Tensor lm_logprobs(const at::Tensor& hidden, const at::Tensor& weight,
                        const at::Tensor& labels);
 void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
@@ -87,4 +91,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("skinny_gemm", &skinny_gemm);
   mod.def("lm_logprobs_v2_with_lse", &lm_logprobs_v2_with_lse);
   mod.def("ce_dlogits", &ce_dlogits);
+  mod.def("decode_advance", &decode_advance);
 }

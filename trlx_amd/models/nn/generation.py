@@ -72,6 +72,7 @@ class DecodeEngine:
         self.cur_tok = torch.zeros(batch, 1, dtype=torch.long, device=device)
         self.key_starts = torch.zeros(batch, dtype=torch.int32, device=device)
         self.seq_lens = torch.zeros(batch, dtype=torch.int32, device=device)
+        self.pos_ids = torch.zeros(batch, 1, dtype=torch.int32, device=device)
         self.cache_idx = torch.zeros(1, dtype=torch.long, device=device)
         self.rng_offset = torch.zeros(1, dtype=torch.long, device=device)
         self.step_col = torch.zeros(1, dtype=torch.long, device=device)
@@ -92,7 +93,22 @@ class DecodeEngine:
                                 seed=self.seed, offset=self.rng_offset)
 
     def _advance(self, tok):
-        """Post-sampling state updates (shared by graph step and prefill)."""
+        """Post-sampling state updates (shared by graph step and prefill).
+
+        One fused HIP kernel (csrc/decode_advance.hip) replaces ~11
+        elementwise launches of bookkeeping per token; it also pre-computes
+        the NEXT step's position ids and bumps seq_lens/cache_idx, so the
+        step body is model + sample + advance and nothing else.  The torch
+        fallback reproduces the same semantics op by op."""
+        ext = ops._load_ext()
+        if ext is not None and hasattr(ext, "decode_advance") and tok.is_cuda:
+            ext.decode_advance(
+                tok.contiguous(), self.out_tokens, self.cur_tok.view(-1), self.finished,
+                self.rng_offset, self.step_col, self.cache_idx, self.seq_lens,
+                self.pos_ids.view(-1), self.key_starts,
+                -1 if self.gen.eos_token_id is None else self.gen.eos_token_id, self.pad_id,
+            )
+            return
         if self.gen.eos_token_id is not None:
             tok = torch.where(self.finished, torch.full_like(tok, self.pad_id), tok)
             self.finished |= tok == self.gen.eos_token_id
@@ -100,13 +116,18 @@ class DecodeEngine:
         self.cur_tok.copy_(tok.unsqueeze(1))
         self.rng_offset += 1
         self.step_col += 1
+        self.cache_idx += 1
+        self.seq_lens += 1
+        self.pos_ids.copy_(
+            (self.cache_idx - self.key_starts.to(torch.long)).to(torch.int32).unsqueeze(1))
 
     def _step(self):
-        """One decode token — everything device-side (hipGraph body)."""
-        self.seq_lens += 1
-        pos_ids = (self.cache_idx - self.key_starts.to(torch.long)).to(torch.int32).unsqueeze(1)
+        """One decode token — everything device-side (hipGraph body).
+
+        State convention: on entry, seq_lens/cache_idx/pos_ids already
+        describe THIS step (the previous _advance set them)."""
         out = self.model(
-            self.cur_tok, kv_cache=self.kv, position_ids=pos_ids, seq_lens=self.seq_lens,
+            self.cur_tok, kv_cache=self.kv, position_ids=self.pos_ids, seq_lens=self.seq_lens,
             key_starts=self.key_starts, cache_idx=self.cache_idx, start_pos=0,
             return_logits=False,
         )
@@ -115,14 +136,14 @@ class DecodeEngine:
             logits = self.shaping_fn(logits, out.last_hidden_state[:, -1], self.cur_tok[:, 0])
         tok = self._sample(logits)
         self._advance(tok)
-        self.cache_idx += 1
 
     def _scratch_state(self):
         """In-bounds dummy state for warmup/capture runs (cache row 0 gets
         scribbled; the real prefill overwrites it afterwards)."""
         self.step_col.zero_()
         self.cache_idx.zero_()
-        self.seq_lens.zero_()
+        self.seq_lens.fill_(1)
+        self.pos_ids.zero_()
         self.rng_offset.zero_()
         self.finished.zero_()
 
@@ -147,10 +168,14 @@ class DecodeEngine:
         if self.graph is None and max_new_tokens > 1:
             self.capture()
 
-        # reset per-call state
+        # reset per-call state.  The fused advance will bump cache_idx/
+        # seq_lens and derive pos_ids for the first replay, so they start
+        # one step "behind": after the prefill-token advance the first
+        # replay sees seq_lens=T+1, cache_idx=T, pos=T-key_starts (the same
+        # values the pre-fusion engine computed inside its step).
         self.key_starts.copy_((T - attention_mask.sum(-1)).to(torch.int32))
         self.seq_lens.fill_(T)
-        self.cache_idx.fill_(T)
+        self.cache_idx.fill_(T - 1)
         self.step_col.zero_()
         self.finished.zero_()
         self.out_tokens.fill_(self.pad_id)
