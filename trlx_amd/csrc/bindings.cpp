@@ -56,6 +56,13 @@ std::vector<at::Tensor> lm_logprobs_v2_with_lse(const at::Tensor& hidden,
                                                 const at::Tensor& labels, bool want_lse);
 at::Tensor ce_dlogits(const at::Tensor& hidden, const at::Tensor& weight,
                       const at::Tensor& labels, const at::Tensor& lse, const at::Tensor& dlp);
+at::Tensor fused_decode_attention(const at::Tensor& qkv, at::Tensor& kcache, at::Tensor& vcache,
+                                  const at::Tensor& seq_lens,
+                                  const c10::optional<at::Tensor>& seq_starts,
+                                  const c10::optional<at::Tensor>& cos,
+                                  const c10::optional<at::Tensor>& sin,
+                                  const at::Tensor& cache_idx, long rot, bool interleaved,
+                                  double scale);
 void decode_advance(const at::Tensor& tok, at::Tensor& out_tokens, at::Tensor& cur_tok,
                     at::Tensor& finished, at::Tensor& rng_offset, at::Tensor& step_col,
                     at::Tensor& cache_idx, at::Tensor& seq_lens, at::Tensor& pos_ids,
@@ -92,4 +99,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("lm_logprobs_v2_with_lse", &lm_logprobs_v2_with_lse);
   mod.def("ce_dlogits", &ce_dlogits);
   mod.def("decode_advance", &decode_advance);
+  mod.def("fused_decode_attention", &fused_decode_attention);
 }

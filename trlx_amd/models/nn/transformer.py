@@ -151,13 +151,23 @@ class Attention(nn.Module):
         if (T == 1 and kv_cache is not None and ctx.cache_idx is not None
                 and ctx.seq_lens is not None and x.is_cuda and ctx.alibi is None):
             cos_sin = rope_tables if self.cfg.position_encoding == "rope" else (None, None)
-            q = ops.decode_prep(
-                qkv, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx], ctx.cache_idx,
-                self.num_heads, cos_sin[0], cos_sin[1], ctx.key_starts, self.rot,
-                self.cfg.rope_interleaved,
-            )
-            out = ops.attention_decode(q, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx],
-                                       ctx.seq_lens, self.scale, seq_starts=ctx.key_starts)
+            if self.num_heads == self.num_kv_heads:
+                # MHA: split+RoPE+append and attention in ONE kernel (the
+                # per-head k/v append and the attention read are same-block)
+                out = ops.fused_decode_attention(
+                    qkv, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx],
+                    ctx.seq_lens, ctx.key_starts, cos_sin[0], cos_sin[1], ctx.cache_idx,
+                    self.rot, self.cfg.rope_interleaved, self.scale,
+                )
+            else:
+                q = ops.decode_prep(
+                    qkv, kv_cache.k[self.layer_idx], kv_cache.v[self.layer_idx], ctx.cache_idx,
+                    self.num_heads, cos_sin[0], cos_sin[1], ctx.key_starts, self.rot,
+                    self.cfg.rope_interleaved,
+                )
+                out = ops.attention_decode(q, kv_cache.k[self.layer_idx],
+                                           kv_cache.v[self.layer_idx],
+                                           ctx.seq_lens, self.scale, seq_starts=ctx.key_starts)
             return dense(self.o_proj, out.transpose(1, 2).reshape(B, T, -1))
 
         if x.is_cuda and x.dtype == torch.bfloat16 and ops.extension_available():

@@ -476,6 +476,29 @@ def decode_prep(
                            num_heads, rot, interleaved)
 
 
+def fused_decode_attention(
+    qkv: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    seq_lens: torch.Tensor,
+    key_starts: Optional[torch.Tensor],
+    cos: Optional[torch.Tensor],
+    sin: Optional[torch.Tensor],
+    cache_idx: torch.Tensor,
+    rot: int,
+    interleaved: bool,
+    scale: float,
+) -> torch.Tensor:
+    """MHA decode step in ONE kernel: split fused qkv, RoPE q/k, append k/v
+    to the cache at *cache_idx, flash-decode attention — out [B,H,1,D].
+    GPU-only (the eager path covers CPU); requires Hq == Hkv."""
+    ext = _require_ext("fused_decode_attention")
+    ks = key_starts.to(torch.int32).contiguous() if key_starts is not None else None
+    return ext.fused_decode_attention(qkv.contiguous(), k_cache, v_cache,
+                                      seq_lens.to(torch.int32).contiguous(), ks, cos, sin,
+                                      cache_idx, rot, interleaved, scale)
+
+
 class _QKVPrep(torch.autograd.Function):
     @staticmethod
     def forward(ctx, qkv, num_heads, num_kv_heads, head_dim, cos, sin, pos, qscale, rot,
