@@ -27,7 +27,8 @@ DEV unsigned int float_orderable(float x) {
   return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
 }
 
-__global__ void gumbel_sample_kernel(const float* __restrict__ logits,
+template <typename T>
+__global__ void gumbel_sample_kernel(const T* __restrict__ logits,
                                      const float* __restrict__ thresholds,
                                      unsigned long long* __restrict__ packed, int V, float invTemp,
                                      unsigned long long seed, const long* __restrict__ offset_ptr,
@@ -36,7 +37,7 @@ __global__ void gumbel_sample_kernel(const float* __restrict__ logits,
   const long row = blockIdx.y;
   const int lo = blockIdx.x * CHUNK;
   const int hi = min(lo + CHUNK, V);
-  const float* x = logits + (size_t)row * V;
+  const T* x = logits + (size_t)row * V;
   const float thr = thresholds ? thresholds[row] : -INFINITY;
   const unsigned long long off = offset_ptr ? (unsigned long long)(*offset_ptr) : (unsigned long long)host_offset;
   const unsigned long long key = splitmix64(seed ^ (0x9e3779b97f4a7c15ull * (off + 1)));
@@ -44,7 +45,7 @@ __global__ void gumbel_sample_kernel(const float* __restrict__ logits,
   float best = -INFINITY;
   int best_i = lo;
   for (int i = lo + threadIdx.x; i < hi; i += BLOCK) {
-    const float xi = x[i];
+    const float xi = ScalarIO<T>::load(x + i);
     if (xi < thr || xi == -INFINITY) continue;
     const float u = rng_uniform(key, (unsigned long long)row, (unsigned long long)i);
     const float g = -__logf(-__logf(u));
@@ -97,8 +98,8 @@ __global__ void unpack_kernel(const unsigned long long* __restrict__ packed, lon
 at::Tensor gumbel_sample_impl(const at::Tensor& logits, double temperature,
                               const c10::optional<at::Tensor>& thresholds, long seed,
                               const long* offset_ptr, long host_offset, hipStream_t stream) {
-  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 && logits.dtype() == at::kFloat &&
-              logits.is_contiguous());
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 && logits.is_contiguous() &&
+              (logits.dtype() == at::kFloat || logits.dtype() == at::kBFloat16));
   const long B = logits.size(0);
   const int V = logits.size(1);
   auto out = at::empty({B}, logits.options().dtype(at::kLong));
@@ -113,11 +114,18 @@ at::Tensor gumbel_sample_impl(const at::Tensor& logits, double temperature,
   }
   const int nchunks = (V + CHUNK - 1) / CHUNK;
   dim3 grid(nchunks, B);
-  gumbel_sample_kernel<<<grid, BLOCK, 0, stream>>>(
-      logits.data_ptr<float>(), thr,
-      reinterpret_cast<unsigned long long*>(packed.data_ptr<long>()), V,
-      1.0f / (float)std::max(temperature, 1e-6), (unsigned long long)seed, offset_ptr,
-      host_offset);
+  if (logits.dtype() == at::kBFloat16)
+    gumbel_sample_kernel<bf16_t><<<grid, BLOCK, 0, stream>>>(
+        reinterpret_cast<const bf16_t*>(logits.data_ptr()), thr,
+        reinterpret_cast<unsigned long long*>(packed.data_ptr<long>()), V,
+        1.0f / (float)std::max(temperature, 1e-6), (unsigned long long)seed, offset_ptr,
+        host_offset);
+  else
+    gumbel_sample_kernel<float><<<grid, BLOCK, 0, stream>>>(
+        logits.data_ptr<float>(), thr,
+        reinterpret_cast<unsigned long long*>(packed.data_ptr<long>()), V,
+        1.0f / (float)std::max(temperature, 1e-6), (unsigned long long)seed, offset_ptr,
+        host_offset);
   const int ub = (int)((B + 255) / 256);
   unpack_kernel<<<ub, 256, 0, stream>>>(
       reinterpret_cast<const unsigned long long*>(packed.data_ptr<long>()), out.data_ptr<long>(),

@@ -131,9 +131,12 @@ class DecodeEngine:
             key_starts=self.key_starts, cache_idx=self.cache_idx, start_pos=0,
             return_logits=False,
         )
-        logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
+        logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0]
         if self.shaping_fn is not None:
-            logits = self.shaping_fn(logits, out.last_hidden_state[:, -1], self.cur_tok[:, 0])
+            logits = self.shaping_fn(logits.float(), out.last_hidden_state[:, -1],
+                                     self.cur_tok[:, 0])
+        # bf16 logits go straight to the sampler (same values the fp32 cast
+        # would carry — the lm_head output is already bf16-rounded)
         tok = self._sample(logits)
         self._advance(tok)
 

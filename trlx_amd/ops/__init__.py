@@ -579,7 +579,9 @@ def sample_token(
                 if dev_offset:
                     return ext.gumbel_sample_dev(lg, 1.0, thr, seed, offset)
                 return ext.gumbel_sample(lg, 1.0, thr, seed, offset)
-            lg = logits.float().contiguous()
+            # fp32 and bf16 logits both sample natively (no cast kernel)
+            lg = logits.contiguous() if logits.dtype in (torch.float32, torch.bfloat16) \
+                else logits.float().contiguous()
             if dev_offset:
                 return ext.gumbel_sample_dev(lg, float(temperature), None, seed, offset)
             return ext.gumbel_sample(lg, float(temperature), None, seed, offset)
