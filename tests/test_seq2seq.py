@@ -165,3 +165,32 @@ def test_seq2seq_gpu_forward_and_generate():
     assert out.logits.isfinite().all()
     gen = m.generate(ids, mask, max_new_tokens=6, do_sample=True)
     assert gen.shape[0] == 2 and gen.shape[1] <= 7
+
+
+def test_seq2seq_value_branch():
+    """Trainable T5 value branch (num_value_layers_unfrozen > 0)."""
+    from trlx_amd.models.modeling_seq2seq import (
+        AutoModelForSeq2SeqLMWithHydraValueHead,
+        AutoModelForSeq2SeqLMWithValueHead,
+    )
+    from trlx_amd.models.nn.seq2seq import Seq2SeqConfig, Seq2SeqTransformer
+
+    torch.manual_seed(5)
+    cfg = Seq2SeqConfig(vocab_size=120, d_model=32, d_ff=64, num_heads=2,
+                        num_layers=2, num_decoder_layers=2)
+    base = Seq2SeqTransformer(cfg)
+    m = AutoModelForSeq2SeqLMWithValueHead(base, num_value_layers_unfrozen=1)
+    assert m.v_branch is not None
+    ids = torch.randint(3, 100, (2, 5))
+    dec = torch.randint(3, 100, (2, 4))
+    out = m(ids, torch.ones_like(ids), dec, torch.ones_like(dec))
+    assert out.values.shape == (2, 4)
+    out.values.sum().backward()
+    assert any(p.grad is not None and p.grad.abs().sum() > 0
+               for p in m.v_branch.blocks.parameters())
+
+    torch.manual_seed(5)
+    hm = AutoModelForSeq2SeqLMWithHydraValueHead(
+        Seq2SeqTransformer(cfg), num_layers_unfrozen=1, num_value_layers_unfrozen=1)
+    hout = hm(ids, torch.ones_like(ids), dec, torch.ones_like(dec), return_ref_logits=True)
+    assert hout.ref_logits is not None and hout.values.shape == (2, 4)

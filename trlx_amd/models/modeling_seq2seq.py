@@ -104,6 +104,44 @@ class Seq2SeqModelWrapper(nn.Module):
         return self.generate(input_ids, attention_mask, **kwargs)
 
 
+class T5ValueBranch(nn.Module):
+    """TRAINABLE value branch for seq2seq: copies of the top
+    ``num_value_layers_unfrozen`` decoder blocks + final norm + scalar head
+    (reference make_value_branch applied to the T5 branch class,
+    modeling_ppo.py:255-263/1483-1592)."""
+
+    def __init__(self, base: Seq2SeqTransformer, num_layers: int):
+        super().__init__()
+        self.cfg = base.config
+        self.num_layers = num_layers
+        self.blocks = nn.ModuleList(
+            copy.deepcopy(b) for b in base.decoder_blocks[-num_layers:]
+        )
+        self.final_norm = copy.deepcopy(base.decoder_final_norm)
+        self.v_head = make_head(self.cfg.d_model, 1, dtype=torch.float32)
+        for p in self.parameters():
+            p.requires_grad_(True)
+
+    def forward(self, hidden, enc_out, attention_mask, decoder_attention_mask,
+                position_bias, logits_slice=None):
+        from .nn.seq2seq import _extend_mask
+
+        B, T = hidden.shape[:2]
+        causal = torch.ones(T, T, device=hidden.device).tril()
+        self_mask = (1.0 - causal[None, None]) * torch.finfo(torch.float32).min
+        if decoder_attention_mask is not None:
+            self_mask = self_mask + _extend_mask(decoder_attention_mask)
+        cross_mask = _extend_mask(attention_mask)
+        h = hidden
+        for block in self.blocks:
+            h, position_bias, _, _ = block(h, enc_out=enc_out, self_mask=self_mask,
+                                           cross_mask=cross_mask, position_bias=position_bias)
+        h = self.final_norm(h)
+        if logits_slice is not None:
+            h = h[:, logits_slice[0] : logits_slice[1]]
+        return self.v_head(h.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+
+
 class AutoModelForSeq2SeqLMWithValueHead(Seq2SeqModelWrapper):
     """Seq2seq LM + scalar value head on decoder hidden states
     (reference modeling_ppo.py:1242-1350)."""
@@ -111,11 +149,30 @@ class AutoModelForSeq2SeqLMWithValueHead(Seq2SeqModelWrapper):
     def __init__(self, base_model, peft_config=None, num_value_layers_unfrozen: int = 0):
         super().__init__(base_model)
         self.peft_config = peft_config
+        self.num_value_layers_unfrozen = num_value_layers_unfrozen
         self.v_head = make_head(self.config.d_model, 1, dtype=torch.float32)
+        self.v_branch = None
+        if num_value_layers_unfrozen > 0:
+            self.v_branch = T5ValueBranch(base_model, num_value_layers_unfrozen)
+
+    def _value_bias(self, T, device):
+        return self.base_model.decoder_blocks[0].self_attn.compute_bias(T, T, device)
 
     def forward(self, input_ids, attention_mask=None, decoder_input_ids=None,
                 decoder_attention_mask=None, return_ref_logits: bool = False,
                 logits_slice=None, **kwargs):
+        if self.v_branch is not None:
+            enc = self.base_model.encode(input_ids, attention_mask)
+            dec, _, hidden_at = self.base_model.decode(
+                decoder_input_ids, enc, attention_mask, decoder_attention_mask,
+                hidden_at_layer=-self.num_value_layers_unfrozen)
+            ds = dec if logits_slice is None else dec[:, logits_slice[0] : logits_slice[1]]
+            logits = self.base_model.project(ds)
+            values = self.v_branch(hidden_at, enc, attention_mask, decoder_attention_mask,
+                                   self._value_bias(decoder_input_ids.shape[1], dec.device),
+                                   logits_slice=logits_slice)
+            return CausalLMOutputWithValue(logits=logits, values=values,
+                                           last_hidden_state=dec)
         out = self.base_model(input_ids, attention_mask, decoder_input_ids,
                               decoder_attention_mask, logits_slice=logits_slice)
         hs = out.last_hidden_state
@@ -171,7 +228,8 @@ class AutoModelForSeq2SeqLMWithHydraValueHead(AutoModelForSeq2SeqLMWithValueHead
 
     def __init__(self, base_model, peft_config=None, num_layers_unfrozen: int = -1,
                  num_value_layers_unfrozen: int = 0):
-        super().__init__(base_model, peft_config=peft_config)
+        super().__init__(base_model, peft_config=peft_config,
+                         num_value_layers_unfrozen=num_value_layers_unfrozen)
         self.num_layers_unfrozen = num_layers_unfrozen
         self.frozen_head = None
         if num_layers_unfrozen > 0 and peft_config is None:
@@ -180,21 +238,32 @@ class AutoModelForSeq2SeqLMWithHydraValueHead(AutoModelForSeq2SeqLMWithValueHead
     def forward(self, input_ids, attention_mask=None, decoder_input_ids=None,
                 decoder_attention_mask=None, return_ref_logits: bool = False,
                 logits_slice=None, **kwargs):
-        stash = -self.num_layers_unfrozen if (return_ref_logits and self.frozen_head) else None
+        stash = []
+        if return_ref_logits and self.frozen_head is not None:
+            stash.append(-self.num_layers_unfrozen)
+        if self.v_branch is not None:
+            stash.append(-self.num_value_layers_unfrozen)
         enc = self.base_model.encode(input_ids, attention_mask)
         dec, _, hidden_at = self.base_model.decode(
             decoder_input_ids, enc, attention_mask, decoder_attention_mask,
-            hidden_at_layer=stash)
+            hidden_at_layer=stash or None)
         ds = dec if logits_slice is None else dec[:, logits_slice[0] : logits_slice[1]]
         logits = self.base_model.project(ds)
-        values = self.v_head(ds.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
+        T = decoder_input_ids.shape[1]
+        if self.v_branch is not None:
+            values = self.v_branch(hidden_at[-self.num_value_layers_unfrozen], enc,
+                                   attention_mask, decoder_attention_mask,
+                                   self._value_bias(T, dec.device), logits_slice=logits_slice)
+        else:
+            values = self.v_head(ds.to(self.v_head[0].weight.dtype)).squeeze(-1).float()
         ref_logits = None
         if return_ref_logits and self.frozen_head is not None:
             # recompute the first unfrozen block's incoming position bias
             bias_layer = self.base_model.decoder_blocks[0].self_attn
-            T = decoder_input_ids.shape[1]
             position_bias = bias_layer.compute_bias(T, T, dec.device)
-            ref_logits = self.frozen_head(hidden_at, enc, attention_mask,
+            ref_hidden = hidden_at[-self.num_layers_unfrozen] if isinstance(hidden_at, dict) \
+                else hidden_at
+            ref_logits = self.frozen_head(ref_hidden, enc, attention_mask,
                                           decoder_attention_mask, position_bias,
                                           logits_slice=logits_slice)
         return CausalLMOutputWithValue(logits=logits, values=values, ref_logits=ref_logits,

@@ -276,12 +276,22 @@ class Seq2SeqTransformer(nn.Module):
 
         bias = None
         new_past = []
-        hidden_at = None
         n = len(self.decoder_blocks)
-        stash_at = hidden_at_layer % n if hidden_at_layer is not None else None
+        # int -> single stash (tensor); list -> dict keyed by requested values
+        multi = isinstance(hidden_at_layer, (list, tuple))
+        wanted = list(hidden_at_layer) if multi else (
+            [hidden_at_layer] if hidden_at_layer is not None else [])
+        stash_for = {}
+        for w in wanted:
+            stash_for.setdefault(w % n, []).append(w)
+        hidden_at = {} if multi else None
         for i, block in enumerate(self.decoder_blocks):
-            if stash_at is not None and i == stash_at:
-                hidden_at = h
+            if i in stash_for:
+                if multi:
+                    for w in stash_for[i]:
+                        hidden_at[w] = h
+                else:
+                    hidden_at = h
             p_self = past[i][0] if past is not None else None
             p_cross = past[i][1] if past is not None else None
             h, bias, self_kv, cross_kv = block(
