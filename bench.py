@@ -14,7 +14,7 @@ collect num_rollouts experiences (generate + reward + logprob/value/ref pass)
 and run ppo_epochs optimization epochs over them — so samples/sec is the true
 end-to-end RLHF pipeline throughput, nothing skipped.
 
-Data: synthetic prompts over a 50257-token synthetic vocabulary with perfect
+Data: synthetic prompts over a model-vocab-sized synthetic vocabulary with perfect
 decode/encode round-trip; weights: random-init GPT-2-small; reward: cheap
 deterministic function of the sample string on rank 0 (scatter to ranks, as
 the real protocol does).
@@ -40,9 +40,13 @@ def build_trainer(args):
 
     config = default_ppo_config()
     config.model.model_path = args.model
-    config.model.model_extra_configs = {"config": preset(args.model).to_dict()}
+    model_cfg = preset(args.model)
+    config.model.model_extra_configs = {"config": model_cfg.to_dict()}
     config.model.num_layers_unfrozen = args.num_layers_unfrozen
-    config.tokenizer.tokenizer_path = "synthetic"
+    # tokenizer vocab MUST match the model vocab: sampled/synthetic ids at or
+    # above vocab_size index out of the embedding (llama V=32000 < the
+    # default 50257 -> device memory fault)
+    config.tokenizer.tokenizer_path = f"synthetic:{model_cfg.vocab_size}"
     config.train.seq_length = args.seq_len
     config.train.batch_size = args.batch_size
     config.train.total_steps = 10**9
@@ -64,7 +68,8 @@ def build_trainer(args):
     trainer = get_trainer(config.train.trainer)(config=config, reward_fn=reward_fn)
 
     torch.manual_seed(1234 + int(os.environ.get("RANK", 0)))
-    prompt_tokens = torch.randint(3, 50257, (args.num_prompts, args.prompt_len)).tolist()
+    prompt_tokens = torch.randint(3, model_cfg.vocab_size,
+                                  (args.num_prompts, args.prompt_len)).tolist()
     prompts = [" ".join(f"t{t}" for t in row) for row in prompt_tokens]
     pipeline = PromptPipeline(prompts, args.prompt_len + 2, trainer.tokenizer)
     trainer.add_prompt_pipeline(pipeline)
@@ -180,7 +185,7 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16" if torch.cuda.is_available() else "fp32",
-            "data": "synthetic prompts (vocab 50257), random-init weights, deterministic synthetic reward",
+            "data": "synthetic prompts (model-sized synthetic vocab), random-init weights, deterministic synthetic reward",
             "config": {
                 "model": args.model,
                 "global_batch": args.batch_size * world,
@@ -208,7 +213,7 @@ def run_ilql(args):
     config = default_ilql_config()
     config.model.model_path = args.model
     config.model.model_extra_configs = {"config": preset(args.model).to_dict()}
-    config.tokenizer.tokenizer_path = "synthetic"
+    config.tokenizer.tokenizer_path = f"synthetic:{preset(args.model).vocab_size}"
     config.train.seq_length = 64
     config.train.batch_size = 128
     config.train.tracker = None
@@ -217,7 +222,7 @@ def run_ilql(args):
 
     torch.manual_seed(1234 + rank)
     # synthetic reward-labeled samples at the canonical ILQL shape (seq 64)
-    toks = torch.randint(3, 50257, (512, 60)).tolist()
+    toks = torch.randint(3, preset(args.model).vocab_size, (512, 60)).tolist()
     samples = [" ".join(f"t{t}" for t in row) for row in toks]
     rewards = [((i * 2654435761) % 1000) / 1000.0 - 0.5 for i in range(len(samples))]
     trainer.make_experience(samples, rewards, config.train.seq_length)

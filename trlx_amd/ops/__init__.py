@@ -494,6 +494,8 @@ def fused_decode_attention(
     GPU-only (the eager path covers CPU); requires Hq == Hkv."""
     ext = _require_ext("fused_decode_attention")
     ks = key_starts.to(torch.int32).contiguous() if key_starts is not None else None
+    if cos is not None and cos.dtype != torch.float32:
+        cos, sin = cos.float(), sin.float()  # kernels require fp32 trig tables
     return ext.fused_decode_attention(qkv.contiguous(), k_cache, v_cache,
                                       seq_lens.to(torch.int32).contiguous(), ks, cos, sin,
                                       cache_idx, rot, interleaved, scale)
