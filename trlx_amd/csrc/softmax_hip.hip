@@ -138,6 +138,10 @@ template <typename T>
 __global__ void causal_softmax_fwd_wave(const T* __restrict__ scores, T* __restrict__ probs,
                                         const int* __restrict__ key_starts, int HTq, int Tq,
                                         int Tk, int start_pos, long rows) {
+  // single pass over the row: each lane caches its <=16 strided elements in
+  // registers (Tk <= 1024 guaranteed by the caller), so the normalize step
+  // re-reads nothing; fully unrolled so xv[] stays in VGPRs.  The previous
+  // two-pass scalar version measured 24.3 us (0.7 TB/s) at the bench shape.
   const int wpb = blockDim.x / WAVE;
   const long row0 = (long)blockIdx.x * wpb + threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
@@ -148,23 +152,35 @@ __global__ void causal_softmax_fwd_wave(const T* __restrict__ scores, T* __restr
     const int valid = min(Tk, start_pos + tq + 1);
     const T* xr = scores + (size_t)row * Tk;
     T* pr = probs + (size_t)row * Tk;
+    float xv[16];
     MS ms{-INFINITY, 0.f};
-    for (int i = kstart + lane; i < valid; i += WAVE) {
-      const float xi = ScalarIO<T>::load(xr + i);
-      if (xi > ms.m) {
-        ms.s = ms.s * expf(ms.m - xi) + 1.f;
-        ms.m = xi;
-      } else {
-        ms.s += expf(xi - ms.m);
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int i = lane + j * WAVE;
+      if (i < Tk) {
+        const float xi = ScalarIO<T>::load(xr + i);
+        xv[j] = xi;
+        if (i >= kstart && i < valid) {
+          if (xi > ms.m) {
+            ms.s = (ms.m > -INFINITY ? ms.s * __expf(ms.m - xi) : 0.f) + 1.f;
+            ms.m = xi;
+          } else {
+            ms.s += __expf(xi - ms.m);
+          }
+        }
       }
     }
     ms = wave_ms(ms);
     const float m = ms.m;
     const float rs = (ms.s > 0.f) ? 1.0f / ms.s : 0.f;
-    for (int i = lane; i < Tk; i += WAVE) {
-      float pv = 0.f;
-      if (i >= kstart && i < valid) pv = expf(ScalarIO<T>::load(xr + i) - m) * rs;
-      ScalarIO<T>::store(pr + i, pv);
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int i = lane + j * WAVE;
+      if (i < Tk) {
+        float pv = 0.f;
+        if (i >= kstart && i < valid) pv = __expf(xv[j] - m) * rs;
+        ScalarIO<T>::store(pr + i, pv);
+      }
     }
   }
 }
