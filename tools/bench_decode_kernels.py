@@ -64,6 +64,11 @@ bf = (torch.randn(3072, device=dev) * 0.1).bfloat16()
 f_blaslt = lambda: torch.nn.functional.gelu(torch.nn.functional.linear(x, wf, bf), approximate="tanh")
 f_skinny = lambda: ext.skinny_gemm(x, wf, bf, 2)
 print(f"fc_in+gelu blaslt graphed {tg(f_blaslt):6.2f}us | skinny graphed {tg(f_skinny):6.2f}us")
+wq = (torch.randn(2304, HID, device=dev) * 0.02).bfloat16()
+bq = (torch.randn(2304, device=dev) * 0.1).bfloat16()
+f_bl_q = lambda: torch.nn.functional.linear(x, wq, bq)
+f_sk_q = lambda: ext.skinny_gemm(x, wq, bq, 0)
+print(f"qkv blaslt graphed {tg(f_bl_q):6.2f}us | skinny graphed {tg(f_sk_q):6.2f}us")
 wo = (torch.randn(HID, HID, device=dev) * 0.02).bfloat16()
 f_blaslt2 = lambda: torch.nn.functional.linear(x, wo)
 f_skinny2 = lambda: ext.skinny_gemm(x, wo, None, 0)

@@ -174,9 +174,11 @@ at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
     bp = reinterpret_cast<const bf16_t*>(bc.data_ptr());
   }
   const int ntiles = ((M + 15) / 16) * ((N + 15) / 16);
-  // fill the chip: target >= 1024 waves (256 CUs x 4); cap splits at K/64
+  // fill the chip AND the memory pipeline: ~4 waves/CU only keeps ~0.5 KB
+  // of loads in flight per CU (13 us for a 3.5 MB weight read); target
+  // >= 4096 waves (16/CU) via K-splits so the latency-bound stream overlaps
   int nsplit = 1;
-  while (nsplit < 8 && ntiles * nsplit < 1024 && (K / 32) / (nsplit * 2) >= 2) nsplit *= 2;
+  while (nsplit < 8 && ntiles * nsplit < 4096 && (K / 32) / (nsplit * 2) >= 2) nsplit *= 2;
   auto stream = c10::hip::getCurrentHIPStream();
   const auto ap = reinterpret_cast<const bf16_t*>(a.data_ptr());
   const auto wp = reinterpret_cast<const bf16_t*>(w.data_ptr());
