@@ -3,6 +3,7 @@
 # Fast pair: randomwalks ILQL + PPO (tiny, no network); then the flagship
 # GPT-2 PPO bench.  Results land in $BENCH_OUTPUT_DIR (default: bench_out/).
 set -e
+export PYTHONPATH="$(pwd)${PYTHONPATH:+:$PYTHONPATH}"
 OUT="${BENCH_OUTPUT_DIR:-bench_out}"
 mkdir -p "$OUT"
 

@@ -231,7 +231,11 @@ def generate(
     try:
         shaping_graphable = shaping_fn is None or gen.graph_safe_shaping
         alibi = getattr(model.config, "position_encoding", None) == "alibi"
+        # the graph engine needs the fused decode kernels (device-side cache
+        # index); unsupported head dims use the eager loop's host-side state
+        head_ok = getattr(model.config, "head_dim", 64) in (32, 64, 128, 256)
         if (device.type == "cuda" and shaping_graphable and gen.use_graph and not alibi
+                and head_ok
                 and _graphs_enabled() and gen.max_new_tokens > 1 and gen.min_new_tokens == 0):
             engine = getattr(model, "_decode_engine", None)
             needed = T + gen.max_new_tokens

@@ -149,7 +149,9 @@ class Attention(nn.Module):
         # fused single-token decode path: decode_prep (split+RoPE+cache
         # append) + flash-decode attention, fully device-side (hipGraph-safe)
         if (T == 1 and kv_cache is not None and ctx.cache_idx is not None
-                and ctx.seq_lens is not None and x.is_cuda and ctx.alibi is None):
+                and ctx.seq_lens is not None and x.is_cuda and ctx.alibi is None
+                and self.head_dim in (32, 64, 128, 256)):
+            # (unsupported head dims fall through to the eager T=1 math below)
             cos_sin = rope_tables if self.cfg.position_encoding == "rope" else (None, None)
             if self.num_heads == self.num_kv_heads:
                 # MHA: split+RoPE+append and attention in ONE kernel (the
