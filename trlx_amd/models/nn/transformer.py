@@ -200,7 +200,8 @@ class Attention(nn.Module):
                 # cache stores unscaled k (decode kernel scales q itself)
                 pass
             k_full, v_full = kv_cache.update(self.layer_idx, k, v, ctx.start_pos)
-            if T == 1 and ctx.seq_lens is not None:
+            if (T == 1 and ctx.seq_lens is not None
+                    and (not x.is_cuda or self.head_dim in (32, 64, 128, 256))):
                 # fused flash-decode kernel
                 out = ops.attention_decode(q, k_full, v_full, ctx.seq_lens, self.scale,
                                            seq_starts=ctx.key_starts)
