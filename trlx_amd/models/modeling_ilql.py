@@ -74,7 +74,8 @@ class ILQLConfig(MethodConfig):
         (semantics of reference modeling_ilql.py:94-166)."""
         logits, (qs, target_qs, vs) = outputs
         terminal_mask = labels.dones[:, :-1]
-        n_nonterminal = max(1, terminal_mask.sum())
+        n_nonterminal = terminal_mask.sum().clamp(min=1)  # device tensor: no
+        # host sync in the hot loop, and safe under hipGraph capture
 
         if isinstance(labels, ILQLBatch):
             actions = labels.input_ids[:, 1:].gather(dim=1, index=labels.actions_ixs).unsqueeze(-1)

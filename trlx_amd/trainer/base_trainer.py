@@ -469,24 +469,31 @@ class NativeRLTrainer(BaseRLTrainer):
             static = {k: v.to(self.device).clone() for k, v in fields.items()}
             static_batch = type(microbatch)(**static)
             torch.cuda.synchronize()
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(2):
-                    loss, stats = self.loss(static_batch)
-                    if not all(torch.is_tensor(v) for v in stats.values()):
-                        torch.cuda.current_stream().wait_stream(side)
-                        self._train_graphs[key] = False
-                        logger.info("train-step graph disabled: non-tensor stats %s",
-                                    [k for k, v in stats.items() if not torch.is_tensor(v)])
-                        return None
-                    loss.backward()
-            torch.cuda.current_stream().wait_stream(side)
-            torch.cuda.synchronize()
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                loss_out, stats_out = self.loss(static_batch)
-                loss_out.backward()
+            try:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(2):
+                        loss, stats = self.loss(static_batch)
+                        if not all(torch.is_tensor(v) for v in stats.values()):
+                            torch.cuda.current_stream().wait_stream(side)
+                            self._train_graphs[key] = False
+                            logger.info("train-step graph disabled: non-tensor stats %s",
+                                        [k for k, v in stats.items() if not torch.is_tensor(v)])
+                            return None
+                        loss.backward()
+                torch.cuda.current_stream().wait_stream(side)
+                torch.cuda.synchronize()
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    loss_out, stats_out = self.loss(static_batch)
+                    loss_out.backward()
+            except Exception as e:  # capture-unsafe loss (host sync etc)
+                logger.info("train-step graph disabled for this shape: %s", e)
+                torch.cuda.synchronize()
+                self._train_graphs[key] = False
+                self.opt.zero_grad()
+                return None
             # warmup + capture polluted the grads; this is the start of an
             # optimizer window (num_mb == 1), so zeroing is safe
             self.opt.zero_grad()
