@@ -51,7 +51,8 @@ class RewardModel(nn.Module):
 
     def _rewards(self, input_ids: torch.Tensor) -> torch.Tensor:
         out = self.transformer(input_ids, return_logits=False)
-        return self.v_head(out.last_hidden_state).squeeze(-1).float()  # [B, T]
+        h = out.last_hidden_state.to(self.v_head.weight.dtype)
+        return self.v_head(h).squeeze(-1).float()  # [B, T]
 
     def _end_index(self, ids: torch.Tensor) -> torch.Tensor:
         """First pad position per row, or T if un-padded (reference

