@@ -34,3 +34,19 @@ def tiny_config(**overrides):
 @pytest.fixture
 def tiny_cfg():
     return tiny_config()
+
+
+@pytest.fixture(autouse=True)
+def _cuda_test_teardown():
+    """Deterministic GPU cleanup between tests: each e2e test builds hipGraph
+    decode/train engines whose pools otherwise get released by GC in the
+    middle of a LATER test's allocations (observed segfault in
+    test_ppo_dense_rewards after the examples suite on a GPU machine)."""
+    yield
+    if torch.cuda.is_available():
+        import gc
+
+        torch.cuda.synchronize()
+        gc.collect()
+        torch.cuda.synchronize()
+        torch.cuda.empty_cache()
