@@ -400,9 +400,10 @@ def test_sequence_parallel_equivalence():
     _spawn(_worker_sp_forward, 29519)
 
 
-def _worker_pp_train(rank):
-    """PP=2 fill-drain fwd+bwd must equal the single-process model: same
-    losses and same gradients on every stage's parameters."""
+def _worker_pp_train(rank, tied=False):
+    """PP=2 fwd+bwd must equal the single-process model: same losses and
+    same gradients on every stage's parameters (incl. the summed tied
+    embedding grad when tie_word_embeddings)."""
     import torch.nn.functional as F
 
     from trlx_amd.models.nn.config import TransformerConfig
@@ -412,7 +413,7 @@ def _worker_pp_train(rank):
     torch.manual_seed(0)
     cfg = TransformerConfig(vocab_size=150, hidden_size=48, num_layers=4, num_heads=4,
                             max_position_embeddings=64, arch_name="gpt2",
-                            tie_word_embeddings=False)
+                            tie_word_embeddings=tied)
     full = CausalTransformer(cfg)
     full_sd = {k: v for k, v in full.state_dict().items() if not k.startswith("rope_")}
 
@@ -463,6 +464,10 @@ def _worker_pp_train(rank):
         if name.startswith("layers."):
             idx = int(name.split(".")[1]) + stage.lo
             ref_name = f"layers.{idx}." + name.split(".", 2)[2]
+        elif name == "lm_head.weight" and tied:
+            # the tied full model exposes the shared weight once, under the
+            # embedding name; its grad is the SUM both stages must carry
+            ref_name = "embed_tokens.weight"
         else:
             ref_name = name
         assert p.grad is not None, name
@@ -472,6 +477,14 @@ def _worker_pp_train(rank):
 
 def test_pipeline_parallel_equivalence():
     _spawn(_worker_pp_train, 29520)
+
+
+def _worker_pp_tied(rank):
+    _worker_pp_train(rank, tied=True)
+
+
+def test_pipeline_parallel_tied_embeddings():
+    _spawn(_worker_pp_tied, 29524)
 
 
 def _worker_tp_checkpoint(rank):

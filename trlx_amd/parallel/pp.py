@@ -241,9 +241,39 @@ class PipelineRunner:
         for _ in range(warmup):
             do_backward()
         self._drain_sends()
+        self._sync_tied_embedding_grads()
         if losses:
             return torch.stack(losses).mean()
         return None
+
+    def _sync_tied_embedding_grads(self):
+        """Tied input/output embeddings live on DIFFERENT stages under PP:
+        stage 0 holds embed_tokens, the last stage holds lm_head (initialized
+        from the same weight).  Their grads must be SUMMED so both copies
+        take the same update and stay tied (the Megatron/NeMo embedding-group
+        all-reduce; reference megatron stack).  Exchange over p2p between the
+        two end stages."""
+        stage = self.stage
+        if not stage.config.tie_word_embeddings or len(self.pp_ranks) < 2:
+            return
+        if stage.is_first and stage.is_last:
+            return
+        first, last = self.pp_ranks[0], self.pp_ranks[-1]
+        if stage.is_first:
+            g = stage.embed_tokens.weight.grad
+            if g is None:
+                g = torch.zeros_like(stage.embed_tokens.weight)
+            dist.send(g.contiguous(), last)
+            total = torch.empty_like(g)
+            dist.recv(total, last)
+            stage.embed_tokens.weight.grad = total
+        elif stage.is_last:
+            other = torch.empty_like(stage.lm_head.weight)
+            dist.recv(other, first)
+            g = stage.lm_head.weight.grad
+            total = other if g is None else other + g
+            dist.send(total.contiguous(), first)
+            stage.lm_head.weight.grad = total
 
     def _forward_backward_gpipe(self, microbatches, loss_fn, hidden_dtype=torch.float32):
         stage = self.stage
@@ -287,6 +317,7 @@ class PipelineRunner:
                 self._send(h_in.grad, self.prev)
 
         self._drain_sends()
+        self._sync_tied_embedding_grads()
         if losses:
             return torch.stack([l.detach() for l in losses]).mean()
         return None
