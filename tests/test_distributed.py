@@ -13,6 +13,10 @@ WORLD = 2
 
 
 def _run(rank, fn, port):
+    # these are gloo/CPU tests: hide any GPU so workers on a GPU machine
+    # don't route tensors onto one shared device (gloo moves CPU tensors)
+    os.environ["HIP_VISIBLE_DEVICES"] = ""
+    os.environ["CUDA_VISIBLE_DEVICES"] = ""
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(WORLD)
@@ -30,6 +34,8 @@ def _spawn(fn, port):
 
 
 def _run_n(rank, fn, port, world):
+    os.environ["HIP_VISIBLE_DEVICES"] = ""
+    os.environ["CUDA_VISIBLE_DEVICES"] = ""
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
