@@ -391,3 +391,53 @@ def test_gpu_generate_alibi_and_mqa(family):
     assert out1.shape[1] == 14
     assert torch.equal(out1, out2)
     assert int(out1.max()) < cfg.vocab_size
+
+
+@pytest.mark.parametrize("family", ["gpt2", "llama", "gptj", "gpt_neox", "opt",
+                                    "bloom", "gpt_bigcode"])
+def test_save_pretrained_roundtrip_all_families(family, tmp_path):
+    """save_pretrained -> from_pretrained round trip through the HF-format
+    conversion (state_dict_to_hf + state_dict_from_hf) for every family:
+    identical logits after reload."""
+    import conftest
+    from trlx_amd.models.modeling_base import PreTrainedModelWrapper
+    from trlx_amd.models.nn.config import TransformerConfig
+
+    torch.manual_seed(17)
+    base_kwargs = dict(vocab_size=160, hidden_size=64, num_layers=2, num_heads=4,
+                       max_position_embeddings=64)
+    per_family = {
+        "gpt2": dict(arch_name="gpt2", norm="layernorm", position_encoding="learned",
+                     activation="gelu_new", attn_bias=True, mlp_bias=True,
+                     tie_word_embeddings=True),
+        "llama": dict(arch_name="llama", norm="rmsnorm", position_encoding="rope",
+                      activation="silu", swiglu=True, attn_bias=False, mlp_bias=False,
+                      tie_word_embeddings=False),
+        "gptj": dict(arch_name="gptj", norm="layernorm", position_encoding="rope",
+                     rope_pct=0.25, rope_interleaved=True, parallel_residual=True,
+                     activation="gelu_new", attn_bias=False, mlp_bias=True,
+                     tie_word_embeddings=False, lm_head_bias=True),
+        "gpt_neox": dict(arch_name="gpt_neox", norm="layernorm", position_encoding="rope",
+                         rope_pct=0.25, parallel_residual=True, activation="gelu_new",
+                         attn_bias=True, mlp_bias=True, tie_word_embeddings=False),
+        "opt": dict(arch_name="opt", norm="layernorm", position_encoding="learned",
+                    activation="relu", attn_bias=True, mlp_bias=True,
+                    tie_word_embeddings=True, extra={"position_offset": 2}),
+        "bloom": dict(arch_name="bloom", norm="layernorm", position_encoding="alibi",
+                      activation="gelu_new", attn_bias=True, mlp_bias=True,
+                      tie_word_embeddings=True, extra={"pre_embed_norm": True}),
+        "gpt_bigcode": dict(arch_name="gpt_bigcode", norm="layernorm",
+                            position_encoding="learned", activation="gelu_new",
+                            num_kv_heads=1, attn_bias=True, mlp_bias=True,
+                            tie_word_embeddings=True),
+    }
+    cfg = TransformerConfig(**base_kwargs, **per_family[family])
+    m = PreTrainedModelWrapper.from_config(cfg)
+    d = str(tmp_path / "hf_model")
+    m.save_pretrained(d)
+    m2 = PreTrainedModelWrapper.from_pretrained(d)
+    ids = torch.randint(3, 150, (2, 9))
+    with torch.no_grad():
+        a = m.base_model(ids).logits
+        b = m2.base_model(ids).logits
+    assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
