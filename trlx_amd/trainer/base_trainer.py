@@ -54,6 +54,18 @@ class NativeRLTrainer(BaseRLTrainer):
         comm.init_distributed()
         from ..parallel import topo
 
+        if config.train.pipeline_parallel_size > 1:
+            # PP exists as an equivalence-tested library (parallel/pp.py:
+            # stage splitting, 1F1B/GPipe schedules, tied-embedding sync,
+            # full-state-dict resharding) but the RL trainer loop is not yet
+            # built on it (rollout generation across stages is the blocker —
+            # NOTES_ROUND2.md).  Refuse loudly rather than silently training
+            # replicated models that never synchronize.
+            raise NotImplementedError(
+                "pipeline_parallel_size > 1 is not wired into the RL trainers yet; "
+                "use parallel.pp.PipelineRunner directly for pipelined supervised "
+                "losses, or TP/DP/ZeRO for RL runs (round-2 item)."
+            )
         topo.init_model_parallel(config.train.tensor_parallel_size,
                                  config.train.pipeline_parallel_size)
         self.device = comm.get_device()
