@@ -184,3 +184,31 @@ def test_hh_example_and_reward_server(tmp_path, monkeypatch):
         "method.num_rollouts": 4, "method.chunk_size": 4, "method.ppo_epochs": 1,
         "method.gen_kwargs": dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True),
     })
+
+
+def test_reward_server_with_trained_rm(tmp_path, monkeypatch):
+    """reward_server serves a trained RM checkpoint (not just the oracle)."""
+    monkeypatch.syspath_prepend("examples/hh")
+    monkeypatch.syspath_prepend("examples/summarize_rlhf")
+    import importlib
+
+    import conftest
+    from fastapi.testclient import TestClient
+
+    rm_lib = importlib.import_module("reward_model")
+    model = rm_lib.RewardModel.from_pretrained(conftest.tiny_config(), pad_token_id=2)
+    model.save_checkpoint(str(tmp_path / "rm"))
+
+    server = importlib.import_module("reward_server")
+    server._rm["model"] = None
+    server.load_rm(str(tmp_path / "rm"))
+    try:
+        client = TestClient(server.app)
+        assert client.get("/health").json()["rm"] is True
+        r = client.post("/reward", json={"samples": ["hello there", "goodbye"]})
+        assert r.status_code == 200
+        scores = r.json()["scores"]
+        assert len(scores) == 2 and all(isinstance(x, float) for x in scores)
+    finally:
+        server._rm["model"] = None
+        server._rm["tok"] = None
