@@ -14,7 +14,10 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 CSRC = Path(__file__).parent / "trlx_amd" / "csrc"
 
-sources = [str(CSRC / "bindings.cpp")] + sorted(str(p) for p in CSRC.glob("*.hip"))
+# *_hip.hip files are hipify byproducts regenerated at build time — never sources.
+sources = [str(CSRC / "bindings.cpp")] + sorted(
+    str(p) for p in CSRC.glob("*.hip") if not p.name.endswith("_hip.hip")
+)
 
 ext = CUDAExtension(
     name="trlx_amd._C",
