@@ -587,3 +587,52 @@ def _worker_tp2dp2(rank):
 
 def test_tp2_dp2_topology():
     _spawn_n(_worker_tp2dp2, 29523, 4)
+
+
+def _worker_dp_sharding(rank):
+    from trlx_amd.parallel import topo
+    from trlx_amd.pipeline.offline_pipeline import PromptPipeline
+    from trlx_amd.utils.tokenizer import ByteTokenizer
+
+    topo.reset()
+    try:
+        topo.init_model_parallel(1, 1)  # pure DP: dp_size == world
+        tok = ByteTokenizer()
+        prompts = [f"prompt number {i}" for i in range(8)]
+        pipe = PromptPipeline(prompts, 16, tok)
+        loader = pipe.create_loader(2, shuffle=False)
+        rows = []
+        for b in loader:
+            rows.extend(tuple(r) for r in b["input_ids"].tolist())
+        # disjoint 4-sample shards whose union covers the dataset
+        assert len(rows) == 4
+        gathered = [None, None]
+        dist.all_gather_object(gathered, rows)
+        all_rows = [r for shard in gathered for r in shard]
+        assert len(set(all_rows)) == 8
+        assert len(set(gathered[0]) & set(gathered[1])) == 0
+
+        # shuffled epochs: infinite_dataloader bumps the sampler epoch
+        from trlx_amd.utils import infinite_dataloader
+
+        loader = pipe.create_loader(4, shuffle=True)
+        it = infinite_dataloader(loader)
+        epoch0 = tuple(next(it)["input_ids"].flatten().tolist())
+        epoch1 = tuple(next(it)["input_ids"].flatten().tolist())
+        orders = [None, None]
+        dist.all_gather_object(orders, (epoch0, epoch1))
+        # at least one rank must see a different order across epochs
+        assert any(a != b for a, b in orders)
+
+        # TP=2 (dp_size == 1): both ranks see the FULL dataset
+        topo.reset()
+        topo.init_model_parallel(2, 1)
+        loader = pipe.create_loader(2, shuffle=False)
+        n = sum(len(b["input_ids"]) for b in loader)
+        assert n == 8
+    finally:
+        topo.reset()
+
+
+def test_dp_dataset_sharding():
+    _spawn(_worker_dp_sharding, 29524)

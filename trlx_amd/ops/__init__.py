@@ -571,8 +571,13 @@ def sample_token(
                 seed = int(torch.randint(0, 2**31 - 1, (1,), generator=generator).item())
             if top_p is not None and 0.0 < top_p < 1.0:
                 # top-p needs a sorted scan: do the filter with library sort,
-                # then fused gumbel sampling on the filtered logits
-                lg = _top_p_filter(logits.float() / max(temperature, 1e-6), top_p)
+                # then fused gumbel sampling on the filtered logits.
+                # HF order: top-k first, then top-p (matches reference.sample_token)
+                lg = logits.float() / max(temperature, 1e-6)
+                if top_k and 0 < top_k < logits.shape[-1]:
+                    kth = torch.topk(lg, top_k, dim=-1).values[..., -1, None]
+                    lg = lg.masked_fill(lg < kth, float("-inf"))
+                lg = _top_p_filter(lg, top_p)
                 return ext.gumbel_sample(lg.contiguous(), 1.0, None, seed, offset)
             dev_offset = isinstance(offset, torch.Tensor)
             if top_k and 0 < top_k < logits.shape[-1]:

@@ -39,6 +39,27 @@ def register_datapipeline(name):
     return cls
 
 
+def dp_sampler(dataset, shuffle: bool = False, seed: int = 0):
+    """A DistributedSampler over the DATA-parallel group, or None when dp=1.
+
+    Each DP rank iterates a disjoint shard per epoch (the reference shards via
+    accelerate.prepare); TP/PP peers share a dp_rank and therefore see
+    identical data.  Callers must pass ``shuffle=False`` to the DataLoader
+    when a sampler is returned and bump ``sampler.set_epoch`` across epochs
+    (``utils.infinite_dataloader`` does this).
+    """
+    from ..parallel import topo
+
+    if topo.dp_size() <= 1:
+        return None
+    from torch.utils.data import DistributedSampler
+
+    return DistributedSampler(
+        dataset, num_replicas=topo.dp_size(), rank=topo.dp_rank(),
+        shuffle=shuffle, seed=seed, drop_last=False,
+    )
+
+
 class BasePipeline(Dataset):
     def __init__(self, path: str = "dataset"):
         super().__init__()

@@ -21,7 +21,7 @@ from ..data.ilql_types import (
     ILQLSeq2SeqBatch,
     ILQLSeq2SeqElement,
 )
-from . import BasePipeline, BaseRolloutStore, register_datapipeline
+from . import BasePipeline, BaseRolloutStore, dp_sampler, register_datapipeline
 
 
 @dataclass
@@ -115,7 +115,9 @@ class DialogStore(BaseRolloutStore):
                 labels=pad_sequence([e["labels"] for e in elems], batch_first=True, padding_value=-100),
             )
 
-        return DataLoader(self, batch_size=batch_size, collate_fn=collate_fn, shuffle=shuffle)
+        sampler = dp_sampler(self, shuffle=shuffle)
+        return DataLoader(self, batch_size=batch_size, collate_fn=collate_fn,
+                          shuffle=shuffle and sampler is None, sampler=sampler)
 
 
 @register_datapipeline
@@ -168,6 +170,10 @@ class PromptPipeline(BasePipeline):
                     out[key] = [x[key] for x in xs]
             return out
 
+        if sampler is None:
+            sampler = dp_sampler(self, shuffle=shuffle)
+            if sampler is not None:
+                shuffle = False
         return DataLoader(self, batch_size=batch_size, collate_fn=collate_fn, shuffle=shuffle,
                           sampler=sampler, num_workers=0, drop_last=drop_last)
 
@@ -205,9 +211,10 @@ class ILQLRolloutStorage(BaseRolloutStore):
         return len(self.input_ids)
 
     def create_loader(self, batch_size: int):
+        sampler = dp_sampler(self, shuffle=True)
         return DataLoader(
-            self, batch_size=batch_size, shuffle=True, collate_fn=ilql_collate_fn,
-            drop_last=dist.is_initialized(),
+            self, batch_size=batch_size, shuffle=sampler is None, sampler=sampler,
+            collate_fn=ilql_collate_fn, drop_last=dist.is_initialized(),
         )
 
 
@@ -246,7 +253,8 @@ class ILQLSeq2SeqRolloutStorage(BaseRolloutStore):
         return len(self.input_ids)
 
     def create_loader(self, batch_size: int):
+        sampler = dp_sampler(self, shuffle=True)
         return DataLoader(
-            self, batch_size=batch_size, shuffle=True, collate_fn=ilql_seq2seq_collate_fn,
-            drop_last=dist.is_initialized(),
+            self, batch_size=batch_size, shuffle=sampler is None, sampler=sampler,
+            collate_fn=ilql_seq2seq_collate_fn, drop_last=dist.is_initialized(),
         )

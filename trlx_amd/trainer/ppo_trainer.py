@@ -408,52 +408,59 @@ class PPOTrainer(NativeRLTrainer):
                 resp_log_ratio = log_ratio[:, start:]
             else:
                 # causal experience pass — one trunk pass for policy logits +
-                # values (+ ref logits via hydra); the vocab-wide
-                # lm_head/logprob math runs only on the response region
-                # [start, T-1) — prompt-position KL (a logged stat in the
-                # reference, not a training signal) is therefore measured over
-                # the response region
+                # values (+ ref logits via hydra).  With the FIXED KL
+                # controller the vocab-wide lm_head/logprob math runs only on
+                # the response region [start, T-1) (mean_kl is then a logged
+                # stat, measured over the response region).  With the ADAPTIVE
+                # controller mean_kl drives kl_ctl.update, so the full-sequence
+                # extent is computed to match the reference's
+                # kl.sum(1).mean() over prompt+response
+                # (accelerate_ppo_trainer.py:455-460,506).
                 all_tokens = torch.cat((prompt_tensors, sample_outputs), dim=1)
                 attention_mask = all_tokens.not_equal(self.tokenizer.pad_token_id).long().to(device)
                 n_samples = samples.shape[0]
                 start = prompt_tensors.shape[1] - 1
                 T_all = all_tokens.shape[1]
-                labels = all_tokens[:, start + 1 :]
+                kl_lo = 0 if isinstance(self.kl_ctl, AdaptiveKLController) else start
+                labels = all_tokens[:, kl_lo + 1 :]
                 with torch.no_grad():
                     if hasattr(self.model, "forward_experience"):
                         # fused hand-MFMA path: hidden -> logprobs directly,
                         # the [B, T, V] logits never materialize
                         logprobs, ref_logprobs, values = self.model.forward_experience(
-                            all_tokens, attention_mask, start, T_all - 1, labels)
+                            all_tokens, attention_mask, kl_lo, T_all - 1, labels)
                         if ref_logprobs is None:
                             if self.ref_model is not None:
                                 ref_logprobs, _, _ = self.ref_model.forward_experience(
-                                    all_tokens, attention_mask, start, T_all - 1, labels)
+                                    all_tokens, attention_mask, kl_lo, T_all - 1, labels)
                             else:
                                 # num_layers_unfrozen == -1, no ref: KL vs itself
                                 ref_logprobs = logprobs
                     else:
                         outputs = self.model(all_tokens, attention_mask=attention_mask,
                                              return_ref_logits=True,
-                                             logits_slice=(start, T_all - 1))
+                                             logits_slice=(kl_lo, T_all - 1))
                         logits, values = outputs.logits, outputs.values
                         if outputs.ref_logits is not None:
                             ref_logits = outputs.ref_logits
                         elif self.ref_model is not None:
                             ref_logits = self.ref_model(all_tokens, attention_mask=attention_mask,
-                                                        logits_slice=(start, T_all - 1)).logits
+                                                        logits_slice=(kl_lo, T_all - 1)).logits
                         else:
                             ref_logits = logits
                         logprobs = logprobs_of_labels(logits, labels)
                         ref_logprobs = logprobs_of_labels(ref_logits, labels)
 
-                log_ratio = (logprobs - ref_logprobs) * attention_mask[:, start:-1]
+                log_ratio = (logprobs - ref_logprobs) * attention_mask[:, kl_lo:-1]
                 kl = log_ratio.exp() - 1 - log_ratio
                 mean_kl_per_token = kl.mean()
                 mean_kl = kl.sum(1).mean()
 
+                off = start - kl_lo  # 0 on the fixed-KL fast path
                 ends_t = attention_mask[:, start:].sum(1) + 1
-                resp_logprobs, resp_values, resp_log_ratio = logprobs, values, log_ratio
+                resp_logprobs = logprobs[:, off:]
+                resp_values = values[:, off:]
+                resp_log_ratio = log_ratio[:, off:]
 
             if torch.cuda.is_available():
                 torch.cuda.synchronize()

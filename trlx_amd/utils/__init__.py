@@ -167,11 +167,15 @@ def get_git_tag() -> Tuple[str, str]:
 
 def infinite_dataloader(dataloader: Iterable, sampler=None) -> Iterable:
     """Cycle a dataloader forever, bumping the sampler epoch each wrap so
-    distributed shuffles differ per epoch."""
+    distributed shuffles differ per epoch (reference utils/__init__.py:240-250)."""
     epoch = 0
+    if sampler is None:
+        sampler = getattr(dataloader, "sampler", None)
     while True:
         for batch in dataloader:
             yield batch
         epoch += 1
+        if hasattr(sampler, "set_epoch"):
+            sampler.set_epoch(epoch)
         if sampler is not None and hasattr(sampler, "set_epoch"):
             sampler.set_epoch(epoch)

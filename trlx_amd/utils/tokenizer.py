@@ -216,14 +216,27 @@ class SyntheticVocabTokenizer(ByteTokenizer):
 
 
 def get_tokenizer(path: str, padding_side: str = "left", truncation_side: str = "right", **kwargs):
-    """Load a tokenizer from a local path; fall back to ByteTokenizer offline."""
+    """Load a tokenizer from a local path; ``byte``/``synthetic[:V]`` name the
+    built-in offline tokenizers explicitly.
+
+    A local directory that exists but fails to load RAISES (a typo'd or
+    corrupt tokenizer dir must not silently train on byte tokens); any other
+    unresolvable name (e.g. a hub id with no network) falls back to
+    ByteTokenizer with a prominent warning.
+    """
     import os
+
+    from . import logging
+
+    logger = logging.get_logger(__name__)
 
     if path and path.startswith("synthetic"):
         # "synthetic" or "synthetic:VOCAB"
         vocab = int(path.split(":")[1]) if ":" in path else 50257
         return SyntheticVocabTokenizer(vocab, padding_side, truncation_side)
-    if path and os.path.isdir(path):
+    if not path or path == "byte":
+        return ByteTokenizer(padding_side=padding_side, truncation_side=truncation_side)
+    if os.path.isdir(path):
         byte_cfg = os.path.join(path, "byte_tokenizer.json")
         if os.path.exists(byte_cfg):
             import json
@@ -241,6 +254,14 @@ def get_tokenizer(path: str, padding_side: str = "left", truncation_side: str = 
             if tok.pad_token is None:
                 tok.pad_token = tok.eos_token
             return tok
-        except Exception:
-            pass
+        except Exception as e:
+            raise ValueError(
+                f"tokenizer_path {path!r} is a directory but no tokenizer could be "
+                f"loaded from it: {e}"
+            ) from e
+    logger.warning(
+        "tokenizer_path %r is not a local directory (and there is no network to "
+        "download it); falling back to the byte-level ByteTokenizer. Set "
+        "tokenizer_path='byte' to make this explicit.", path,
+    )
     return ByteTokenizer(padding_side=padding_side, truncation_side=truncation_side)
