@@ -435,10 +435,13 @@ def _worker_pp_train(rank, tied=False):
         mask = torch.ones_like(ids)
         mbs.append({"input_ids": ids, "attention_mask": mask})
 
-    def loss_fn(logits, mb):
+    def ce(logits, mb):
         ids = mb["input_ids"]
         return F.cross_entropy(logits[:, :-1].reshape(-1, 150).float(),
                                ids[:, 1:].reshape(-1))
+
+    def loss_fn(h, mb):  # runner passes the post-norm hidden; project here
+        return ce(stage.project(h), mb)
 
     mean_loss = runner.forward_backward(mbs, loss_fn)  # default: 1F1B
 
@@ -457,7 +460,7 @@ def _worker_pp_train(rank, tied=False):
     ref_losses = []
     for mb in mbs:
         out = full(mb["input_ids"], attention_mask=mb["attention_mask"])
-        ref_losses.append(loss_fn(out.logits, mb))
+        ref_losses.append(ce(out.logits, mb))
         ref_losses[-1].backward()
     ref_mean = torch.stack([l.detach() for l in ref_losses]).mean()
 
@@ -636,3 +639,250 @@ def _worker_dp_sharding(rank):
 
 def test_dp_dataset_sharding():
     _spawn(_worker_dp_sharding, 29524)
+
+
+def _worker_pp_hydra_model(rank):
+    """PipelinedPPOModel (pp=2) must match the single-process hydra model:
+    forward_experience numerics and greedy pipelined generation."""
+    from trlx_amd.models.modeling_ppo import AutoModelForCausalLMWithHydraValueHead
+    from trlx_amd.models.modeling_pp import PipelinedPPOModel
+    from trlx_amd.models.nn.config import TransformerConfig
+    from trlx_amd.parallel import topo
+    from trlx_amd.utils.modeling import logprobs_of_labels
+
+    topo.reset()
+    try:
+        torch.manual_seed(0)
+        cfg = TransformerConfig(vocab_size=160, hidden_size=48, num_layers=4, num_heads=4,
+                                max_position_embeddings=64, arch_name="gpt2")
+        full = AutoModelForCausalLMWithHydraValueHead.from_config(cfg, num_layers_unfrozen=1)
+        full.eval()
+        full_sd = full.state_dict()
+
+        topo.init_model_parallel(1, 2)
+        model = PipelinedPPOModel(cfg, num_layers_unfrozen=1)
+        model.load_full_base_state_dict(
+            {k: v for k, v in full_sd.items() if k.startswith("base_model.")})
+        if model.stage.is_last:
+            model.v_head.load_state_dict(full.v_head.state_dict())
+        model.eval()
+
+        g = torch.Generator().manual_seed(3)
+        ids = torch.randint(3, 160, (2, 9), generator=g)
+        mask = torch.ones_like(ids)
+        mask[0, :2] = 0
+        T = ids.shape[1]
+        lo, hi = 3, T - 1
+        labels = ids[:, lo + 1 : hi + 1]
+        lp, rlp, vals = model.forward_experience(ids, mask, lo, hi, labels)
+
+        out = full(ids, attention_mask=mask, return_ref_logits=True, logits_slice=(lo, hi))
+        ref_lp = logprobs_of_labels(out.logits, labels)
+        ref_rlp = logprobs_of_labels(out.ref_logits, labels)
+        assert torch.allclose(lp, ref_lp, atol=1e-4), (lp - ref_lp).abs().max()
+        assert torch.allclose(rlp, ref_rlp, atol=1e-4), (rlp - ref_rlp).abs().max()
+        assert torch.allclose(vals, out.values, atol=1e-4), (vals - out.values).abs().max()
+
+        samples_pp = model.generate(ids, mask, max_new_tokens=5, do_sample=False,
+                                    use_graph=False)
+        samples_full = full.generate(ids, mask, max_new_tokens=5, do_sample=False,
+                                     use_graph=False)
+        assert torch.equal(samples_pp, samples_full), (samples_pp, samples_full)
+
+        # sharded checkpoint round-trip through the PP merge
+        out_dir = f"/tmp/pp_ckpt_test"
+        model.save_pretrained(out_dir)
+        dist.barrier()
+        if rank == 0:
+            from trlx_amd.models.modeling_pp import merge_pp_checkpoint
+
+            merged = merge_pp_checkpoint(out_dir)
+            for k, v in full_sd.items():
+                if k.startswith("rope_") or ".rope_" in k:
+                    continue
+                assert k in merged, k
+                assert torch.allclose(merged[k], v, atol=1e-6), k
+    finally:
+        topo.reset()
+
+
+def test_pp_hydra_model_equivalence():
+    _spawn(_worker_pp_hydra_model, 29525)
+
+
+def _ppo_cfg_for_pp(rank, pp, tp=1):
+    import trlx_amd  # noqa: F401
+    from trlx_amd.data.default_configs import default_ppo_config
+    from trlx_amd.models.nn.config import TransformerConfig
+
+    cfg = default_ppo_config()
+    tiny = TransformerConfig(vocab_size=300, hidden_size=32, num_layers=4, num_heads=2,
+                             max_position_embeddings=128, arch_name="gpt2")
+    cfg.model.model_path = "tiny"
+    cfg.model.model_extra_configs = {"config": tiny.to_dict()}
+    cfg.model.num_layers_unfrozen = 1
+    cfg.tokenizer.tokenizer_path = "byte"
+    cfg.train.seq_length = 32
+    cfg.train.batch_size = 2
+    cfg.train.total_steps = 2
+    cfg.train.eval_interval = 2
+    cfg.train.checkpoint_interval = 100
+    cfg.train.tracker = None
+    cfg.train.save_best = False
+    cfg.train.pipeline_parallel_size = pp
+    cfg.train.tensor_parallel_size = tp
+    cfg.train.checkpoint_dir = f"/tmp/dist_ppo_pp_{rank}"
+    cfg.method.num_rollouts = 4
+    cfg.method.chunk_size = 2
+    cfg.method.ppo_epochs = 1
+    cfg.method.gen_kwargs = dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True)
+    return cfg
+
+
+def _worker_pp_ppo_train(rank):
+    """PP=2 PPO end-to-end on gloo: pipelined generation, experience and the
+    1F1B train step driven by the trainer."""
+    import trlx_amd
+    from trlx_amd.parallel import topo
+
+    cfg = _ppo_cfg_for_pp(rank, pp=2)
+
+    def reward_fn(samples, prompts, outputs, **kw):
+        return [float(len(o)) for o in outputs]
+
+    try:
+        trainer = trlx_amd.train(
+            reward_fn=reward_fn,
+            prompts=["aa", "bb", "cc", "dd"],
+            eval_prompts=["aa", "bb"],
+            config=cfg,
+        )
+        assert trainer.iter_count == 2
+        # both stages must have taken real optimizer steps on trainable params
+        n_trainable = sum(p.numel() for p in trainer.model.parameters() if p.requires_grad)
+        if trainer.model.stage.is_last:
+            assert n_trainable > 0
+    finally:
+        topo.reset()
+
+
+def test_pp_ppo_end_to_end():
+    _spawn(_worker_pp_ppo_train, 29526)
+
+
+def _worker_pp2_dp2_ppo(rank):
+    """world 4 = PP2 x DP2 PPO end-to-end: DP replicas of each stage must end
+    with identical weights (grads all-reduced over the DP group)."""
+    import trlx_amd
+    from trlx_amd.parallel import topo
+
+    cfg = _ppo_cfg_for_pp(rank, pp=2)
+
+    def reward_fn(samples, prompts, outputs, **kw):
+        return [float(len(o)) for o in outputs]
+
+    try:
+        trainer = trlx_amd.train(
+            reward_fn=reward_fn,
+            prompts=["aa", "bb", "cc", "dd", "ee", "ff", "gg", "hh"],
+            eval_prompts=["aa", "bb"],
+            config=cfg,
+        )
+        assert trainer.iter_count == 2
+        for p in trainer.model.parameters():
+            if p.requires_grad:
+                buf = [torch.empty_like(p) for _ in range(2)]
+                dist.all_gather(buf, p.detach(), group=topo.dp_group())
+                assert torch.allclose(buf[0], buf[1], atol=1e-6)
+    finally:
+        topo.reset()
+
+
+def test_pp2_dp2_ppo_end_to_end():
+    _spawn_n(_worker_pp2_dp2_ppo, 29527, 4)
+
+
+def _worker_tp2_pp2_ppo(rank):
+    """world 4 = TP2 x PP2 PPO end-to-end: the full hybrid model-parallel
+    path (BASELINE config #5's shape at test scale)."""
+    import trlx_amd
+    from trlx_amd.parallel import topo
+
+    cfg = _ppo_cfg_for_pp(rank, pp=2, tp=2)
+
+    def reward_fn(samples, prompts, outputs, **kw):
+        return [float(len(o)) for o in outputs]
+
+    try:
+        trainer = trlx_amd.train(
+            reward_fn=reward_fn,
+            prompts=["aa", "bb", "cc", "dd"],
+            eval_prompts=["aa", "bb"],
+            config=cfg,
+        )
+        assert trainer.iter_count == 2
+        # TP peers of the same stage must hold identical REPLICATED params
+        # (norms); sanity-check one on the last stage
+        if trainer.model.stage.is_last:
+            p = trainer.model.stage.final_norm.weight.detach()
+            buf = [torch.empty_like(p) for _ in range(2)]
+            dist.all_gather(buf, p, group=topo.tp_group())
+            assert torch.allclose(buf[0], buf[1], atol=1e-6)
+    finally:
+        topo.reset()
+
+
+def test_tp2_pp2_ppo_end_to_end():
+    _spawn_n(_worker_tp2_pp2_ppo, 29528, 4)
+
+
+def _worker_pp_sft_train(rank):
+    """PP=2 SFT end-to-end (dialog store, CE loss on the last stage)."""
+    import trlx_amd
+    from trlx_amd.data.default_configs import default_sft_config
+    from trlx_amd.models.nn.config import TransformerConfig
+    from trlx_amd.parallel import topo
+
+    cfg = default_sft_config()
+    tiny = TransformerConfig(vocab_size=300, hidden_size=32, num_layers=4, num_heads=2,
+                             max_position_embeddings=128, arch_name="gpt2")
+    cfg.model.model_path = "tiny"
+    cfg.model.model_extra_configs = {"config": tiny.to_dict()}
+    cfg.tokenizer.tokenizer_path = "byte"
+    cfg.train.seq_length = 32
+    cfg.train.batch_size = 2
+    cfg.train.total_steps = 2
+    cfg.train.eval_interval = 10
+    cfg.train.checkpoint_interval = 100
+    cfg.train.tracker = None
+    cfg.train.save_best = False
+    cfg.train.pipeline_parallel_size = 2
+    cfg.train.checkpoint_dir = f"/tmp/dist_sft_pp_{rank}"
+    cfg.method.gen_kwargs = dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True)
+
+    try:
+        trainer = trlx_amd.train(
+            samples=[["question a", "answer x"], ["question b", "answer y"],
+                     ["question c", "answer z"], ["question d", "answer w"]],
+            eval_prompts=["question a"],
+            config=cfg,
+        )
+        assert trainer.iter_count == 2
+        # tied embeddings: stage0's embed and stage1's lm_head must stay equal
+        st = trainer.model.stage
+        if st.is_first:
+            w = st.embed_tokens.weight.detach()
+            dist.send(w.contiguous(), 1)
+        else:
+            other = torch.empty_like(st.lm_head.weight)
+            dist.recv(other, 0)
+            # separate per-rank optimizer states apply the same update up to
+            # fp rounding; ties stay within ~1e-4 over a few steps
+            assert torch.allclose(other, st.lm_head.weight.detach(), atol=3e-4), \
+                (other - st.lm_head.weight).abs().max()
+    finally:
+        topo.reset()
+
+
+def test_pp_sft_end_to_end():
+    _spawn(_worker_pp_sft_train, 29529)

@@ -150,10 +150,17 @@ class GradReducer:
         finally:
             self._sync = prev
 
-    def broadcast_parameters(self, model: torch.nn.Module, src: int = 0):
-        """Ensure identical initial weights across ranks (random-init runs)."""
+    def broadcast_parameters(self, model: torch.nn.Module, src: int = None):
+        """Ensure identical initial weights across ranks (random-init runs).
+        ``src`` is a GLOBAL rank; defaults to the group's first member (the
+        group may not contain global rank 0 under model parallelism)."""
         if not self.enabled:
             return
+        if src is None:
+            if self.group is not None:
+                src = dist.get_process_group_ranks(self.group)[0]
+            else:
+                src = 0
         for p in model.state_dict().values():
             if isinstance(p, torch.Tensor):
                 dist.broadcast(p, src, group=self.group)

@@ -186,11 +186,50 @@ class FusedAdamW(torch.optim.Optimizer):
         return out
 
 
+class NullOptimizer:
+    """Stand-in for ranks with NO trainable parameters (e.g. a fully frozen
+    pipeline stage when ``num_layers_unfrozen`` keeps all trainable layers on
+    the last stage).  step/zero_grad are no-ops; pairs with NullScheduler."""
+
+    def __init__(self):
+        self.param_groups = []
+
+    def step(self, closure=None):
+        pass
+
+    def zero_grad(self, set_to_none: bool = True):
+        pass
+
+    def state_dict(self):
+        return {}
+
+    def load_state_dict(self, state):
+        pass
+
+
+class NullScheduler:
+    """LR scheduler stand-in paired with NullOptimizer."""
+
+    def step(self):
+        pass
+
+    def get_last_lr(self):
+        return []
+
+    def state_dict(self):
+        return {}
+
+    def load_state_dict(self, state):
+        pass
+
+
 def build_optimizer(model: torch.nn.Module, name: str, kwargs: dict, world: int = 1,
                     zero: bool = False):
     """Construct the optimizer named in the config (reference
     utils/__init__.py get_optimizer_class registry)."""
     params = [p for p in model.parameters() if p.requires_grad]
+    if not params:
+        return NullOptimizer()
     kwargs = dict(kwargs)
     if name in ("fused_adamw", "adamw"):
         from . import topo

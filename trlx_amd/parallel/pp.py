@@ -60,7 +60,14 @@ class PipelineStage(nn.Module):
                 self.embed_positions = nn.Embedding(cfg.max_position_embeddings + off, cfg.hidden_size)
             else:
                 self.embed_positions = None
-        self.layers = nn.ModuleList(Block(cfg, i) for i in range(self.lo, self.hi))
+        # LOCAL layer indices: Block.layer_idx is only the KV-cache slot, and
+        # each stage owns its own cache of len(self.layers) entries
+        self.layers = nn.ModuleList(Block(cfg, i - self.lo) for i in range(self.lo, self.hi))
+        # set to a local layer index to stash the true residual stream fed
+        # into that layer (consumed by the hydra frozen branch on the last
+        # stage); read back from ``self.last_stash`` right after forward()
+        self.stash_local_layer: Optional[int] = None
+        self.last_stash = None
         if self.is_last:
             self.final_norm = Norm(cfg)
             self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=cfg.lm_head_bias)
@@ -111,7 +118,7 @@ class PipelineStage(nn.Module):
                             .unsqueeze(0).expand(B, T).contiguous())
         return AttentionContext(position_ids=position_ids, key_starts=key_starts, start_pos=0)
 
-    def forward(self, hidden_or_ids, ctx):
+    def forward(self, hidden_or_ids, ctx, kv_cache=None):
         if self.is_first:
             h = self.embed_tokens(hidden_or_ids)
             if self.embed_positions is not None:
@@ -120,8 +127,13 @@ class PipelineStage(nn.Module):
         else:
             h = hidden_or_ids
         res = None
-        for layer in self.layers:
-            h, res = layer(h, ctx, self.rope_tables, res=res)
+        self.last_stash = None
+        for i, layer in enumerate(self.layers):
+            if self.stash_local_layer is not None and i == self.stash_local_layer:
+                self.last_stash = h if res is None else h + res
+            h, res = layer(h, ctx, self.rope_tables, kv_cache, res=res)
+        if self.stash_local_layer is not None and self.stash_local_layer == len(self.layers):
+            self.last_stash = h if res is None else h + res
         if res is not None:
             # stage boundaries exchange the materialized stream
             h = h + res
@@ -132,6 +144,14 @@ class PipelineStage(nn.Module):
     def project(self, h):
         assert self.is_last
         return self.lm_head(h)
+
+    def new_kv_cache(self, batch: int, max_len: int, device=None, dtype=None):
+        from ..models.nn.transformer import KVCache
+
+        p = next(self.parameters())
+        kv_heads_local = self.config.num_kv_heads // topo.tp_size()
+        return KVCache(len(self.layers), batch, kv_heads_local, max_len,
+                       self.config.head_dim, device or p.device, dtype or p.dtype)
 
 
 class PipelineRunner:
@@ -178,13 +198,52 @@ class PipelineRunner:
     def forward_backward(self, microbatches, loss_fn, hidden_dtype=torch.float32):
         """Run pipelined fwd+bwd.  ``microbatches``: list of dicts with
         input_ids/attention_mask (used on stage 0 for embeddings and on every
-        stage for the attention context).  ``loss_fn(logits, mb) -> loss`` is
-        evaluated on the last stage.  Returns mean loss (last stage) or None.
-        Gradients accumulate into stage parameters; the caller averages over
-        microbatch count via the optimizer's grad scale."""
+        stage for the attention context).  ``loss_fn(h_last, mb)`` is
+        evaluated on the last stage with the POST-final-norm hidden state
+        (callers project to logits/values themselves); it returns a loss or a
+        ``(loss, stats_dict)`` pair.  Returns mean loss (last stage) or None;
+        per-microbatch stats land in ``self.last_stats``.  Gradients
+        accumulate into stage parameters (summed over microbatches, like the
+        single-process accumulation path)."""
+        self.last_stats = []
         if self.schedule == "1f1b":
             return self._forward_backward_1f1b(microbatches, loss_fn, hidden_dtype)
         return self._forward_backward_gpipe(microbatches, loss_fn, hidden_dtype)
+
+    def _eval_loss(self, h, mb, loss_fn, losses):
+        out = loss_fn(h, mb)
+        if isinstance(out, tuple):
+            loss, stats = out
+            self.last_stats.append(stats)
+        else:
+            loss = out
+        losses.append(loss.detach())
+        return loss
+
+    @torch.no_grad()
+    def forward_inference(self, input_ids, attention_mask=None, kv_cache=None,
+                          ctx=None, hidden_dtype=torch.float32):
+        """One no-grad chained forward through the pipeline (experience phase,
+        generation prefill/decode).  Every PP rank calls this with the SAME
+        input_ids; returns the post-final-norm hidden on the last stage, None
+        elsewhere."""
+        stage = self.stage
+        device = next(stage.parameters()).device
+        ids = input_ids.to(device)
+        if ctx is None:
+            mask = attention_mask.to(device) if attention_mask is not None else None
+            ctx = stage.make_context(ids, mask)
+        if stage.is_first:
+            inp = ids
+        else:
+            B, T = ids.shape[:2]
+            inp = self._recv((B, T, stage.config.hidden_size), hidden_dtype, device, self.prev)
+        out = stage(inp, ctx, kv_cache=kv_cache)
+        if not stage.is_last:
+            self._send(out, self.next)
+            self._drain_sends()
+            return None
+        return out
 
     def _forward_backward_1f1b(self, microbatches, loss_fn, hidden_dtype=torch.float32):
         stage = self.stage
@@ -215,8 +274,7 @@ class PipelineRunner:
                 inp = h_in
             out = stage(inp, ctx)
             if stage.is_last:
-                loss = loss_fn(stage.project(out), mb)
-                losses.append(loss.detach())
+                loss = self._eval_loss(out, mb, loss_fn, losses)
                 stashes.append((h_in, loss))
             else:
                 self._send(out.detach(), self.next)
@@ -228,7 +286,9 @@ class PipelineRunner:
                 out_or_loss.backward()
             else:
                 grad = self._recv(out_or_loss.shape, hidden_dtype, device, self.next)
-                out_or_loss.backward(grad)
+                if out_or_loss.requires_grad:
+                    out_or_loss.backward(grad)
+                # else: fully frozen first stage — grad is dropped
             if not stage.is_first:
                 self._send(h_in.grad, self.prev)
 
@@ -257,6 +317,12 @@ class PipelineRunner:
         if not stage.config.tie_word_embeddings or len(self.pp_ranks) < 2:
             return
         if stage.is_first and stage.is_last:
+            return
+        # frozen tied pairs (e.g. num_layers_unfrozen freezing the embeddings
+        # AND the tied lm_head) take no updates — nothing to synchronize
+        if stage.is_first and not stage.embed_tokens.weight.requires_grad:
+            return
+        if stage.is_last and not stage.lm_head.weight.requires_grad:
             return
         first, last = self.pp_ranks[0], self.pp_ranks[-1]
         if stage.is_first:
@@ -298,9 +364,7 @@ class PipelineRunner:
                 inp = h_in
             out = stage(inp, ctx)
             if stage.is_last:
-                logits = stage.project(out)
-                loss = loss_fn(logits, mb)
-                losses.append(loss)
+                loss = self._eval_loss(out, mb, loss_fn, losses)
                 stashes.append((h_in, loss))
             else:
                 self._send(out.detach(), self.next)
@@ -312,7 +376,9 @@ class PipelineRunner:
                 out_or_loss.backward()
             else:
                 grad = self._recv(out_or_loss.shape, hidden_dtype, device, self.next)
-                out_or_loss.backward(grad)
+                if out_or_loss.requires_grad:
+                    out_or_loss.backward(grad)
+                # else: fully frozen first stage — grad is dropped
             if not stage.is_first:
                 self._send(h_in.grad, self.prev)
 

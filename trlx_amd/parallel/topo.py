@@ -33,11 +33,11 @@ def state() -> ParallelState:
 
 
 def init_model_parallel(tp_size: int = 1, pp_size: int = 1) -> ParallelState:
-    """Create TP/DP groups.  Safe to call with tp_size=1 (no-op topology)."""
+    """Create TP/PP/DP groups.  Safe to call with tp=pp=1 (no-op topology)."""
     s = _STATE
     s.tp_size = tp_size
     s.pp_size = pp_size
-    if not dist.is_initialized() or tp_size <= 1:
+    if not dist.is_initialized() or (tp_size <= 1 and pp_size <= 1):
         s.tp_group = None
         s.dp_group = None
         s.tp_rank = 0
@@ -54,12 +54,15 @@ def init_model_parallel(tp_size: int = 1, pp_size: int = 1) -> ParallelState:
     block = tp_size * pp_size  # ranks per model replica
 
     # TP groups: contiguous rank blocks (intra-node xGMI locality)
-    for b in range(world // tp_size):
-        ranks = list(range(b * tp_size, (b + 1) * tp_size))
-        g = dist.new_group(ranks)
-        if rank in ranks:
-            s.tp_group = g
-            s.tp_rank = ranks.index(rank)
+    s.tp_group = None
+    s.tp_rank = 0
+    if tp_size > 1:
+        for b in range(world // tp_size):
+            ranks = list(range(b * tp_size, (b + 1) * tp_size))
+            g = dist.new_group(ranks)
+            if rank in ranks:
+                s.tp_group = g
+                s.tp_rank = ranks.index(rank)
     # PP groups: stages tp_size apart within a replica block
     s.pp_group = None
     s.pp_rank = 0

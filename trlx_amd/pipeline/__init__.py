@@ -40,17 +40,24 @@ def register_datapipeline(name):
 
 
 def dp_sampler(dataset, shuffle: bool = False, seed: int = 0):
-    """A DistributedSampler over the DATA-parallel group, or None when dp=1.
+    """A DistributedSampler over the DATA-parallel group, or None when
+    single-process.
 
     Each DP rank iterates a disjoint shard per epoch (the reference shards via
     accelerate.prepare); TP/PP peers share a dp_rank and therefore see
-    identical data.  Callers must pass ``shuffle=False`` to the DataLoader
-    when a sampler is returned and bump ``sampler.set_epoch`` across epochs
+    identical data.  Returned for ANY multi-process run (even dp_size == 1,
+    where it degenerates to a deterministic full-coverage shuffler): model-
+    parallel peers must draw identical batches, and the global torch RNG has
+    diverged across ranks by init time (per-stage/per-shard weight init).
+    Callers must pass ``shuffle=False`` to the DataLoader when a sampler is
+    returned and bump ``sampler.set_epoch`` across epochs
     (``utils.infinite_dataloader`` does this).
     """
+    import torch.distributed as dist
+
     from ..parallel import topo
 
-    if topo.dp_size() <= 1:
+    if not dist.is_initialized() or dist.get_world_size() <= 1:
         return None
     from torch.utils.data import DistributedSampler
 

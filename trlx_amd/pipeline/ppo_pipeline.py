@@ -55,7 +55,8 @@ class BatchedRolloutLoader:
     """
 
     def __init__(self, batches: List[Tuple[PPORLBatch, torch.Tensor]], batch_size: int,
-                 shuffle: bool, pad_token_id: int):
+                 shuffle: bool, pad_token_id: int, generator: torch.Generator = None):
+        self.generator = generator
         assert batches
         qw = max(b.query_tensors.shape[1] for b, _ in batches)
         rw = max(b.response_tensors.shape[1] for b, _ in batches)
@@ -85,7 +86,8 @@ class BatchedRolloutLoader:
         return (self.n + self.batch_size - 1) // self.batch_size
 
     def __iter__(self):
-        order = torch.randperm(self.n) if self.shuffle else torch.arange(self.n)
+        order = (torch.randperm(self.n, generator=self.generator) if self.shuffle
+                 else torch.arange(self.n))
         device = self.query_tensors.device
         for i in range(0, self.n, self.batch_size):
             idx = order[i : i + self.batch_size]
@@ -168,11 +170,18 @@ class PPORolloutStorage(BaseRolloutStore):
     def __len__(self) -> int:
         return len(self.history) + sum(b.query_tensors.shape[0] for b, _ in self.batches)
 
-    def create_loader(self, batch_size: int, shuffle: bool):
+    def create_loader(self, batch_size: int, shuffle: bool, seed: int = None):
+        """``seed`` pins the shuffle order — REQUIRED under model parallelism
+        (TP/PP peers hold identical rollouts and must draw identical
+        minibatches; the global RNG has diverged across ranks)."""
+        generator = None
+        if seed is not None:
+            generator = torch.Generator().manual_seed(seed)
         if self.batches:
             assert not self.history, "mixing push() and push_batch() is unsupported"
-            return BatchedRolloutLoader(self.batches, batch_size, shuffle, self.pad_token_id)
+            return BatchedRolloutLoader(self.batches, batch_size, shuffle, self.pad_token_id,
+                                        generator=generator)
         return DataLoader(
-            self, batch_size, shuffle=shuffle,
+            self, batch_size, shuffle=shuffle, generator=generator,
             collate_fn=partial(ppo_collate_fn, self.padding_side, self.pad_token_id),
         )

@@ -54,17 +54,11 @@ class NativeRLTrainer(BaseRLTrainer):
         comm.init_distributed()
         from ..parallel import topo
 
-        if config.train.pipeline_parallel_size > 1:
-            # PP exists as an equivalence-tested library (parallel/pp.py:
-            # stage splitting, 1F1B/GPipe schedules, tied-embedding sync,
-            # full-state-dict resharding) but the RL trainer loop is not yet
-            # built on it (rollout generation across stages is the blocker —
-            # NOTES_ROUND2.md).  Refuse loudly rather than silently training
-            # replicated models that never synchronize.
+        if (config.train.pipeline_parallel_size > 1
+                and config.train.sequence_parallel):
             raise NotImplementedError(
-                "pipeline_parallel_size > 1 is not wired into the RL trainers yet; "
-                "use parallel.pp.PipelineRunner directly for pipelined supervised "
-                "losses, or TP/DP/ZeRO for RL runs (round-2 item)."
+                "sequence_parallel + pipeline_parallel_size > 1 is not supported yet; "
+                "disable one of them."
             )
         topo.init_model_parallel(config.train.tensor_parallel_size,
                                  config.train.pipeline_parallel_size)
@@ -73,6 +67,7 @@ class NativeRLTrainer(BaseRLTrainer):
         self.dp_size = topo.dp_size()
         self.dp_group = topo.dp_group()
         self.tp_size = topo.tp_size()
+        self.pp_size = topo.pp_size()
         self.local_rank = comm.local_rank()
         # re-seed with the DP rank so TP peers share RNG streams (identical
         # rollouts within a TP group, decorrelated across DP replicas —
@@ -156,6 +151,14 @@ class NativeRLTrainer(BaseRLTrainer):
 
     def setup_model(self):
         model = self.get_arch(self.config)
+        if getattr(model, "is_pipelined", False):
+            # PP model owns its stage-aware freezing; SP/checkpointing toggles
+            # are guarded at config validation
+            model.freeze_bottom(self.config.model.num_layers_unfrozen)
+            model = model.to(self.device)
+            if self.dtype != torch.float32:
+                model.cast_compute(self.dtype)
+            return model
         base = model.base_model if hasattr(model, "base_model") else model
         if self.config.model.peft_config is not None:
             pass  # LoRA already froze the base
@@ -186,6 +189,10 @@ class NativeRLTrainer(BaseRLTrainer):
         )
 
     def setup_scheduler(self):
+        from ..parallel.optim import NullOptimizer, NullScheduler
+
+        if isinstance(self.opt, NullOptimizer):
+            return NullScheduler()
         cls = get_scheduler_class(self.config.scheduler.name)
         return cls(self.opt, **self.config.scheduler.kwargs)
 
@@ -272,20 +279,42 @@ class NativeRLTrainer(BaseRLTrainer):
     # --- persistence ----------------------------------------------------------
 
     def save_pretrained(self, directory: Optional[str] = None, **kwargs):
+        from ..parallel import topo
+
         if directory is None:
             directory = os.path.join(self.config.train.checkpoint_dir, "hf_model")
         comm.barrier()
-        if comm.is_main_process():
+        if self.tp_size > 1 or self.pp_size > 1:
+            # sharded formats: every model-parallel rank of DP replica 0
+            # writes its own mp_rank_XX[_YYY] shard
+            if topo.dp_rank() == 0:
+                self.unwrapped_model.save_pretrained(directory)
+            if comm.is_main_process() and self.tokenizer:
+                self.tokenizer.save_pretrained(directory)
+        elif comm.is_main_process():
             self.unwrapped_model.save_pretrained(directory)
             if self.tokenizer:
                 self.tokenizer.save_pretrained(directory)
         comm.barrier()
 
+    def _state_file(self) -> str:
+        """Per-model-parallel-rank training-state filename (plain ``state.pt``
+        when there is no model parallelism)."""
+        from ..parallel import topo
+
+        if self.tp_size > 1 or self.pp_size > 1:
+            return f"state_mp_{topo.tp_rank():02d}_{topo.pp_rank():03d}.pt"
+        return "state.pt"
+
     def save(self, directory: Optional[str] = None, **kwargs):
-        """Training-state checkpoint: model + optimizer + scheduler + step."""
+        """Training-state checkpoint: model + optimizer + scheduler + step.
+        Under TP/PP every model-parallel rank of DP replica 0 writes its own
+        state shard."""
+        from ..parallel import topo
+
         dst_dir = directory or self.config.train.checkpoint_dir
         comm.barrier()
-        if comm.is_main_process():
+        if topo.dp_rank() == 0:
             os.makedirs(dst_dir, exist_ok=True)
             state = {
                 "model": self.unwrapped_model.state_dict(),
@@ -294,14 +323,15 @@ class NativeRLTrainer(BaseRLTrainer):
             }
             if self.config.train.save_optimizer:
                 state["optimizer"] = self.opt.state_dict()
-            torch.save(state, os.path.join(dst_dir, "state.pt"))
-            with open(os.path.join(dst_dir, "state.json"), "w") as f:
-                json.dump({"iter_count": self.iter_count}, f)
+            torch.save(state, os.path.join(dst_dir, self._state_file()))
+            if comm.is_main_process():
+                with open(os.path.join(dst_dir, "state.json"), "w") as f:
+                    json.dump({"iter_count": self.iter_count}, f)
         comm.barrier()
 
     def load(self, directory: Optional[str] = None, **kwargs):
         src_dir = directory or self.config.train.checkpoint_dir
-        path = os.path.join(src_dir, "state.pt")
+        path = os.path.join(src_dir, self._state_file())
         state = torch.load(path, map_location=self.device, weights_only=True)
         self.unwrapped_model.load_state_dict(state["model"])
         if "optimizer" in state and self.config.train.save_optimizer:
@@ -548,25 +578,35 @@ class NativeRLTrainer(BaseRLTrainer):
                     forward_time = 0.0
                     backward_time = 0.0
                     stats_accum = []
-                    for microbatch in minibatch:
-                        with self._accumulate():
-                            graphed = self._graphed_loss_backward(microbatch)
-                            if graphed is not None:
+                    if self.pp_size > 1:
+                        # pipelined fwd+bwd over the whole minibatch (1F1B);
+                        # grads accumulate inside, reduced once in finalize()
+                        forward_time -= time()
+                        with self.reducer.no_sync():
+                            _loss, stats = self.pp_train_minibatch(list(minibatch))
+                        forward_time += time()
+                        stats_accum.append(stats)
+                    else:
+                        for microbatch in minibatch:
+                            with self._accumulate():
+                                graphed = self._graphed_loss_backward(microbatch)
+                                if graphed is not None:
+                                    forward_time -= time()
+                                    loss, stats = graphed
+                                    forward_time += time()
+                                    stats_accum.append(stats)
+                                    continue
                                 forward_time -= time()
-                                loss, stats = graphed
+                                loss, stats = self.loss(microbatch)
                                 forward_time += time()
+                                backward_time -= time()
+                                self.backward(loss)
+                                backward_time += time()
                                 stats_accum.append(stats)
-                                continue
-                            forward_time -= time()
-                            loss, stats = self.loss(microbatch)
-                            forward_time += time()
-                            backward_time -= time()
-                            self.backward(loss)
-                            backward_time += time()
-                            stats_accum.append(stats)
                     forward_time /= self.num_mb
                     backward_time /= self.num_mb
-                    stats = {k: sum(s[k] for s in stats_accum) / self.num_mb for k in stats_accum[0]}
+                    stats = {k: sum(s[k] for s in stats_accum) / len(stats_accum)
+                             for k in stats_accum[0]}
 
                     self.reducer.finalize()
                     self.opt.step()
@@ -638,6 +678,13 @@ class NativeRLTrainer(BaseRLTrainer):
     @abstractmethod
     def loss(self, batch) -> Tuple[torch.Tensor, Dict]:
         pass
+
+    def pp_train_minibatch(self, microbatches) -> Tuple[float, Dict]:
+        """Pipelined fwd+bwd for one minibatch (list of microbatches); only
+        trainers that support pipeline parallelism implement this."""
+        raise NotImplementedError(
+            f"{type(self).__name__} does not support pipeline_parallel_size > 1"
+        )
 
     @abstractmethod
     def prepare_learning(self):

@@ -60,7 +60,16 @@ class PPOTrainer(NativeRLTrainer):
 
         # separate full reference model only without a hydra branch or peft
         # (reference accelerate_ppo_trainer.py:74-77)
-        if (hasattr(self.model, "frozen_head") and self.model.frozen_head is not None) or \
+        if getattr(self.model, "is_pipelined", False):
+            # PP: the hydra branch on the last stage is the reference; a
+            # second pipelined ref model is not supported
+            if self.config.model.num_layers_unfrozen <= 0:
+                raise NotImplementedError(
+                    "PPO under pipeline parallelism requires num_layers_unfrozen > 0 "
+                    "(the hydra frozen branch on the last stage is the reference model)"
+                )
+            self.ref_model = None
+        elif (hasattr(self.model, "frozen_head") and self.model.frozen_head is not None) or \
                 self.config.model.peft_config is not None:
             self.ref_model = None
         else:
@@ -83,6 +92,18 @@ class PPOTrainer(NativeRLTrainer):
         self.mean_kl = 0.0
 
     def get_arch(self, config: TRLConfig):
+        if config.train.pipeline_parallel_size > 1:
+            if config.model.model_arch_type == "seq2seq":
+                raise NotImplementedError("seq2seq + pipeline parallelism is not supported")
+            from ..models.modeling_pp import PipelinedPPOModel
+            from ..models.nn.config import TransformerConfig
+
+            path = config.model.model_path
+            if (isinstance(path, str) and not os.path.isdir(path)
+                    and config.model.model_extra_configs.get("config")):
+                path = TransformerConfig.from_dict(config.model.model_extra_configs["config"])
+            return PipelinedPPOModel.from_any(
+                path, num_layers_unfrozen=config.model.num_layers_unfrozen)
         if config.model.model_arch_type == "seq2seq":
             from ..models.modeling_seq2seq import (
                 AutoModelForSeq2SeqLMWithHydraValueHead,
@@ -204,6 +225,49 @@ class PPOTrainer(NativeRLTrainer):
         )
         return loss, stats
 
+    def pp_train_minibatch(self, microbatches):
+        """PPO loss under pipeline parallelism: 1F1B microbatched fwd+bwd with
+        the loss evaluated on the LAST stage from its post-norm hidden states
+        (reference nemo training_step + get_forward_output_and_loss_func,
+        modeling_nemo_ppo.py:652-786, 945-1026).  Returns (loss, stats) —
+        host floats, identical on every PP rank."""
+        method = self.config.method
+        model = self.model
+        pad_id = self.tokenizer.pad_token_id
+        device = self.device
+
+        mbs = []
+        for batch in microbatches:
+            tokens = torch.cat((batch.query_tensors.to(device),
+                                batch.response_tensors.to(device)), dim=1)
+            mask = tokens.not_equal(pad_id).long()
+            mbs.append({"input_ids": tokens, "attention_mask": mask, "_batch": batch})
+
+        def loss_fn(h, mb):
+            batch = mb["_batch"]
+            old_logprobs = batch.logprobs.to(device)
+            old_values = batch.values.to(device)
+            old_rewards = batch.rewards.to(device)
+            response_length = old_rewards.shape[1]
+            advantages, returns = method.get_advantages_and_returns(
+                old_values, old_rewards, response_length)
+            start = batch.query_tensors.shape[1] - 1
+            end = start + response_length
+            hs = h[:, start:end].contiguous()
+            labels = mb["input_ids"][:, start + 1 : end + 1]
+            logprobs = logprobs_of_labels(model.stage.project(hs), labels)
+            values_pred = model.v_head(hs.to(model.v_head[0].weight.dtype)).squeeze(-1).float()
+            mask_sl = mb["attention_mask"][:, start + 1 : end + 1]
+            loss, stats = method.loss(
+                logprobs=logprobs, values=values_pred, old_logprobs=old_logprobs,
+                old_values=old_values, advantages=advantages, returns=returns, mask=mask_sl,
+            )
+            # host floats: the (loss, stats) pair is object-broadcast to the
+            # other PP stages after the pipeline drains
+            return loss, {k: float(v) for k, v in stats.items()}
+
+        return model.forward_backward(mbs, loss_fn)
+
     # --- plumbing ----------------------------------------------------------------
 
     def setup_rollout_logging(self, config):
@@ -227,7 +291,13 @@ class PPOTrainer(NativeRLTrainer):
         self.kl_ctl.update(self.mean_kl, n_steps=self.config.train.batch_size)
 
     def create_train_dataloader(self):
-        return self.store.create_loader(self.config.train.batch_size, shuffle=True)
+        # deterministic per-inner-epoch shuffle seed, identical across ranks:
+        # TP/PP peers must draw identical minibatch orders from their
+        # identical stores (each DP rank's store holds different data, so DP
+        # decorrelation is unaffected)
+        self._loader_seed = getattr(self, "_loader_seed", self.config.train.seed) + 1
+        return self.store.create_loader(self.config.train.batch_size, shuffle=True,
+                                        seed=self._loader_seed)
 
     def prepare_learning(self):
         self.eval_dataloader = self.eval_pipeline.create_loader(self.config.method.chunk_size)
@@ -270,10 +340,10 @@ class PPOTrainer(NativeRLTrainer):
             metadata_local = {k: v for k, v in batch.items()
                               if k not in ("input_ids", "attention_mask")}
 
-            if topo.tp_size() > 1:
-                # tensor-parallel mode: every rank scores its own samples
-                # locally (TP peers hold identical rollouts) — the NeMo-path
-                # protocol (reference nemo_ppo_trainer.py:195-197)
+            if topo.tp_size() > 1 or topo.pp_size() > 1:
+                # model-parallel mode: every rank scores its own samples
+                # locally (TP/PP peers hold identical rollouts) — the
+                # NeMo-path protocol (reference nemo_ppo_trainer.py:195-197)
                 l_samples, l_prompts, l_outputs = self.decode(
                     prompt_tensors, samples, append_eos_token=True)
                 rollout_score_time = time()
@@ -324,7 +394,7 @@ class PPOTrainer(NativeRLTrainer):
                     scores = all_scores[0].clone().detach()
             scores_mask = scores != -np.inf
 
-            if topo.tp_size() <= 1 and self.world_size == 1:
+            if topo.tp_size() <= 1 and topo.pp_size() <= 1 and self.world_size == 1:
                 # single process: the gathered batch IS the local batch
                 str_samples, str_prompts, str_outputs = (
                     all_str_samples, all_str_prompts, all_str_outputs)

@@ -51,6 +51,15 @@ class SFTTrainer(NativeRLTrainer):
 
     def get_arch(self, config: TRLConfig):
         path = config.model.model_path
+        if config.train.pipeline_parallel_size > 1:
+            from ..models.modeling_pp import PipelinedPPOModel
+            from ..models.nn.config import TransformerConfig
+
+            if isinstance(path, str) and config.model.model_extra_configs.get("config"):
+                path = TransformerConfig.from_dict(config.model.model_extra_configs["config"])
+            return PipelinedPPOModel.from_any(
+                path, num_layers_unfrozen=config.model.num_layers_unfrozen,
+                with_value_head=False)
         if isinstance(path, str) and config.model.model_extra_configs.get("config"):
             from ..models.nn.config import TransformerConfig
 
@@ -58,6 +67,37 @@ class SFTTrainer(NativeRLTrainer):
                 TransformerConfig.from_dict(config.model.model_extra_configs["config"])
             )
         return CausalLMWrapper.from_pretrained(path)
+
+    def pp_train_minibatch(self, microbatches):
+        """SFT CE loss under pipeline parallelism (reference NeMo SFT path,
+        modeling_nemo_sft.py training_step skeleton)."""
+        model = self.model
+        device = self.device
+        mbs = []
+        for batch in microbatches:
+            if isinstance(batch, dict):
+                ids = batch["input_ids"].to(device)
+                mask = batch.get("attention_mask")
+                labels = batch.get("labels", ids)
+            else:
+                ids = batch.input_ids.to(device)
+                mask = getattr(batch, "attention_mask", None)
+                labels = ids
+            mask = mask.to(device) if mask is not None else torch.ones_like(ids)
+            mbs.append({"input_ids": ids, "attention_mask": mask,
+                        "_labels": labels.to(device)})
+
+        def loss_fn(h, mb):
+            logits = model.stage.project(h)[:, :-1, :].float()
+            shift_labels = mb["_labels"][:, 1:].clone()
+            shift_labels[mb["attention_mask"][:, 1:] == 0] = -100
+            loss = F.cross_entropy(
+                logits.reshape(-1, logits.shape[-1]), shift_labels.reshape(-1),
+                ignore_index=-100,
+            )
+            return loss, {"loss": float(loss)}
+
+        return model.forward_backward(mbs, loss_fn)
 
     def loss(self, batch):
         if isinstance(batch, dict):
