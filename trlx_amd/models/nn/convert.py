@@ -640,6 +640,7 @@ def load_hf_dir(path: str) -> Tuple[TransformerConfig, Dict[str, torch.Tensor]]:
     cfg = config_from_hf(hf_cfg)
     sd = {}
     st_index = os.path.join(path, "model.safetensors.index.json")
+    pt_index = os.path.join(path, "pytorch_model.bin.index.json")
     st_single = os.path.join(path, "model.safetensors")
     pt_single = os.path.join(path, "pytorch_model.bin")
     if os.path.exists(st_index):
@@ -649,6 +650,14 @@ def load_hf_dir(path: str) -> Tuple[TransformerConfig, Dict[str, torch.Tensor]]:
             index = json.load(f)
         for shard in sorted(set(index["weight_map"].values())):
             sd.update(safetensors.torch.load_file(os.path.join(path, shard)))
+    elif os.path.exists(pt_index):
+        # sharded torch checkpoint (reference modeling_base.py:276-311 merges
+        # the index's shards the same way)
+        with open(pt_index) as f:
+            index = json.load(f)
+        for shard in sorted(set(index["weight_map"].values())):
+            sd.update(torch.load(os.path.join(path, shard), map_location="cpu",
+                                 weights_only=True))
     elif os.path.exists(st_single):
         import safetensors.torch
 

@@ -143,3 +143,43 @@ def all_reduce_max_flag(flag: bool, device=None) -> bool:
     t = torch.tensor(int(flag), device=device or get_device())
     dist.all_reduce(t, dist.ReduceOp.MAX)
     return bool(t.item())
+
+
+def preflight(log=print) -> bool:
+    """Communication-environment sanity check at job startup (bench/trainer):
+    logs the topology and verifies one small all-reduce / broadcast /
+    all-gather with value checks over the default group.  Cheap (three
+    scalar-ish collectives) and loud: first contact with a fresh multi-GPU
+    node fails HERE with a clear message instead of deep inside training.
+    Returns True when the collectives verify."""
+    if not dist.is_initialized():
+        return True
+    w = world_size()
+    r = rank()
+    dev = get_device()
+    if r == 0:
+        info = {
+            "backend": dist.get_backend(),
+            "world_size": w,
+            "devices_visible": torch.cuda.device_count() if torch.cuda.is_available() else 0,
+            "rccl_env": {k: v for k, v in os.environ.items()
+                         if k.startswith(("NCCL_", "RCCL_", "HSA_"))},
+        }
+        log(f"[comm.preflight] {info}")
+    t = torch.tensor([float(r + 1)], device=dev)
+    dist.all_reduce(t)
+    expect = w * (w + 1) / 2
+    if abs(float(t.item()) - expect) > 1e-4:
+        raise RuntimeError(f"[comm.preflight] all_reduce wrong: got {t.item()}, want {expect}")
+    b = torch.tensor([42.5 if r == 0 else 0.0], device=dev)
+    dist.broadcast(b, 0)
+    if abs(float(b.item()) - 42.5) > 1e-6:
+        raise RuntimeError(f"[comm.preflight] broadcast wrong: got {b.item()}")
+    buf = [torch.zeros(1, device=dev) for _ in range(w)]
+    dist.all_gather(buf, torch.tensor([float(r)], device=dev))
+    got = [int(x.item()) for x in buf]
+    if got != list(range(w)):
+        raise RuntimeError(f"[comm.preflight] all_gather wrong: {got}")
+    if r == 0:
+        log(f"[comm.preflight] OK: all_reduce/broadcast/all_gather verified over {w} ranks")
+    return True
