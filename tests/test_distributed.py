@@ -886,3 +886,47 @@ def _worker_pp_sft_train(rank):
 
 def test_pp_sft_end_to_end():
     _spawn(_worker_pp_sft_train, 29529)
+
+
+def _worker_cross_tp_reshard(rank):
+    """Checkpoint saved at TP=2 reloads at TP=4 (merge -> re-cut): the
+    cross-TP-size resharding the NeMo path implies but round 1 lacked."""
+    import os as _os
+
+    from trlx_amd.models.modeling_ppo import AutoModelForCausalLMWithValueHead
+    from trlx_amd.models.nn.config import TransformerConfig
+    from trlx_amd.models.nn.transformer import CausalTransformer
+    from trlx_amd.parallel import topo
+    from trlx_amd.parallel.tp import shard_state_dict_tp
+
+    torch.manual_seed(0)
+    cfg = TransformerConfig(vocab_size=128, hidden_size=64, num_layers=2, num_heads=4,
+                            intermediate_size=128, max_position_embeddings=64,
+                            arch_name="gpt2", tie_word_embeddings=False)
+    full = CausalTransformer(cfg)
+    full_sd = {k: v for k, v in full.state_dict().items() if not k.startswith("rope_")}
+    out_dir = "/tmp/cross_tp_ckpt_test"
+
+    topo.reset()
+    try:
+        topo.init_model_parallel(tp_size=2)
+        sharded = CausalTransformer(cfg)
+        sharded.load_state_dict(
+            shard_state_dict_tp(full_sd, cfg, topo.tp_rank(), 2), strict=False)
+        model = AutoModelForCausalLMWithValueHead(sharded)
+        model.save_pretrained(out_dir)
+        dist.barrier()
+        topo.reset()
+
+        topo.init_model_parallel(tp_size=4)
+        loaded = AutoModelForCausalLMWithValueHead.from_pretrained(out_dir)
+        want = shard_state_dict_tp(full_sd, cfg, topo.tp_rank(), 4)
+        got = loaded.base_model.state_dict()
+        for k, v in want.items():
+            assert torch.allclose(got[k], v, atol=1e-6), k
+    finally:
+        topo.reset()
+
+
+def test_cross_tp_size_resharding():
+    _spawn_n(_worker_cross_tp_reshard, 29530, 4)

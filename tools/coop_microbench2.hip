@@ -25,9 +25,12 @@
 // All threads of the block participate via __syncthreads around thread 0's
 // atomic.  gen/cnt live in one cacheline-separated pair.
 struct GridBar {
-  unsigned int cnt;
+  unsigned int cnt;  // add-only arrival counter (never reset: reset races
+                     // with the next round's atomics across per-XCD L2s)
   unsigned int pad[31];
   unsigned int gen;
+  unsigned int pad2[31];
+  unsigned int fail;
 };
 
 template <bool FENCE>
@@ -37,14 +40,17 @@ __device__ __forceinline__ void grid_bar(GridBar* b, unsigned int nblocks,
   __syncthreads();
   if (threadIdx.x == 0) {
     const unsigned int g = *local_gen;
-    const unsigned int arrived =
-        __atomic_fetch_add(&b->cnt, 1u, __ATOMIC_ACQ_REL) + 1;
-    if (arrived == nblocks) {
-      b->cnt = 0;
+    const unsigned int arrived = __atomic_fetch_add(&b->cnt, 1u, __ATOMIC_ACQ_REL) + 1;
+    if (arrived == (g + 1) * nblocks) {
       __atomic_store_n(&b->gen, g + 1, __ATOMIC_RELEASE);
     } else {
-      while (__atomic_load_n(&b->gen, __ATOMIC_ACQUIRE) == g) {
-        __builtin_amdgcn_s_sleep(1);
+      long spins = 0;
+      while (__atomic_load_n(&b->gen, __ATOMIC_ACQUIRE) < g + 1) {
+        __builtin_amdgcn_s_sleep(8);
+        if (++spins > (1L << 24)) {  // bounded: a logic bug must not hang the box
+          atomicAdd(&b->fail, 1u);
+          break;
+        }
       }
     }
     *local_gen = g + 1;
@@ -55,10 +61,9 @@ __device__ __forceinline__ void grid_bar(GridBar* b, unsigned int nblocks,
 
 template <bool FENCE>
 __global__ void bar_cost_kernel(GridBar* bar, int iters, float* dummy) {
-  __shared__ unsigned int lg;
-  if (threadIdx.x == 0) lg = 0;
-  __syncthreads();
-  unsigned int local_gen = 0;
+  // continue from the persistent generation: back-to-back launches share the
+  // bar without a host-side reset (cnt is add-only)
+  unsigned int local_gen = __atomic_load_n(&bar->gen, __ATOMIC_ACQUIRE);
   for (int i = 0; i < iters; ++i) grid_bar<FENCE>(bar, gridDim.x, &local_gen);
   if (blockIdx.x == 0 && threadIdx.x == 0) *dummy = local_gen;
 }
@@ -70,7 +75,7 @@ __global__ void fence_cost_kernel(int iters, float* dummy) {
 
 // visibility under barrier A (same pattern as round 1 but hand barrier)
 __global__ void vis_kernel(GridBar* bar, int iters, unsigned int* buf, long n, int* errors) {
-  unsigned int local_gen = 0;
+  unsigned int local_gen = __atomic_load_n(&bar->gen, __ATOMIC_ACQUIRE);
   const long stride = (long)gridDim.x * blockDim.x;
   const long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
   for (int i = 0; i < iters; ++i) {

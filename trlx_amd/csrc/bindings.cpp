@@ -72,6 +72,22 @@ at::Tensor lm_logprobs(const at::Tensor& hidden, const at::Tensor& weight,
 void fused_adamw(at::Tensor& p, at::Tensor& master, const at::Tensor& g, at::Tensor& m,
                  at::Tensor& v, long step, double lr, double beta1, double beta2, double eps,
                  double weight_decay, double grad_scale);
+void stage_gemm(const at::Tensor& a, const at::Tensor& w, const c10::optional<at::Tensor>& bias,
+                at::Tensor& c, const c10::optional<at::Tensor>& nstats,
+                const c10::optional<at::Tensor>& nw, const c10::optional<at::Tensor>& nb,
+                bool norm_rms, double eps, long act, const c10::optional<at::Tensor>& resid,
+                const c10::optional<at::Tensor>& out_stats);
+void embed_stats(const at::Tensor& wte, const c10::optional<at::Tensor>& wpe,
+                 const at::Tensor& cur_tok, const at::Tensor& pos_ids, long pos_offset,
+                 at::Tensor& x, at::Tensor& stats, at::Tensor& packed);
+void lm_sample(const at::Tensor& x, const at::Tensor& wlm, const c10::optional<at::Tensor>& blm,
+               const at::Tensor& nstats, const at::Tensor& nw,
+               const c10::optional<at::Tensor>& nb, at::Tensor& packed, bool norm_rms,
+               double eps, double temperature, long seed, const at::Tensor& rng_offset);
+void advance_packed(const at::Tensor& packed, at::Tensor& out_tokens, at::Tensor& cur_tok,
+                    at::Tensor& finished, at::Tensor& rng_offset, at::Tensor& step_col,
+                    at::Tensor& cache_idx, at::Tensor& seq_lens, at::Tensor& pos_ids,
+                    const c10::optional<at::Tensor>& key_starts, long eos, long pad);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("logprobs_fwd", &logprobs_fwd, "fused logsumexp + label gather fwd");
@@ -100,4 +116,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ce_dlogits", &ce_dlogits);
   mod.def("decode_advance", &decode_advance);
   mod.def("fused_decode_attention", &fused_decode_attention);
+  mod.def("stage_gemm", &stage_gemm);
+  mod.def("embed_stats", &embed_stats);
+  mod.def("lm_sample", &lm_sample);
+  mod.def("advance_packed", &advance_packed);
 }

@@ -98,8 +98,12 @@ class PreTrainedModelWrapper(nn.Module):
                                  f"mp_rank_{topo.tp_rank():02d}", "model_weights.pt"),
                     map_location="cpu", weights_only=True)
                 model.load_state_dict(shard, strict=False)
-            elif topo.tp_size() == 1:
-                from ..parallel.tp import merge_state_dicts_tp
+            else:
+                # cross-TP-size resharding: merge the saved shards to a full
+                # state dict, then (for tp > 1) cut this rank's new shard —
+                # the NeMo-path checkpoint resharding generalized to any
+                # saved/live TP pair
+                from ..parallel.tp import merge_state_dicts_tp, shard_state_dict_tp
 
                 shards = [
                     torch.load(os.path.join(pretrained_model_name_or_path,
@@ -107,10 +111,10 @@ class PreTrainedModelWrapper(nn.Module):
                                map_location="cpu", weights_only=True)
                     for r in range(saved_tp)
                 ]
-                model.load_state_dict(merge_state_dicts_tp(shards, cfg, saved_tp), strict=False)
-            else:
-                raise ValueError(
-                    f"checkpoint TP size {saved_tp} != current TP {topo.tp_size()}")
+                full = merge_state_dicts_tp(shards, cfg, saved_tp)
+                if topo.tp_size() > 1:
+                    full = shard_state_dict_tp(full, cfg, topo.tp_rank(), topo.tp_size())
+                model.load_state_dict(full, strict=False)
             return model
         elif os.path.isdir(pretrained_model_name_or_path):
             cfg, sd = load_hf_dir(pretrained_model_name_or_path)

@@ -55,6 +55,78 @@ class GenerateConfig:
                 self.eos_token_id, self.pad_token_id)
 
 
+class FusedStageRefs:
+    """Weight references + scratch buffers for the fused stage-kernel decode
+    step (csrc/decode_mega.hip): 5 kernels per layer instead of ~12."""
+
+    def __init__(self, model, batch: int, device):
+        import torch.nn as nn
+
+        from ...parallel import topo
+
+        cfg = model.config
+        base_ok = (
+            not cfg.parallel_residual
+            and cfg.num_heads == cfg.num_kv_heads
+            and cfg.head_dim in (64, 128, 256)
+            and cfg.position_encoding in ("learned", "rope")
+            and not cfg.extra.get("pre_embed_norm")
+            and not cfg.swiglu
+            and cfg.hidden_size % 32 == 0
+            and cfg.intermediate_size % 32 == 0
+            and topo.tp_size() == 1
+            and model.embed_norm is None
+        )
+        if not base_ok:
+            raise ValueError("arch not fused-decode eligible")
+        from ... import ops
+
+        self.act_code = ops.ACT_CODES.get(cfg.activation)
+        if self.act_code is None:
+            raise ValueError(f"activation {cfg.activation} not fused-decode eligible")
+        for layer in model.layers:
+            for mod in (layer.attn.qkv_proj, layer.attn.o_proj, layer.mlp.fc_in,
+                        layer.mlp.down_proj):
+                if type(mod) is not nn.Linear or mod.weight.dtype != torch.bfloat16:
+                    raise ValueError("fused decode needs plain bf16 linears")
+        if model.lm_head.weight.dtype != torch.bfloat16:
+            raise ValueError("fused decode needs a bf16 lm_head")
+
+        H, I = cfg.hidden_size, cfg.intermediate_size
+        L = cfg.num_layers
+        self.cfg = cfg
+        self.rms = cfg.norm == "rmsnorm"
+        self.eps = cfg.norm_eps
+        self.scale = model.layers[0].attn.scale
+        self.rot = model.layers[0].attn.rot
+        self.pos_offset = cfg.extra.get("position_offset", 0)
+        self.wte = model.embed_tokens.weight
+        self.wpe = model.embed_positions.weight if model.embed_positions is not None else None
+        self.rcos, self.rsin = (model.rope_tables if cfg.position_encoding == "rope"
+                                else (None, None))
+        self.lnf_w = model.final_norm.weight
+        self.lnf_b = model.final_norm.bias
+        self.lm_w = model.lm_head.weight
+        self.lm_b = model.lm_head.bias
+        self.layers = []
+        for layer in model.layers:
+            self.layers.append(dict(
+                ln1_w=layer.ln_1.weight, ln1_b=layer.ln_1.bias,
+                qkv_w=layer.attn.qkv_proj.weight, qkv_b=layer.attn.qkv_proj.bias,
+                o_w=layer.attn.o_proj.weight, o_b=layer.attn.o_proj.bias,
+                ln2_w=layer.ln_2.weight, ln2_b=layer.ln_2.bias,
+                fc_w=layer.mlp.fc_in.weight, fc_b=layer.mlp.fc_in.bias,
+                down_w=layer.mlp.down_proj.weight, down_b=layer.mlp.down_proj.bias,
+            ))
+        bf = torch.bfloat16
+        self.x = torch.empty(batch, H, dtype=bf, device=device)
+        self.x1 = torch.empty(batch, H, dtype=bf, device=device)
+        self.qkv = torch.empty(batch, cfg.qkv_out, dtype=bf, device=device)
+        self.act = torch.empty(batch, I, dtype=bf, device=device)
+        self.stats = torch.zeros(2 * L + 1, batch, 2, dtype=torch.float32, device=device)
+        self.packed = torch.zeros(batch, dtype=torch.long, device=device)
+
+
 class DecodeEngine:
     """Persistent hipGraph-captured decode step."""
 
@@ -69,6 +141,16 @@ class DecodeEngine:
         self.device = device
         self.seed = gen.seed if gen.seed is not None else int(torch.randint(0, 2**31 - 1, (1,)).item())
         self.kv = model.new_kv_cache(batch, cache_len, device=device)
+        # fused stage-kernel step (5 kernels/layer): eligible archs only, and
+        # only for the plain sampler the lm_sample kernel implements
+        self.fused = None
+        if (shaping_fn is None and os.environ.get("TRLX_AMD_NO_FUSED_DECODE") != "1"
+                and gen.top_k in (0, None) and gen.top_p in (1.0, None)
+                and gen.min_new_tokens == 0):
+            try:
+                self.fused = FusedStageRefs(model, batch, device)
+            except (ValueError, AttributeError):
+                self.fused = None
         self.cur_tok = torch.zeros(batch, 1, dtype=torch.long, device=device)
         self.key_starts = torch.zeros(batch, dtype=torch.int32, device=device)
         self.seq_lens = torch.zeros(batch, dtype=torch.int32, device=device)
@@ -121,11 +203,50 @@ class DecodeEngine:
         self.pos_ids.copy_(
             (self.cache_idx - self.key_starts.to(torch.long)).to(torch.int32).unsqueeze(1))
 
+    def _fused_step(self):
+        """One decode token through the fused stage kernels
+        (csrc/decode_mega.hip): embed+stats, then per layer
+        [ln1-folded qkv GEMM] -> [fused decode attention] ->
+        [o GEMM + residual + ln2 stats] -> [ln2-folded fc GEMM + act] ->
+        [down GEMM + residual + next-ln1 stats], then the final-norm-folded
+        lm_head GEMM fused with Gumbel-max sampling, then advance — 63
+        launches/token for GPT-2 vs ~150 on the module path, each folding
+        the norm/bias/act/residual elementwise work into a GEMM epilogue."""
+        from ... import ops
+
+        ext = ops._load_ext()
+        f = self.fused
+        cur = self.cur_tok.view(-1)
+        pos = self.pos_ids.view(-1)
+        ext.embed_stats(f.wte, f.wpe, cur, pos, f.pos_offset, f.x, f.stats, f.packed)
+        for li, lay in enumerate(f.layers):
+            ext.stage_gemm(f.x, lay["qkv_w"], lay["qkv_b"], f.qkv, f.stats[2 * li],
+                           lay["ln1_w"], lay["ln1_b"], f.rms, f.eps, 0, None, None)
+            attn = ext.fused_decode_attention(
+                f.qkv, self.kv.k[li], self.kv.v[li], self.seq_lens, self.key_starts,
+                f.rcos, f.rsin, self.cache_idx, f.rot, f.cfg.rope_interleaved, f.scale,
+            ).view(self.batch, -1)
+            ext.stage_gemm(attn, lay["o_w"], lay["o_b"], f.x1, None, None, None,
+                           False, 0.0, 0, f.x, f.stats[2 * li + 1])
+            ext.stage_gemm(f.x1, lay["fc_w"], lay["fc_b"], f.act, f.stats[2 * li + 1],
+                           lay["ln2_w"], lay["ln2_b"], f.rms, f.eps, f.act_code, None, None)
+            ext.stage_gemm(f.act, lay["down_w"], lay["down_b"], f.x, None, None, None,
+                           False, 0.0, 0, f.x1, f.stats[2 * li + 2])
+        temperature = self.gen.temperature if self.gen.do_sample else 0.0
+        ext.lm_sample(f.x, f.lm_w, f.lm_b, f.stats[2 * len(f.layers)], f.lnf_w, f.lnf_b,
+                      f.packed, f.rms, f.eps, temperature, self.seed, self.rng_offset)
+        ext.advance_packed(f.packed, self.out_tokens, cur, self.finished, self.rng_offset,
+                           self.step_col, self.cache_idx, self.seq_lens, pos, self.key_starts,
+                           -1 if self.gen.eos_token_id is None else self.gen.eos_token_id,
+                           self.pad_id)
+
     def _step(self):
         """One decode token — everything device-side (hipGraph body).
 
         State convention: on entry, seq_lens/cache_idx/pos_ids already
         describe THIS step (the previous _advance set them)."""
+        if self.fused is not None:
+            return self._fused_step()
         out = self.model(
             self.cur_tok, kv_cache=self.kv, position_ids=self.pos_ids, seq_lens=self.seq_lens,
             key_starts=self.key_starts, cache_idx=self.cache_idx, start_pos=0,
