@@ -1,18 +1,30 @@
 """Hyperparameter sweep runner: ``python -m trlx_amd.sweep config.yml script.py``.
 
-Parity target: reference trlx/sweep.py (Ray Tune + W&B).  Ray and W&B are not
-available offline, so this implements the same YAML search-space grammar
-(strategy: grid / random / bayes-fallback-to-random) with a local
-subprocess-per-trial executor and a jsonl results summary.  Trials run
-sequentially (or torchrun-launched for multi-GPU) and report the tracker's
-last logged metrics.
+Parity target: reference trlx/sweep.py (Ray Tune search algorithms +
+schedulers + W&B reports).  Ray and W&B are not available offline, so the
+same YAML search-space grammar runs on a local subprocess executor with:
+
+- search_alg: grid | random | bayes   (bayes = Gaussian-process expected
+  improvement on the encoded space, scikit-learn — the reference's BayesOpt
+  role, trlx/sweep.py:103-133)
+- scheduler: hyperband                 (ASHA-style successive halving over a
+  budget parameter, default train.total_steps — the reference's HyperBand,
+  trlx/sweep.py:136-158)
+- --parallel N                         (N concurrent trials; single-GPU
+  trials are pinned round-robin to visible devices)
+
+Results land in <output>/results.jsonl + best.json.
 
 Search-space grammar (same keys as the reference's YAML):
   tune_config:
     mode: max | min
     metric: reward/mean
-    search_alg: grid | random
-    num_samples: 16            # random only
+    search_alg: grid | random | bayes
+    num_samples: 16
+    scheduler: hyperband          # optional
+    budget_param: train.total_steps
+    max_budget: 64
+    eta: 3
   method.init_kl_coef:
     strategy: loguniform
     values: [1e-4, 1e-1]
@@ -24,12 +36,14 @@ Search-space grammar (same keys as the reference's YAML):
 import argparse
 import itertools
 import json
+import math
 import os
 import random
 import subprocess
 import sys
 import time
-from typing import Any, Dict, List
+from concurrent.futures import ThreadPoolExecutor
+from typing import Any, Dict, List, Optional
 
 
 def parse_space(config: Dict[str, Any]):
@@ -48,8 +62,6 @@ def sample_param(strategy: str, values: List[Any], rng: random.Random):
     if strategy == "uniform":
         return rng.uniform(values[0], values[1])
     if strategy == "loguniform":
-        import math
-
         return math.exp(rng.uniform(math.log(values[0]), math.log(values[1])))
     if strategy == "quniform":
         lo, hi, q = values
@@ -68,11 +80,78 @@ def generate_trials(tune_config: Dict, dims: Dict, seed: int = 0):
         keys = list(grid_dims)
         for combo in itertools.product(*grid_dims.values()):
             yield dict(zip(keys, combo))
-    else:  # random / bayesopt fallback
+    else:  # random (bayes adds its own proposal loop on top)
         rng = random.Random(seed)
         n = int(tune_config.get("num_samples", 8))
         for _ in range(n):
             yield {p: sample_param(s, v, rng) for p, (s, v) in dims.items()}
+
+
+# ---- encoded space for the GP (bayes) ---------------------------------------
+
+
+def _encode(dims: Dict, hparams: Dict) -> List[float]:
+    """Map hparams onto [0, 1]^d (log scale for loguniform; choice -> index)."""
+    x = []
+    for p, (s, v) in dims.items():
+        val = hparams[p]
+        if s == "choice":
+            x.append(v.index(val) / max(len(v) - 1, 1))
+        elif s == "uniform":
+            x.append((val - v[0]) / (v[1] - v[0]))
+        elif s == "loguniform":
+            x.append((math.log(val) - math.log(v[0])) / (math.log(v[1]) - math.log(v[0])))
+        elif s == "quniform":
+            x.append((val - v[0]) / (v[1] - v[0]))
+    return x
+
+
+def _decode(dims: Dict, x: List[float]) -> Dict:
+    out = {}
+    for (p, (s, v)), xi in zip(dims.items(), x):
+        xi = min(max(xi, 0.0), 1.0)
+        if s == "choice":
+            out[p] = v[round(xi * (len(v) - 1))]
+        elif s == "uniform":
+            out[p] = v[0] + xi * (v[1] - v[0])
+        elif s == "loguniform":
+            out[p] = math.exp(math.log(v[0]) + xi * (math.log(v[1]) - math.log(v[0])))
+        elif s == "quniform":
+            lo, hi, q = v
+            out[p] = round((lo + xi * (hi - lo)) / q) * q
+    return out
+
+
+def propose_bayes(dims: Dict, observed: List, mode: str, rng: random.Random,
+                  n_candidates: int = 512) -> Dict:
+    """GP expected-improvement proposal over the encoded space; pending trials
+    enter with the observed mean ("constant liar") so parallel batches spread
+    out.  Falls back to random until 2 scored observations exist."""
+    scored = [(x, y) for x, y in observed if y is not None]
+    if len(scored) < 2:
+        return {p: sample_param(s, v, rng) for p, (s, v) in dims.items()}
+    liar = sum(y for _, y in scored) / len(scored)
+    X = [x for x, _ in observed]
+    y = [(yv if yv is not None else liar) for _, yv in observed]
+    sign = 1.0 if mode == "max" else -1.0
+    y = [sign * v for v in y]
+
+    import numpy as np
+    from sklearn.gaussian_process import GaussianProcessRegressor
+    from sklearn.gaussian_process.kernels import RBF, ConstantKernel, WhiteKernel
+
+    kernel = ConstantKernel(1.0) * RBF(length_scale=0.25) + WhiteKernel(1e-3)
+    gp = GaussianProcessRegressor(kernel=kernel, normalize_y=True)
+    gp.fit(np.asarray(X), np.asarray(y))
+    best = max(y)
+    cand = np.random.RandomState(rng.randrange(2**31)).rand(n_candidates, len(dims))
+    mu, sigma = gp.predict(cand, return_std=True)
+    sigma = np.maximum(sigma, 1e-9)
+    z = (mu - best) / sigma
+    from scipy.stats import norm
+
+    ei = (mu - best) * norm.cdf(z) + sigma * norm.pdf(z)
+    return _decode(dims, list(cand[int(np.argmax(ei))]))
 
 
 def read_last_metrics(logging_dir: str) -> Dict[str, float]:
@@ -88,14 +167,133 @@ def read_last_metrics(logging_dir: str) -> Dict[str, float]:
     return last
 
 
-def main():
+class Executor:
+    """Runs trials as subprocesses, optionally in parallel with round-robin
+    single-GPU pinning (the reference parallelizes through Ray workers)."""
+
+    def __init__(self, args, metric: str):
+        self.args = args
+        self.metric = metric
+        self.counter = 0
+        try:
+            import torch
+
+            self.n_devices = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        except Exception:
+            self.n_devices = 0
+
+    def run(self, idx: int, hparams: Dict) -> Dict:
+        args = self.args
+        trial_dir = os.path.join(args.output, f"trial_{idx:03d}")
+        hparams = dict(hparams)
+        hparams["train.logging_dir"] = trial_dir
+        hparams["train.tracker"] = "jsonl"
+        hparams["train.checkpoint_dir"] = os.path.join(trial_dir, "ckpts")
+        print(f"[sweep] trial {idx}: {hparams}")
+        env = dict(os.environ)
+        if args.num_gpus > 1:
+            cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+                   f"--nproc-per-node={args.num_gpus}", "--master-addr", "127.0.0.1",
+                   "--master-port", str(29400 + idx % 100),
+                   args.script, json.dumps(hparams)]
+        else:
+            cmd = [sys.executable, args.script, json.dumps(hparams)]
+            if args.parallel > 1 and self.n_devices > 1:
+                env["HIP_VISIBLE_DEVICES"] = str(idx % self.n_devices)
+                env["CUDA_VISIBLE_DEVICES"] = str(idx % self.n_devices)
+        t0 = time.time()
+        proc = subprocess.run(cmd, env=env)
+        metrics = read_last_metrics(trial_dir)
+        result = {
+            "trial": idx,
+            "hparams": hparams,
+            "metrics": metrics,
+            "returncode": proc.returncode,
+            "wallclock_s": round(time.time() - t0, 1),
+        }
+        with open(os.path.join(args.output, "results.jsonl"), "a") as f:
+            f.write(json.dumps(result) + "\n")
+        return result
+
+    def score(self, result: Dict) -> Optional[float]:
+        if result["returncode"] != 0:
+            return None
+        v = result["metrics"].get(self.metric)
+        return float(v) if v is not None else None
+
+
+def run_hyperband(ex: Executor, tune_config: Dict, dims: Dict, seed: int, mode: str,
+                  parallel: int) -> List[Dict]:
+    """ASHA-style successive halving: num_samples random configs start at
+    max_budget / eta^(rungs-1); the top 1/eta of each rung re-run with eta x
+    the budget (reference HyperBand role, trlx/sweep.py:136-158)."""
+    budget_param = tune_config.get("budget_param", "train.total_steps")
+    max_budget = int(tune_config.get("max_budget", 64))
+    eta = int(tune_config.get("eta", 3))
+    n = int(tune_config.get("num_samples", 9))
+    rungs = max(1, int(math.log(n, eta)) + 1)
+    b0 = max(1, max_budget // (eta ** (rungs - 1)))
+
+    rng = random.Random(seed)
+    configs = [{p: sample_param(s, v, rng) for p, (s, v) in dims.items()} for _ in range(n)]
+    results = []
+    idx = 0
+    budget = b0
+    sign = 1.0 if mode == "max" else -1.0
+    for rung in range(rungs):
+        print(f"[sweep] hyperband rung {rung}: {len(configs)} configs at "
+              f"{budget_param}={budget}")
+        batch = []
+        for cfg in configs:
+            h = dict(cfg)
+            h[budget_param] = budget
+            batch.append((idx, h))
+            idx += 1
+        with ThreadPoolExecutor(max_workers=max(parallel, 1)) as pool:
+            rung_results = list(pool.map(lambda t: ex.run(*t), batch))
+        results.extend(rung_results)
+        scored = [(r, ex.score(r)) for r in rung_results]
+        scored = [(r, s) for r, s in scored if s is not None]
+        if not scored or rung == rungs - 1:
+            break
+        scored.sort(key=lambda rs: -sign * rs[1])
+        keep = max(1, len(scored) // eta)
+        configs = [{p: r["hparams"][p] for p in dims} for r, _ in scored[:keep]]
+        budget = min(budget * eta, max_budget)
+    return results
+
+
+def run_bayes(ex: Executor, tune_config: Dict, dims: Dict, seed: int, mode: str,
+              parallel: int) -> List[Dict]:
+    n = int(tune_config.get("num_samples", 16))
+    rng = random.Random(seed)
+    observed = []  # (encoded_x, score_or_None)
+    results = []
+    idx = 0
+    while idx < n:
+        batch = []
+        for _ in range(min(max(parallel, 1), n - idx)):
+            h = propose_bayes(dims, observed, mode, rng)
+            observed.append([_encode(dims, h), None])  # pending (constant liar)
+            batch.append((idx, h, len(observed) - 1))
+            idx += 1
+        with ThreadPoolExecutor(max_workers=max(parallel, 1)) as pool:
+            outs = list(pool.map(lambda t: (t[2], ex.run(t[0], t[1])), batch))
+        for slot, r in outs:
+            observed[slot][1] = ex.score(r)
+            results.append(r)
+    return results
+
+
+def main(argv=None):
     parser = argparse.ArgumentParser(description="trlx_amd hyperparameter sweep")
     parser.add_argument("config", help="sweep YAML (search space + tune_config)")
     parser.add_argument("script", help="training script taking JSON hparams as argv[1]")
     parser.add_argument("--num-gpus", type=int, default=1, help="GPUs per trial")
+    parser.add_argument("--parallel", type=int, default=1, help="concurrent trials")
     parser.add_argument("--output", default="sweep_results", help="results directory")
     parser.add_argument("--seed", type=int, default=0)
-    args = parser.parse_args()
+    args = parser.parse_args(argv)
 
     import yaml
 
@@ -106,37 +304,24 @@ def main():
     mode = tune_config.get("mode", "max")
 
     os.makedirs(args.output, exist_ok=True)
-    results = []
-    for i, hparams in enumerate(generate_trials(tune_config, dims, args.seed)):
-        trial_dir = os.path.join(args.output, f"trial_{i:03d}")
-        hparams = dict(hparams)
-        hparams["train.logging_dir"] = trial_dir
-        hparams["train.tracker"] = "jsonl"
-        hparams["train.checkpoint_dir"] = os.path.join(trial_dir, "ckpts")
-        print(f"[sweep] trial {i}: {hparams}")
-        if args.num_gpus > 1:
-            cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-                   f"--nproc-per-node={args.num_gpus}", "--master-addr", "127.0.0.1",
-                   args.script, json.dumps(hparams)]
-        else:
-            cmd = [sys.executable, args.script, json.dumps(hparams)]
-        t0 = time.time()
-        proc = subprocess.run(cmd)
-        metrics = read_last_metrics(trial_dir)
-        results.append({
-            "trial": i,
-            "hparams": hparams,
-            "metrics": metrics,
-            "returncode": proc.returncode,
-            "wallclock_s": round(time.time() - t0, 1),
-        })
-        with open(os.path.join(args.output, "results.jsonl"), "a") as f:
-            f.write(json.dumps(results[-1]) + "\n")
+    ex = Executor(args, metric)
 
-    scored = [r for r in results if metric in r["metrics"] and r["returncode"] == 0]
+    alg = tune_config.get("search_alg", "grid")
+    scheduler = tune_config.get("scheduler")
+    if scheduler == "hyperband":
+        results = run_hyperband(ex, tune_config, dims, args.seed, mode, args.parallel)
+    elif alg == "bayes":
+        results = run_bayes(ex, tune_config, dims, args.seed, mode, args.parallel)
+    else:
+        trials = list(enumerate(generate_trials(tune_config, dims, args.seed)))
+        with ThreadPoolExecutor(max_workers=max(args.parallel, 1)) as pool:
+            results = list(pool.map(lambda t: ex.run(*t), trials))
+
+    scored = [(r, ex.score(r)) for r in results]
+    scored = [(r, s) for r, s in scored if s is not None]
     if scored:
-        best = (max if mode == "max" else min)(scored, key=lambda r: r["metrics"][metric])
-        print(f"[sweep] best trial {best['trial']}: {metric}={best['metrics'][metric]}")
+        best, best_score = (max if mode == "max" else min)(scored, key=lambda rs: rs[1])
+        print(f"[sweep] best trial {best['trial']}: {metric}={best_score}")
         print(json.dumps(best["hparams"], indent=2))
         with open(os.path.join(args.output, "best.json"), "w") as f:
             json.dump(best, f, indent=2)
