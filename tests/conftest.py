@@ -46,7 +46,11 @@ def _cuda_test_teardown():
     if torch.cuda.is_available():
         import gc
 
-        torch.cuda.synchronize()
+        import trlx_amd
+
+        # destroy captured graphs DETERMINISTICALLY before gc frees their
+        # pools mid-way through the next test's allocations
+        trlx_amd.release_graphs()
         gc.collect()
         torch.cuda.synchronize()
         torch.cuda.empty_cache()

@@ -19,12 +19,39 @@ Left-padded prompts are handled with per-row key-start offsets everywhere.
 """
 
 import os
+import weakref
 from dataclasses import dataclass
 from typing import Callable, Optional
 
 import torch
 
 from ... import ops
+
+# live decode engines (weak): release_graphs() destroys their captured
+# hipGraphs DETERMINISTICALLY instead of leaving destruction to GC, whose
+# graph-pool frees can interleave with a later context's allocations
+# (observed mixed-suite segfault, NOTES_ROUND2 "Known issue")
+_ENGINES = weakref.WeakSet()
+
+
+def release_graphs():
+    """Destroy all captured decode graphs + engine buffers now (ordered,
+    synchronized).  Safe to call any time; engines re-capture on next use."""
+    if not torch.cuda.is_available():
+        return
+    torch.cuda.synchronize()
+    for engine in list(_ENGINES):
+        g = getattr(engine, "graph", None)
+        if g is not None:
+            try:
+                g.reset()
+            except (RuntimeError, AttributeError):
+                pass
+            engine.graph = None
+        model = getattr(engine, "model", None)
+        if model is not None and getattr(model, "_decode_engine", None) is engine:
+            model._decode_engine = None
+    torch.cuda.synchronize()
 
 
 @dataclass
@@ -163,6 +190,7 @@ class DecodeEngine:
         self.pad_id = pad
         self.out_tokens = torch.full((batch, max_new), pad, dtype=torch.long, device=device)
         self.graph = None
+        _ENGINES.add(self)
 
     def matches(self, batch, needed_cache, max_new, gen: GenerateConfig) -> bool:
         return (batch == self.batch and needed_cache <= self.cache_len

@@ -43,6 +43,31 @@ from ..utils.trackers import make_tracker
 
 logger = logging.get_logger(__name__)
 
+# live trainers (weak) for deterministic hipGraph teardown — see
+# models/nn/generation.release_graphs
+import weakref  # noqa: E402
+
+_TRAINERS = weakref.WeakSet()
+
+
+def release_train_graphs():
+    """Destroy all captured train-step graphs now (ordered, synchronized)."""
+    if not torch.cuda.is_available():
+        return
+    torch.cuda.synchronize()
+    for trainer in list(_TRAINERS):
+        graphs = getattr(trainer, "_train_graphs", None)
+        if not graphs:
+            continue
+        for key, entry in list(graphs.items()):
+            if entry and entry is not False:
+                try:
+                    entry[0].reset()
+                except (RuntimeError, AttributeError):
+                    pass
+        trainer._train_graphs = {}
+    torch.cuda.synchronize()
+
 
 @register_trainer
 class NativeRLTrainer(BaseRLTrainer):
@@ -146,6 +171,7 @@ class NativeRLTrainer(BaseRLTrainer):
 
         self.iter_count = 0
         self.nth_evaluation = 0
+        _TRAINERS.add(self)
 
     # --- setup ---------------------------------------------------------------
 
