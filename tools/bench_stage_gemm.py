@@ -4,9 +4,14 @@ hipBLASLt on the GPT-2 decode shapes, plus the per-kernel launch floor.
 Run on the GPU box: python tools/bench_stage_gemm.py
 """
 
+import os
+import sys
+
 import torch
 
-from trlx_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from trlx_amd import ops  # noqa: E402
 
 EXT = ops._load_ext()
 

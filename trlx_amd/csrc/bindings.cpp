@@ -88,6 +88,14 @@ void advance_packed(const at::Tensor& packed, at::Tensor& out_tokens, at::Tensor
                     at::Tensor& finished, at::Tensor& rng_offset, at::Tensor& step_col,
                     at::Tensor& cache_idx, at::Tensor& seq_lens, at::Tensor& pos_ids,
                     const c10::optional<at::Tensor>& key_starts, long eos, long pad);
+void stage_gemm_v2(const at::Tensor& a, const at::Tensor& w, const c10::optional<at::Tensor>& bias,
+                   at::Tensor& c, bool norm, const c10::optional<at::Tensor>& nw,
+                   const c10::optional<at::Tensor>& nb, bool norm_rms, double eps, long act,
+                   const c10::optional<at::Tensor>& resid);
+void lm_sample_v2(const at::Tensor& x, const at::Tensor& wlm,
+                  const c10::optional<at::Tensor>& blm, const at::Tensor& nw,
+                  const c10::optional<at::Tensor>& nb, at::Tensor& packed, bool norm_rms,
+                  double eps, double temperature, long seed, const at::Tensor& rng_offset);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("logprobs_fwd", &logprobs_fwd, "fused logsumexp + label gather fwd");
@@ -120,4 +128,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embed_stats", &embed_stats);
   mod.def("lm_sample", &lm_sample);
   mod.def("advance_packed", &advance_packed);
+  mod.def("stage_gemm_v2", &stage_gemm_v2);
+  mod.def("lm_sample_v2", &lm_sample_v2);
 }

@@ -145,6 +145,8 @@ class FusedStageRefs:
                 fc_w=layer.mlp.fc_in.weight, fc_b=layer.mlp.fc_in.bias,
                 down_w=layer.mlp.down_proj.weight, down_b=layer.mlp.down_proj.bias,
             ))
+        # v2 staged kernels need full 16-row tiles and 128-aligned K
+        self.use_v2 = batch >= 16 and H % 128 == 0 and I % 128 == 0
         bf = torch.bfloat16
         self.x = torch.empty(batch, H, dtype=bf, device=device)
         self.x1 = torch.empty(batch, H, dtype=bf, device=device)
@@ -247,6 +249,30 @@ class DecodeEngine:
         cur = self.cur_tok.view(-1)
         pos = self.pos_ids.view(-1)
         ext.embed_stats(f.wte, f.wpe, cur, pos, f.pos_offset, f.x, f.stats, f.packed)
+        temperature = self.gen.temperature if self.gen.do_sample else 0.0
+        if f.use_v2:
+            # LDS-staged v2 kernels: consumers compute their own norm stats
+            for li, lay in enumerate(f.layers):
+                ext.stage_gemm_v2(f.x, lay["qkv_w"], lay["qkv_b"], f.qkv, True,
+                                  lay["ln1_w"], lay["ln1_b"], f.rms, f.eps, 0, None)
+                attn = ext.fused_decode_attention(
+                    f.qkv, self.kv.k[li], self.kv.v[li], self.seq_lens, self.key_starts,
+                    f.rcos, f.rsin, self.cache_idx, f.rot, f.cfg.rope_interleaved, f.scale,
+                ).view(self.batch, -1)
+                ext.stage_gemm_v2(attn, lay["o_w"], lay["o_b"], f.x1, False,
+                                  None, None, False, 0.0, 0, f.x)
+                ext.stage_gemm_v2(f.x1, lay["fc_w"], lay["fc_b"], f.act, True,
+                                  lay["ln2_w"], lay["ln2_b"], f.rms, f.eps, f.act_code, None)
+                ext.stage_gemm_v2(f.act, lay["down_w"], lay["down_b"], f.x, False,
+                                  None, None, False, 0.0, 0, f.x1)
+            ext.lm_sample_v2(f.x, f.lm_w, f.lm_b, f.lnf_w, f.lnf_b, f.packed, f.rms,
+                             f.eps, temperature, self.seed, self.rng_offset)
+            ext.advance_packed(f.packed, self.out_tokens, cur, self.finished,
+                               self.rng_offset, self.step_col, self.cache_idx,
+                               self.seq_lens, pos, self.key_starts,
+                               -1 if self.gen.eos_token_id is None else self.gen.eos_token_id,
+                               self.pad_id)
+            return
         for li, lay in enumerate(f.layers):
             ext.stage_gemm(f.x, lay["qkv_w"], lay["qkv_b"], f.qkv, f.stats[2 * li],
                            lay["ln1_w"], lay["ln1_b"], f.rms, f.eps, 0, None, None)
@@ -260,7 +286,6 @@ class DecodeEngine:
                            lay["ln2_w"], lay["ln2_b"], f.rms, f.eps, f.act_code, None, None)
             ext.stage_gemm(f.act, lay["down_w"], lay["down_b"], f.x, None, None, None,
                            False, 0.0, 0, f.x1, f.stats[2 * li + 2])
-        temperature = self.gen.temperature if self.gen.do_sample else 0.0
         ext.lm_sample(f.x, f.lm_w, f.lm_b, f.stats[2 * len(f.layers)], f.lnf_w, f.lnf_b,
                       f.packed, f.rms, f.eps, temperature, self.seed, self.rng_offset)
         ext.advance_packed(f.packed, self.out_tokens, cur, self.finished, self.rng_offset,
