@@ -212,3 +212,75 @@ def test_reward_server_with_trained_rm(tmp_path, monkeypatch):
     finally:
         server._rm["model"] = None
         server._rm["tok"] = None
+
+
+@pytest.mark.parametrize("which", ["architext", "simulacra", "llama", "peft"])
+def test_examples_tail_run(which, tmp_path, monkeypatch):
+    """Smoke-run the round-2 example tail (architext, simulacra, llama/peft
+    sentiment variants) at tiny settings on CPU."""
+    monkeypatch.syspath_prepend("examples")
+    import importlib
+
+    mod = importlib.import_module({
+        "architext": "architext", "simulacra": "simulacra",
+        "llama": "ppo_sentiments_llama", "peft": "ppo_sentiments_peft",
+    }[which])
+    overrides = {
+        "train.total_steps": 2,
+        "train.epochs": 1,
+        "train.batch_size": 4,
+        "train.eval_interval": 2,
+        "train.checkpoint_interval": 100,
+        "train.checkpoint_dir": str(tmp_path / "ckpt"),
+        "train.tracker": None,
+        "train.save_best": False,
+        "train.seq_length": 32,
+    }
+    if which in ("architext", "peft"):
+        overrides["model.model_extra_configs"] = {
+            "config": __import__("conftest").tiny_config().to_dict()}
+    if which == "llama":
+        overrides["model.model_extra_configs"] = {
+            "config": __import__("conftest").tiny_config(
+                arch_name="llama", norm="rmsnorm", position_encoding="rope",
+                activation="silu", swiglu=True, attn_bias=False, mlp_bias=False,
+                intermediate_size=128, tie_word_embeddings=False).to_dict()}
+    if which in ("architext", "llama", "peft"):
+        overrides.update({"method.num_rollouts": 4, "method.chunk_size": 4,
+                          "method.ppo_epochs": 1,
+                          "method.gen_kwargs": dict(max_new_tokens=4, top_k=0, top_p=1.0,
+                                                    do_sample=True)})
+    else:  # simulacra (ILQL)
+        overrides.update({"method.gen_kwargs": dict(max_new_tokens=4, top_k=5, beta=1,
+                                                    temperature=1.0)})
+    mod.main(overrides)
+
+
+def test_grounded_program_synthesis_run(tmp_path, monkeypatch):
+    monkeypatch.syspath_prepend("examples/grounded_program_synthesis")
+    import importlib
+
+    lang = importlib.import_module("lang")
+    # grounded grading: gold programs score +1, junk scores -1
+    data = lang.make_dataset(8, seed=3)
+    assert lang.reward_fn([f"{p} {g}" for p, g in data]) == [1.0] * 8
+    assert lang.reward_fn(["nonsense"]) == [-1.0]
+
+    mod = importlib.import_module("train_trlx")
+    overrides = {
+        "train.total_steps": 2,
+        "train.epochs": 1,
+        "train.batch_size": 4,
+        "train.eval_interval": 2,
+        "train.checkpoint_interval": 100,
+        "train.checkpoint_dir": str(tmp_path / "ckpt"),
+        "train.tracker": None,
+        "train.save_best": False,
+        "train.seq_length": 48,
+        "model.model_extra_configs": {"config": __import__("conftest").tiny_config().to_dict()},
+        "method.num_rollouts": 4,
+        "method.chunk_size": 4,
+        "method.ppo_epochs": 1,
+        "method.gen_kwargs": dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True),
+    }
+    mod.main(overrides)
