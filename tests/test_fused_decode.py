@@ -179,11 +179,11 @@ def test_stage_gemm_v2_numerics(norm, act):
     c = torch.empty(B, N, device=dev, dtype=torch.bfloat16)
 
     if norm is None:
-        EXT.stage_gemm_v2(a, w, bias, c, False, None, None, False, eps, act, resid)
+        EXT.stage_gemm_v2(a, w, bias, c, False, None, None, False, eps, act, resid, None)
         a_ref = a.float()
     else:
         rms = norm == "rmsnorm"
-        EXT.stage_gemm_v2(a, w, bias, c, True, nw, nb if not rms else None, rms, eps, act, resid)
+        EXT.stage_gemm_v2(a, w, bias, c, True, nw, nb if not rms else None, rms, eps, act, resid, None)
         af = a.float()
         if rms:
             a_ref = af * torch.rsqrt((af * af).mean(-1, keepdim=True) + eps) * nw.float()
@@ -207,7 +207,7 @@ def test_stage_gemm_v2_long_k():
     a = (torch.randn(B, K, device=dev) * 0.05).bfloat16()
     w = (torch.randn(N, K, device=dev) * 0.05).bfloat16()
     c = torch.empty(B, N, device=dev, dtype=torch.bfloat16)
-    EXT.stage_gemm_v2(a, w, None, c, False, None, None, False, 1e-5, 0, None)
+    EXT.stage_gemm_v2(a, w, None, c, False, None, None, False, 1e-5, 0, None, None)
     ref = a.float() @ w.float().t()
     torch.testing.assert_close(c.float(), ref, atol=5e-2, rtol=5e-2)
 
@@ -235,8 +235,8 @@ def test_lm_sample_v2_greedy_matches_torch():
 
 
 def test_fused_engine_v2_matches_module_path():
-    """B >= 16 engages the v2 staged kernels; greedy decode must match the
-    module path."""
+    """A 16-multiple batch engages the v3/v2 staged kernels; greedy decode
+    must match the module path."""
     from trlx_amd.models.nn.config import TransformerConfig
     from trlx_amd.models.nn.generation import generate
     from trlx_amd.models.nn.transformer import CausalTransformer
@@ -246,13 +246,13 @@ def test_fused_engine_v2_matches_module_path():
                             max_position_embeddings=128, arch_name="gpt2")
     model = CausalTransformer(cfg).cuda().bfloat16()
     model.eval()
-    ids = torch.randint(3, 600, (20, 9), device="cuda")
+    ids = torch.randint(3, 600, (32, 9), device="cuda")
     mask = torch.ones_like(ids)
     mask[0, :3] = 0
 
     out_fused = generate(model, ids, mask, max_new_tokens=8, do_sample=False)
     eng = model._decode_engine
-    assert eng.fused is not None and eng.fused.use_v2, "v2 staged path did not engage"
+    assert eng.fused is not None and eng.fused.use_v3, "v3 staged path did not engage"
     del model._decode_engine
     os.environ["TRLX_AMD_NO_FUSED_DECODE"] = "1"
     try:
@@ -261,3 +261,61 @@ def test_fused_engine_v2_matches_module_path():
         del os.environ["TRLX_AMD_NO_FUSED_DECODE"]
     agree = (out_fused == out_mod).float().mean().item()
     assert agree >= 0.97, (agree, out_fused, out_mod)
+
+
+def test_stage_gemm_v3_numerics():
+    """v3 W-stationary GEMM with slab-reduced norm stats vs torch fp32."""
+    torch.manual_seed(11)
+    M, K, N = 32, 128, 96
+    dev = "cuda"
+    a = torch.randn(M, K, device=dev).bfloat16()
+    w = (torch.randn(N, K, device=dev) * 0.1).bfloat16()
+    bias = (torch.randn(N, device=dev) * 0.1).bfloat16()
+    nw = (1 + 0.1 * torch.randn(K, device=dev)).bfloat16()
+    nb = (0.1 * torch.randn(K, device=dev)).bfloat16()
+    eps = 1e-5
+    c = torch.empty(M, N, device=dev, dtype=torch.bfloat16)
+    pstats_out = torch.zeros(N // 16, M, 2, device=dev)
+
+    # split stats into 8 fake 16-col partials to exercise the slab reduce
+    af = a.float()
+    parts = torch.stack([torch.stack([af[:, i*16:(i+1)*16].sum(-1),
+                                      (af[:, i*16:(i+1)*16] ** 2).sum(-1)], -1)
+                         for i in range(K // 16)]).contiguous()
+    EXT.stage_gemm_v3(a, w, bias, c, parts, K // 16, nw, nb, False, eps, 2, pstats_out)
+
+    mu = af.mean(-1, keepdim=True)
+    var = af.var(-1, keepdim=True, unbiased=False)
+    an = ((af - mu) * torch.rsqrt(var + eps) * nw.float() + nb.float()).bfloat16().float()
+    ref = an @ w.float().t() + bias.float()
+    ref = 0.5 * ref * (1 + torch.tanh(0.7978845608 * (ref + 0.044715 * ref ** 3)))
+    torch.testing.assert_close(c.float(), ref, atol=5e-2, rtol=5e-2)
+    # producer slab: per-16-col partials of the stored output
+    cf = c.float()
+    want = torch.stack([torch.stack([cf[:, i*16:(i+1)*16].sum(-1),
+                                     (cf[:, i*16:(i+1)*16] ** 2).sum(-1)], -1)
+                        for i in range(N // 16)])
+    torch.testing.assert_close(pstats_out, want, atol=1e-2, rtol=1e-3)
+
+
+def test_lm_sample_v3_greedy_matches_torch():
+    torch.manual_seed(12)
+    B, K, V = 32, 128, 1000
+    dev = "cuda"
+    x = torch.randn(B, K, device=dev).bfloat16()
+    w = (torch.randn(V, K, device=dev) * 0.2).bfloat16()
+    nw = (1 + 0.1 * torch.randn(K, device=dev)).bfloat16()
+    nb = (0.1 * torch.randn(K, device=dev)).bfloat16()
+    packed = torch.zeros(B, device=dev, dtype=torch.long)
+    off = torch.zeros(1, device=dev, dtype=torch.long)
+    xf = x.float()
+    full = torch.stack([xf.sum(-1), (xf * xf).sum(-1)], -1)[None].contiguous()
+    EXT.lm_sample_v3(x, w, None, full, 1, nw, nb, packed, False, 1e-5, 0.0, 123, off)
+    tok = (~(packed & 0xFFFFFFFF).to(torch.int64)) & 0xFFFFFFFF
+    mu = xf.mean(-1, keepdim=True)
+    xn = ((xf - mu) * torch.rsqrt(xf.var(-1, keepdim=True, unbiased=False) + 1e-5)
+          * nw.float() + nb.float()).bfloat16().float()
+    logits = (xn @ w.float().t()).bfloat16().float()
+    want = logits.argmax(-1)
+    agree = (tok == want).float().mean().item()
+    assert agree >= 0.95, (agree, tok[:8], want[:8])

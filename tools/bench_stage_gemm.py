@@ -66,12 +66,26 @@ def main():
         t_raw = time_fn(lambda: EXT.stage_gemm(a, w, bias, c, None, None, None, False, 1e-5, 0, None, None))
         t_norm = time_fn(lambda: EXT.stage_gemm(a, w, bias, c, nstats, nw, nb, False, 1e-5, 0, None, None))
         t_full = time_fn(lambda: EXT.stage_gemm(a, w, bias, c, None, None, None, False, 1e-5, 0, resid, ostats))
+        t_v2 = time_fn(lambda: EXT.stage_gemm_v2(a, w, bias, c, False, None, None, False, 1e-5, 0, None, None))
+        t_v2n = time_fn(lambda: EXT.stage_gemm_v2(a, w, bias, c, True, nw, nb, False, 1e-5, 0, None, None))
+        t_v2r = time_fn(lambda: EXT.stage_gemm_v2(a, w, bias, c, False, None, None, False, 1e-5, 0, resid, None))
+        ps = torch.zeros(max(K // 16, 1), M, 2, device=dev)
+        af2 = a.float()
+        ps[0, :, 0] = af2.sum(-1)
+        ps[0, :, 1] = (af2 * af2).sum(-1)
+        pso = torch.zeros(N // 16 if N % 16 == 0 else 1, M, 2, device=dev)
+        t_v3 = time_fn(lambda: EXT.stage_gemm_v3(a, w, bias, c, None, 0, None, None, False, 1e-5, 0, None))
+        t_v3n = time_fn(lambda: EXT.stage_gemm_v3(a, w, bias, c, ps, 1, nw, nb, False, 1e-5, 0, None))
+        t_v3p = time_fn(lambda: EXT.stage_gemm_v3(a, w, bias, c, None, 0, None, None, False, 1e-5, 0, pso if N % 16 == 0 else None))
         t_sk = time_fn(lambda: EXT.skinny_gemm(a, w, bias, 0))
         t_bl = time_fn(lambda: torch.nn.functional.linear(a, w, bias))
         gbs = (M * K + N * K + M * N) * 2 / 1e9
-        print(f"  {label:5s} M{M} K{K} N{N}: raw {t_raw:7.2f}  norm {t_norm:7.2f}  "
-              f"resid+stats {t_full:7.2f}  skinny {t_sk:7.2f}  blaslt {t_bl:7.2f} us "
-              f"(raw {gbs / (t_raw * 1e-6):5.2f} TB/s)")
+        wbytes = N * K * 2 / 1e12
+        print(f"  {label:5s} M{M} K{K} N{N}: raw {t_raw:7.2f} norm {t_norm:7.2f} "
+              f"rs {t_full:7.2f} | v2 {t_v2:7.2f} v2n {t_v2n:7.2f} v2r {t_v2r:7.2f} | "
+              f"v3 {t_v3:7.2f} v3n {t_v3n:7.2f} v3p {t_v3p:7.2f} | "
+              f"skinny {t_sk:7.2f} blaslt {t_bl:7.2f} us "
+              f"(v2 {wbytes / (t_v2 * 1e-6):5.2f} TB/s W-stream)")
 
     # lm_sample
     M, K, N = 128, 768, 50257
@@ -95,6 +109,33 @@ def main():
         EXT.lm_sample(x, w, None, nstats, nw, nb, packed, False, 1e-5, 0.0, 7, off)
 
     print(f"  lm_sample (greedy):   {time_fn(lmg, reps=50):8.2f} us")
+
+    def lm2():
+        packed.zero_()
+        EXT.lm_sample_v2(x, w, None, nw, nb, packed, False, 1e-5, 1.0, 7, off)
+
+    print(f"  lm_sample_v2 (sampling): {time_fn(lm2, reps=50):8.2f} us")
+
+    def lm2g():
+        packed.zero_()
+        EXT.lm_sample_v2(x, w, None, nw, nb, packed, False, 1e-5, 0.0, 7, off)
+
+    print(f"  lm_sample_v2 (greedy):   {time_fn(lm2g, reps=50):8.2f} us")
+
+    xf3 = x.float()
+    full = torch.stack([xf3.sum(-1), (xf3 * xf3).sum(-1)], -1)[None].contiguous()
+
+    def lm3():
+        packed.zero_()
+        EXT.lm_sample_v3(x, w, None, full, 1, nw, nb, packed, False, 1e-5, 1.0, 7, off)
+
+    print(f"  lm_sample_v3 (sampling): {time_fn(lm3, reps=50):8.2f} us")
+
+    def lm3g():
+        packed.zero_()
+        EXT.lm_sample_v3(x, w, None, full, 1, nw, nb, packed, False, 1e-5, 0.0, 7, off)
+
+    print(f"  lm_sample_v3 (greedy):   {time_fn(lm3g, reps=50):8.2f} us")
 
 
 if __name__ == "__main__":

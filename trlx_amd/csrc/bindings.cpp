@@ -91,7 +91,18 @@ void advance_packed(const at::Tensor& packed, at::Tensor& out_tokens, at::Tensor
 void stage_gemm_v2(const at::Tensor& a, const at::Tensor& w, const c10::optional<at::Tensor>& bias,
                    at::Tensor& c, bool norm, const c10::optional<at::Tensor>& nw,
                    const c10::optional<at::Tensor>& nb, bool norm_rms, double eps, long act,
-                   const c10::optional<at::Tensor>& resid);
+                   const c10::optional<at::Tensor>& resid,
+                   const c10::optional<at::Tensor>& pstats_out);
+void stage_gemm_v3(const at::Tensor& a, const at::Tensor& w, const c10::optional<at::Tensor>& bias,
+                   at::Tensor& c, const c10::optional<at::Tensor>& pstats, long nparts,
+                   const c10::optional<at::Tensor>& nw, const c10::optional<at::Tensor>& nb,
+                   bool norm_rms, double eps, long act,
+                   const c10::optional<at::Tensor>& pstats_out);
+void lm_sample_v3(const at::Tensor& x, const at::Tensor& wlm,
+                  const c10::optional<at::Tensor>& blm, const at::Tensor& pstats, long nparts,
+                  const at::Tensor& nw, const c10::optional<at::Tensor>& nb, at::Tensor& packed,
+                  bool norm_rms, double eps, double temperature, long seed,
+                  const at::Tensor& rng_offset);
 void lm_sample_v2(const at::Tensor& x, const at::Tensor& wlm,
                   const c10::optional<at::Tensor>& blm, const at::Tensor& nw,
                   const c10::optional<at::Tensor>& nb, at::Tensor& packed, bool norm_rms,
@@ -130,4 +141,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("advance_packed", &advance_packed);
   mod.def("stage_gemm_v2", &stage_gemm_v2);
   mod.def("lm_sample_v2", &lm_sample_v2);
+  mod.def("stage_gemm_v3", &stage_gemm_v3);
+  mod.def("lm_sample_v3", &lm_sample_v3);
 }

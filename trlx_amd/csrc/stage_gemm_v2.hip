@@ -111,7 +111,8 @@ template <bool NORM, bool RMS, bool RESID>
 __global__ __launch_bounds__(V2BLOCK, 2) void stage_gemm_v2_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ W, const bf16_t* __restrict__ bias,
     bf16_t* __restrict__ C, int M, int N, int K, const bf16_t* __restrict__ nw,
-    const bf16_t* __restrict__ nb, float eps, int act, const bf16_t* __restrict__ resid) {
+    const bf16_t* __restrict__ nb, float eps, int act, const bf16_t* __restrict__ resid,
+    float* __restrict__ pstats_out) {
   const int lane = threadIdx.x % WAVE;
   const int wid = threadIdx.x / WAVE;
   const int nM = (M + 15) >> 4;
@@ -205,7 +206,23 @@ __global__ __launch_bounds__(V2BLOCK, 2) void stage_gemm_v2_kernel(
         if (bias) v += bf2f(bias[ccol].u);
         v = v2_act(v, act);
         if (RESID) v += bf2f(resid[(size_t)crow * N + ccol].u);
-        C[(size_t)crow * N + ccol].u = f2bf(v);
+        const unsigned short vb = f2bf(v);
+        C[(size_t)crow * N + ccol].u = vb;
+        if (pstats_out) {
+          // per-row 16-col partial (sum, sumsq) of the STORED value — plain
+          // slab store, reduced by the consumer (stage_gemm_v3 pre-phase)
+          const float vr = bf2f(vb);
+          float s = vr, s2 = vr * vr;
+#pragma unroll
+          for (int off = 8; off > 0; off >>= 1) {
+            s += __shfl_xor(s, off);
+            s2 += __shfl_xor(s2, off);
+          }
+          if ((lane & 15) == 0) {
+            pstats_out[((size_t)nt * M + crow) * 2] = s;
+            pstats_out[((size_t)nt * M + crow) * 2 + 1] = s2;
+          }
+        }
       }
     }
     __syncthreads();
@@ -365,7 +382,8 @@ __global__ __launch_bounds__(V2BLOCK, 2) void lm_sample_v2_kernel(
 void stage_gemm_v2(const at::Tensor& a, const at::Tensor& w, const c10::optional<at::Tensor>& bias,
                    at::Tensor& c, bool norm, const c10::optional<at::Tensor>& nw,
                    const c10::optional<at::Tensor>& nb, bool norm_rms, double eps, long act,
-                   const c10::optional<at::Tensor>& resid) {
+                   const c10::optional<at::Tensor>& resid,
+                   const c10::optional<at::Tensor>& pstats_out) {
   TORCH_CHECK(a.is_cuda() && a.dtype() == at::kBFloat16 && a.dim() == 2 && a.is_contiguous());
   TORCH_CHECK(w.dtype() == at::kBFloat16 && w.is_contiguous());
   const int M = a.size(0), K = a.size(1), N = w.size(0);
@@ -373,6 +391,9 @@ void stage_gemm_v2(const at::Tensor& a, const at::Tensor& w, const c10::optional
   TORCH_CHECK(M >= 16 && N >= 16, "stage_gemm_v2: M and N must be >= 16");
   TORCH_CHECK(c.size(0) == M && c.size(1) == N && c.is_contiguous());
   const bool has_res = resid.has_value();
+  TORCH_CHECK(!pstats_out.has_value() || (M % 16 == 0 && N % 16 == 0),
+              "stage_gemm_v2: pstats_out needs M,N multiples of 16");
+  auto pso = pstats_out.has_value() ? pstats_out->data_ptr<float>() : (float*)nullptr;
   auto bp = bias.has_value() ? reinterpret_cast<const bf16_t*>(bias->data_ptr()) : nullptr;
   auto nwp = nw.has_value() ? reinterpret_cast<const bf16_t*>(nw->data_ptr()) : nullptr;
   auto nbp = nb.has_value() ? reinterpret_cast<const bf16_t*>(nb->data_ptr()) : nullptr;
@@ -385,7 +406,7 @@ void stage_gemm_v2(const at::Tensor& a, const at::Tensor& w, const c10::optional
   stage_gemm_v2_kernel<NORMV, RMSV, RESV><<<grid, V2BLOCK, 0, stream>>>(                  \
       reinterpret_cast<const bf16_t*>(a.data_ptr()),                                      \
       reinterpret_cast<const bf16_t*>(w.data_ptr()), bp,                                  \
-      reinterpret_cast<bf16_t*>(c.data_ptr()), M, N, K, nwp, nbp, (float)eps, (int)act, rp)
+      reinterpret_cast<bf16_t*>(c.data_ptr()), M, N, K, nwp, nbp, (float)eps, (int)act, rp, pso)
   if (norm) {
     if (norm_rms) {
       if (has_res) LAUNCH_V2(true, true, true);

@@ -145,14 +145,22 @@ class FusedStageRefs:
                 fc_w=layer.mlp.fc_in.weight, fc_b=layer.mlp.fc_in.bias,
                 down_w=layer.mlp.down_proj.weight, down_b=layer.mlp.down_proj.bias,
             ))
-        # v2 staged kernels need full 16-row tiles and 128-aligned K
-        self.use_v2 = batch >= 16 and H % 128 == 0 and I % 128 == 0
+        # v3 (W-stationary, partial-stat slabs) needs 16-multiple batch and
+        # 16-multiple widths; everything the flagship runs qualifies
+        self.use_v3 = (batch in (16, 32, 64, 128, 256) and H % 128 == 0 and I % 128 == 0
+                       and cfg.qkv_out % 16 == 0)
         bf = torch.bfloat16
         self.x = torch.empty(batch, H, dtype=bf, device=device)
         self.x1 = torch.empty(batch, H, dtype=bf, device=device)
         self.qkv = torch.empty(batch, cfg.qkv_out, dtype=bf, device=device)
         self.act = torch.empty(batch, I, dtype=bf, device=device)
-        self.stats = torch.zeros(2 * L + 1, batch, 2, dtype=torch.float32, device=device)
+        # partial-stat slabs [N/16 panels, B, 2]; embed fills slot 0 of ps_x
+        # (nparts=1 for the first layer's qkv).  Sized to also cover the v1
+        # fallback's 2L+1 per-norm slots.
+        nslots = max(H // 16, 2 * L + 1, 1)
+        self.ps_x = torch.zeros(nslots, batch, 2, dtype=torch.float32, device=device)
+        self.ps_x1 = torch.zeros(nslots, batch, 2, dtype=torch.float32, device=device)
+        self.stats = self.ps_x  # embed_stats writes slot 0 + zeroes the rest
         self.packed = torch.zeros(batch, dtype=torch.long, device=device)
 
 
@@ -250,23 +258,29 @@ class DecodeEngine:
         pos = self.pos_ids.view(-1)
         ext.embed_stats(f.wte, f.wpe, cur, pos, f.pos_offset, f.x, f.stats, f.packed)
         temperature = self.gen.temperature if self.gen.do_sample else 0.0
-        if f.use_v2:
-            # LDS-staged v2 kernels: consumers compute their own norm stats
+        if f.use_v3:
+            # v3 W-stationary consumers (qkv/fc/lm read each weight byte
+            # ONCE) + v2 producers emitting partial-stat slabs for the next
+            # norm.  nparts: the embed kernel wrote one full-row slot; each
+            # down/o stage writes H/16 16-col partials.
+            H16 = f.cfg.hidden_size // 16
+            nparts_x = 1  # embed slot 0
             for li, lay in enumerate(f.layers):
-                ext.stage_gemm_v2(f.x, lay["qkv_w"], lay["qkv_b"], f.qkv, True,
+                ext.stage_gemm_v3(f.x, lay["qkv_w"], lay["qkv_b"], f.qkv, f.ps_x, nparts_x,
                                   lay["ln1_w"], lay["ln1_b"], f.rms, f.eps, 0, None)
                 attn = ext.fused_decode_attention(
                     f.qkv, self.kv.k[li], self.kv.v[li], self.seq_lens, self.key_starts,
                     f.rcos, f.rsin, self.cache_idx, f.rot, f.cfg.rope_interleaved, f.scale,
                 ).view(self.batch, -1)
                 ext.stage_gemm_v2(attn, lay["o_w"], lay["o_b"], f.x1, False,
-                                  None, None, False, 0.0, 0, f.x)
-                ext.stage_gemm_v2(f.x1, lay["fc_w"], lay["fc_b"], f.act, True,
+                                  None, None, False, 0.0, 0, f.x, f.ps_x1)
+                ext.stage_gemm_v3(f.x1, lay["fc_w"], lay["fc_b"], f.act, f.ps_x1, H16,
                                   lay["ln2_w"], lay["ln2_b"], f.rms, f.eps, f.act_code, None)
                 ext.stage_gemm_v2(f.act, lay["down_w"], lay["down_b"], f.x, False,
-                                  None, None, False, 0.0, 0, f.x1)
-            ext.lm_sample_v2(f.x, f.lm_w, f.lm_b, f.lnf_w, f.lnf_b, f.packed, f.rms,
-                             f.eps, temperature, self.seed, self.rng_offset)
+                                  None, None, False, 0.0, 0, f.x1, f.ps_x)
+                nparts_x = H16
+            ext.lm_sample_v3(f.x, f.lm_w, f.lm_b, f.ps_x, nparts_x, f.lnf_w, f.lnf_b,
+                             f.packed, f.rms, f.eps, temperature, self.seed, self.rng_offset)
             ext.advance_packed(f.packed, self.out_tokens, cur, self.finished,
                                self.rng_offset, self.step_col, self.cache_idx,
                                self.seq_lens, pos, self.key_starts,
