@@ -9,6 +9,8 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
+os.environ["TRLX_AMD_FUSED_DECODE"] = "1"  # the engine path under test is opt-in
+
 if torch.cuda.is_available():
     from trlx_amd import ops
 
@@ -132,12 +134,12 @@ def test_fused_engine_matches_module_path():
     assert hasattr(model, "_decode_engine") and model._decode_engine.fused is not None, \
         "fused stage path did not engage"
     del model._decode_engine
-    os.environ["TRLX_AMD_NO_FUSED_DECODE"] = "1"
+    os.environ["TRLX_AMD_FUSED_DECODE"] = "0"
     try:
         out_mod = generate(model, ids, mask, max_new_tokens=8, do_sample=False)
         assert model._decode_engine.fused is None
     finally:
-        del os.environ["TRLX_AMD_NO_FUSED_DECODE"]
+        os.environ["TRLX_AMD_FUSED_DECODE"] = "1"
     # greedy tokens may differ only at near-exact logit ties (different GEMM
     # accumulation order); require (near-)full agreement
     agree = (out_fused == out_mod).float().mean().item()
@@ -254,11 +256,11 @@ def test_fused_engine_v2_matches_module_path():
     eng = model._decode_engine
     assert eng.fused is not None and eng.fused.use_v3, "v3 staged path did not engage"
     del model._decode_engine
-    os.environ["TRLX_AMD_NO_FUSED_DECODE"] = "1"
+    os.environ["TRLX_AMD_FUSED_DECODE"] = "0"
     try:
         out_mod = generate(model, ids, mask, max_new_tokens=8, do_sample=False)
     finally:
-        del os.environ["TRLX_AMD_NO_FUSED_DECODE"]
+        os.environ["TRLX_AMD_FUSED_DECODE"] = "1"
     agree = (out_fused == out_mod).float().mean().item()
     assert agree >= 0.97, (agree, out_fused, out_mod)
 

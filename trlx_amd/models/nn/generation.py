@@ -179,9 +179,13 @@ class DecodeEngine:
         self.seed = gen.seed if gen.seed is not None else int(torch.randint(0, 2**31 - 1, (1,)).item())
         self.kv = model.new_kv_cache(batch, cache_len, device=device)
         # fused stage-kernel step (5 kernels/layer): eligible archs only, and
-        # only for the plain sampler the lm_sample kernel implements
+        # only for the plain sampler the lm_sample kernel implements.
+        # OPT-IN (TRLX_AMD_FUSED_DECODE=1): the same-box A/B currently favors
+        # the module path (its in-graph hipBLASLt GEMMs run 2x faster than the
+        # stage kernels — see profiles/ and tools/bench_stage_gemm.py); the
+        # fused path stays in for iteration until it wins e2e.
         self.fused = None
-        if (shaping_fn is None and os.environ.get("TRLX_AMD_NO_FUSED_DECODE") != "1"
+        if (shaping_fn is None and os.environ.get("TRLX_AMD_FUSED_DECODE") == "1"
                 and gen.top_k in (0, None) and gen.top_p in (1.0, None)
                 and gen.min_new_tokens == 0):
             try:
