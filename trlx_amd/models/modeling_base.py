@@ -129,11 +129,11 @@ class PreTrainedModelWrapper(nn.Module):
                 logger.warning(f"missing keys loading {pretrained_model_name_or_path}: {missing}")
             adapter_path = os.path.join(pretrained_model_name_or_path, "adapter_model.pt")
             if os.path.exists(adapter_path) and wrapped_kwargs.get("peft_config") is not None:
-                from .lora import apply_lora, load_lora_state_dict
+                from .lora import apply_peft, load_adapter_state_dict
 
-                apply_lora(base, wrapped_kwargs["peft_config"])
-                load_lora_state_dict(base, torch.load(adapter_path, map_location="cpu",
-                                                      weights_only=True))
+                apply_peft(base, wrapped_kwargs["peft_config"])
+                load_adapter_state_dict(base, torch.load(adapter_path, map_location="cpu",
+                                                         weights_only=True))
                 wrapped_kwargs = dict(wrapped_kwargs)
                 wrapped_kwargs["_lora_applied"] = True
         elif pretrained_model_name_or_path in PRESETS:
@@ -150,9 +150,9 @@ class PreTrainedModelWrapper(nn.Module):
         if wrapped_kwargs.pop("_lora_applied", False):
             pass
         elif wrapped_kwargs.get("peft_config") is not None:
-            from .lora import apply_lora
+            from .lora import apply_peft
 
-            apply_lora(base, wrapped_kwargs["peft_config"])
+            apply_peft(base, wrapped_kwargs["peft_config"])
         model = cls(base, **wrapped_kwargs)
         if heads_sd:
             model.post_init(heads_sd)
@@ -185,7 +185,7 @@ class PreTrainedModelWrapper(nn.Module):
         base weights exclude the adapter.  Under tensor parallelism each TP
         rank writes its shard to mp_rank_XX/ (the NeMo format,
         modeling_nemo_ppo.py:445-467); from_pretrained reassembles."""
-        from .lora import has_lora, lora_state_dict
+        from .lora import adapter_state_dict, has_adapter
         from ..parallel import topo
 
         if topo.tp_size() > 1:
@@ -211,8 +211,8 @@ class PreTrainedModelWrapper(nn.Module):
         base_sd = {k.replace(".base.weight", ".weight").replace(".base.bias", ".bias"): v
                    for k, v in base_sd.items()}
         save_hf_dir(save_directory, self.config, base_sd)
-        if has_lora(self.base_model):
-            torch.save(lora_state_dict(self.base_model),
+        if has_adapter(self.base_model):
+            torch.save(adapter_state_dict(self.base_model),
                        os.path.join(save_directory, "adapter_model.pt"))
         heads = {k: v.cpu() for k, v in self.heads_state_dict().items()}
         if heads:

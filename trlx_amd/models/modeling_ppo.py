@@ -395,9 +395,9 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
         elif return_ref_logits and self.peft_config is not None:
             # peft hydra: the base model WITHOUT adapters is the reference
             # (reference accelerate_ppo_trainer.py:74-77 + peft disable_adapter)
-            from .lora import lora_disabled
+            from .lora import adapters_disabled
 
-            with torch.no_grad(), lora_disabled(self.base_model):
+            with torch.no_grad(), adapters_disabled(self.base_model):
                 ref_logits = self.base_model(
                     input_ids, attention_mask=attention_mask, position_ids=position_ids,
                     logits_slice=logits_slice,
@@ -465,9 +465,9 @@ class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):
             else:
                 ref_logprobs = ops.logprobs_of_labels(rlm(rh), labels)
         elif self.peft_config is not None:
-            from .lora import lora_disabled
+            from .lora import adapters_disabled
 
-            with lora_disabled(self.base_model):
+            with adapters_disabled(self.base_model):
                 rout = self.base_model(input_ids, attention_mask=attention_mask,
                                        return_logits=False)
             rh = rout.last_hidden_state[:, lo:hi].contiguous()

@@ -422,12 +422,13 @@ def generate(
 
     try:
         shaping_graphable = shaping_fn is None or gen.graph_safe_shaping
+        has_virtual = getattr(model, "num_virtual_tokens", 0) > 0
         alibi = getattr(model.config, "position_encoding", None) == "alibi"
         # the graph engine needs the fused decode kernels (device-side cache
         # index); unsupported head dims use the eager loop's host-side state
         head_ok = getattr(model.config, "head_dim", 64) in (32, 64, 128, 256)
         if (device.type == "cuda" and shaping_graphable and gen.use_graph and not alibi
-                and head_ok
+                and head_ok and not has_virtual
                 and _graphs_enabled() and gen.max_new_tokens > 1 and gen.min_new_tokens == 0):
             engine = getattr(model, "_decode_engine", None)
             needed = T + gen.max_new_tokens
@@ -444,11 +445,14 @@ def generate(
 
 
 def _generate_eager(model, input_ids, attention_mask, gen: GenerateConfig, shaping_fn):
-    """Reference loop: CPU, shaping fns, or graphs disabled."""
+    """Reference loop: CPU, shaping fns, graphs disabled, or PEFT virtual
+    tokens (the prefill inserts nv prompt/prefix slots into the cache; all
+    decode positions shift by nv)."""
     B, T = input_ids.shape
     device = input_ids.device
+    nv = getattr(model, "num_virtual_tokens", 0)
     key_starts = (T - attention_mask.sum(-1)).to(torch.int32)
-    kv = model.new_kv_cache(B, T + gen.max_new_tokens, device=device)
+    kv = model.new_kv_cache(B, T + nv + gen.max_new_tokens, device=device)
 
     out = model(input_ids, attention_mask=attention_mask, kv_cache=kv, start_pos=0,
                 return_logits=False)
@@ -488,7 +492,7 @@ def _generate_eager(model, input_ids, attention_mask, gen: GenerateConfig, shapi
         last_tokens = next_tok
         if step == gen.max_new_tokens - 1:
             break
-        start_pos = T + step
+        start_pos = T + nv + step
         position_ids = (start_pos - key_starts).to(torch.int32).unsqueeze(1)
         seq_lens = torch.full((B,), start_pos + 1, dtype=torch.int32, device=device)
         # on GPU route through the same fused decode kernels as the graph

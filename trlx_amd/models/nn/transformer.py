@@ -86,6 +86,9 @@ class AttentionContext:
     cache_idx: Optional[torch.Tensor] = None
     # ALiBi additive bias [B, H_local, 1, T_keys] (Bloom)
     alibi: Optional[torch.Tensor] = None
+    # PREFIX_TUNING: per-layer K/V params + the cache positions they occupy
+    prefix_kv: Optional[object] = None
+    virtual_slots: Optional[torch.Tensor] = None  # [B, n] long
 
 
 class KVCache:
@@ -194,6 +197,16 @@ class Attention(nn.Module):
                 cos, sin = rope_tables
                 q, k = ops.apply_rope(q, k, cos, sin, positions=ctx.position_ids,
                                       interleaved=self.cfg.rope_interleaved, rot=self.rot)
+
+        if ctx.prefix_kv is not None:
+            # PREFIX_TUNING: overwrite K/V at the virtual slots with this
+            # layer's trained prefix (post-RoPE positions; prefixes unrotated)
+            pk = ctx.prefix_kv.prefix_k[self.layer_idx].to(k.dtype)
+            pv = ctx.prefix_kv.prefix_v[self.layer_idx].to(v.dtype)
+            B_, n_ = ctx.virtual_slots.shape
+            idx = ctx.virtual_slots.view(B_, 1, n_, 1).expand(B_, k.shape[1], n_, k.shape[3])
+            k = k.scatter(2, idx, pk.unsqueeze(0).expand(B_, -1, -1, -1))
+            v = v.scatter(2, idx, pv.unsqueeze(0).expand(B_, -1, -1, -1))
 
         if kv_cache is not None:
             if pre_scaled:
@@ -322,6 +335,55 @@ class Block(nn.Module):
         return m, s2
 
 
+def insert_virtual_rows(h, attention_mask, n, virt_values=None):
+    """Insert ``n`` virtual slots BETWEEN each row's left pads and its real
+    tokens (the pad-aware analog of peft's prepend: the masked prefix stays
+    contiguous, so key_starts/flash-kernel semantics hold unchanged).
+
+    Returns (h2 [B,T+n,H], mask2 [B,T+n], slots [B,n] cache positions of the
+    virtual tokens, strip_idx [B,T] gather indices recovering the real
+    positions from a T+n-length output).  ``virt_values`` [n, H] fills the
+    slots (zeros when None — prefix tuning overwrites K/V per layer anyway).
+    """
+    B, T, H = h.shape
+    device = h.device
+    if attention_mask is not None:
+        p = (T - attention_mask.sum(-1)).long()  # left-pad count per row
+    else:
+        p = torch.zeros(B, dtype=torch.long, device=device)
+    pos = torch.arange(T + n, device=device).unsqueeze(0)  # [1, T+n]
+    pe = p.unsqueeze(1)
+    is_virt = (pos >= pe) & (pos < pe + n)
+    src = torch.where(pos < pe, pos, (pos - n).clamp(min=0)).clamp(max=T - 1)
+    h2 = torch.gather(h, 1, src.unsqueeze(-1).expand(B, T + n, H))
+    if virt_values is not None:
+        virt_idx = (pos - pe).clamp(0, n - 1)
+        vfill = virt_values[virt_idx]  # [B, T+n, H]
+        h2 = torch.where(is_virt.unsqueeze(-1), vfill.to(h2.dtype), h2)
+    else:
+        h2 = h2.masked_fill(is_virt.unsqueeze(-1), 0)
+    if attention_mask is not None:
+        mask2 = torch.where(is_virt, torch.ones_like(is_virt, dtype=attention_mask.dtype),
+                            torch.gather(attention_mask, 1, src))
+    else:
+        mask2 = is_virt.to(torch.long) | (pos >= pe + n).to(torch.long)
+        mask2 = torch.ones(B, T + n, dtype=torch.long, device=device)
+    slots = pe + torch.arange(n, device=device).unsqueeze(0)  # [B, n]
+    t_pos = torch.arange(T, device=device).unsqueeze(0)
+    strip_idx = torch.where(t_pos < pe, t_pos, t_pos + n)  # [B, T]
+    return h2, mask2, slots, strip_idx
+
+
+def _strip_virtual(t, strip_idx):
+    """Gather the real positions back out of a T+n-length tensor."""
+    if t is None:
+        return None
+    B, T = strip_idx.shape
+    if t.dim() == 3:
+        return torch.gather(t, 1, strip_idx.unsqueeze(-1).expand(B, T, t.shape[-1]))
+    return torch.gather(t, 1, strip_idx)
+
+
 @dataclass
 class TransformerOutput:
     logits: Optional[torch.Tensor] = None
@@ -350,6 +412,11 @@ class CausalTransformer(nn.Module):
             self.lm_head.weight = self.embed_tokens.weight
         self.embd_pdrop = cfg.embd_pdrop
         self.gradient_checkpointing = False
+        # PEFT virtual-token adapters (models/lora.py apply_peft):
+        # PROMPT_TUNING sets soft_prompt, PREFIX_TUNING sets prefix_kv
+        self.soft_prompt = None
+        self.prefix_kv = None
+        self.num_virtual_tokens = 0
         if cfg.position_encoding == "rope":
             rot = int(cfg.head_dim * cfg.rope_pct)
             rot -= rot % 2
@@ -455,10 +522,27 @@ class CausalTransformer(nn.Module):
     ) -> TransformerOutput:
         """hidden_at_layer=k stashes the hidden state FED INTO layer k
         (negative counts from the end: -2 = input of the 2nd-to-last layer)."""
-        ctx = self.make_context(input_ids, attention_mask, start_pos, seq_lens, key_starts,
-                                position_ids)
-        ctx.cache_idx = cache_idx
-        h = self.embed_tokens(input_ids)
+        # PEFT virtual tokens: insert adapter slots between each row's left
+        # pads and real tokens at prefill/training time (decode steps reuse
+        # the prefix already in the KV cache); outputs are stripped back to
+        # the caller's T positions at the end
+        nv = self.num_virtual_tokens if (self.soft_prompt is not None
+                                         or self.prefix_kv is not None) else 0
+        strip_idx = None
+        if nv and start_pos == 0 and cache_idx is None:
+            h = self.embed_tokens(input_ids)
+            virt = self.soft_prompt.embeddings if self.soft_prompt is not None else None
+            h, attention_mask, slots, strip_idx = insert_virtual_rows(h, attention_mask, nv, virt)
+            ctx = self.make_context(h[..., 0], attention_mask, start_pos, seq_lens, None, None)
+            ctx.cache_idx = cache_idx
+            if self.prefix_kv is not None:
+                ctx.prefix_kv = self.prefix_kv
+                ctx.virtual_slots = slots
+        else:
+            ctx = self.make_context(input_ids, attention_mask, start_pos, seq_lens, key_starts,
+                                    position_ids)
+            ctx.cache_idx = cache_idx
+            h = self.embed_tokens(input_ids)
         if self.embed_norm is not None:
             h = self.embed_norm(h)
         if self.embed_positions is not None:
@@ -540,6 +624,13 @@ class CausalTransformer(nn.Module):
                 hidden_at = {k: gather_from_sp_replicated(v) for k, v in hidden_at.items()}
             elif hidden_at is not None:
                 hidden_at = gather_from_sp_replicated(hidden_at)
+        if strip_idx is not None:
+            # drop the virtual positions: callers see their original T
+            h = _strip_virtual(h, strip_idx)
+            if isinstance(hidden_at, dict):
+                hidden_at = {kk: _strip_virtual(v, strip_idx) for kk, v in hidden_at.items()}
+            elif hidden_at is not None:
+                hidden_at = _strip_virtual(hidden_at, strip_idx)
         logits = None
         if return_logits:
             # logits_slice: compute the [V]-wide projection only where the

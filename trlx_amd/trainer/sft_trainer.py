@@ -32,6 +32,12 @@ class SFTConfig(MethodConfig):
 class CausalLMWrapper(PreTrainedModelWrapper):
     """Bare LM wrapper (no heads) with generate."""
 
+    _supported_args = ["peft_config"]
+
+    def __init__(self, base_model, peft_config=None):
+        super().__init__(base_model)
+        self.peft_config = peft_config
+
     def forward(self, input_ids, attention_mask=None, position_ids=None, **kwargs):
         return self.base_model(input_ids, attention_mask=attention_mask, position_ids=position_ids)
 
@@ -64,9 +70,10 @@ class SFTTrainer(NativeRLTrainer):
             from ..models.nn.config import TransformerConfig
 
             return CausalLMWrapper.from_config(
-                TransformerConfig.from_dict(config.model.model_extra_configs["config"])
+                TransformerConfig.from_dict(config.model.model_extra_configs["config"]),
+                peft_config=config.model.peft_config,
             )
-        return CausalLMWrapper.from_pretrained(path)
+        return CausalLMWrapper.from_pretrained(path, peft_config=config.model.peft_config)
 
     def pp_train_minibatch(self, microbatches):
         """SFT CE loss under pipeline parallelism (reference NeMo SFT path,
