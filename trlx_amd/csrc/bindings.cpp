@@ -98,6 +98,9 @@ void stage_gemm_v3(const at::Tensor& a, const at::Tensor& w, const c10::optional
                    const c10::optional<at::Tensor>& nw, const c10::optional<at::Tensor>& nb,
                    bool norm_rms, double eps, long act,
                    const c10::optional<at::Tensor>& pstats_out);
+at::Tensor flash_prefill(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+                         const c10::optional<at::Tensor>& key_starts, long start_pos,
+                         double scale, long tk);
 void lm_sample_v3(const at::Tensor& x, const at::Tensor& wlm,
                   const c10::optional<at::Tensor>& blm, const at::Tensor& pstats, long nparts,
                   const at::Tensor& nw, const c10::optional<at::Tensor>& nb, at::Tensor& packed,
@@ -143,4 +146,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("lm_sample_v2", &lm_sample_v2);
   mod.def("stage_gemm_v3", &stage_gemm_v3);
   mod.def("lm_sample_v3", &lm_sample_v3);
+  mod.def("flash_prefill", &flash_prefill);
 }

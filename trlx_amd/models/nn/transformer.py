@@ -19,6 +19,7 @@ Design points:
 """
 
 import math
+import os
 from dataclasses import dataclass
 from typing import List, Optional, Tuple
 
@@ -208,6 +209,14 @@ class Attention(nn.Module):
             k = k.scatter(2, idx, pk.unsqueeze(0).expand(B_, -1, -1, -1))
             v = v.scatter(2, idx, pv.unsqueeze(0).expand(B_, -1, -1, -1))
 
+        flash_ok = (
+            T > 1 and x.is_cuda and q.dtype == torch.bfloat16
+            and self.head_dim in (64, 128) and ctx.alibi is None
+            and not torch.is_grad_enabled()
+            and (self.attn_pdrop == 0 or not self.training)
+            and ops.extension_available()
+            and os.environ.get("TRLX_AMD_NO_FLASH_PREFILL") != "1"
+        )
         if kv_cache is not None:
             if pre_scaled:
                 # cache stores unscaled k (decode kernel scales q itself)
@@ -220,8 +229,25 @@ class Attention(nn.Module):
                                            seq_starts=ctx.key_starts)
                 out = out.transpose(1, 2).reshape(B, T, -1)
                 return dense(self.o_proj, out)
+            if flash_ok:
+                # cache prefill through the flash kernel: reads the cache
+                # tensors in place (allocated stride Sk, valid start_pos+T)
+                out = ops.flash_prefill(q, k_full, v_full, ctx.key_starts, ctx.start_pos,
+                                        1.0 if pre_scaled else self.scale,
+                                        tk=ctx.start_pos + T)
+                out = out.transpose(1, 2).reshape(B, T, -1)
+                return dense(self.o_proj, out)
             k = k_full[:, :, : ctx.start_pos + T]
             v = v_full[:, :, : ctx.start_pos + T]
+        elif flash_ok:
+            # no_grad experience/training-free forward: online-softmax flash
+            # attention — the [B, H, T, T] scores never materialize
+            # (the memory wall at seq 1024-2048)
+            out = ops.flash_prefill(q.contiguous(), k.contiguous(), v.contiguous(),
+                                    ctx.key_starts, ctx.start_pos,
+                                    1.0 if pre_scaled else self.scale)
+            out = out.transpose(1, 2).reshape(B, T, -1)
+            return dense(self.o_proj, out)
 
         # prefill / training: rocBLAS batched GEMMs + fused causal softmax
         if self.num_kv_heads != self.num_heads:

@@ -687,3 +687,18 @@ def causal_softmax(
                 key_starts = key_starts.to(torch.int32).contiguous()
             return _CausalSoftmax.apply(scores, start_pos, key_starts, ext)
     return reference.causal_softmax(scores, start_pos, key_starts)
+
+
+def flash_prefill(q, k, v, key_starts=None, start_pos: int = 0, scale: float = 1.0,
+                  tk=None):
+    """Forward-only flash attention for prefill / the no_grad experience pass
+    (csrc/flash_prefill.hip): q [B,Hq,T,D] (bf16), k/v [B,Hkv,Sk,D] with the
+    first ``tk`` tokens valid (cache prefill passes the whole cache), causal +
+    left-pad masking via ``key_starts``.  Returns [B,Hq,T,D] bf16; the
+    [B,H,T,T] scores never materialize."""
+    ext = _require_ext("flash_prefill")
+    if ext is None or not q.is_cuda:
+        return reference.flash_prefill(q, k, v, key_starts, start_pos, scale, tk)
+    ks = key_starts.to(torch.int32).contiguous() if key_starts is not None else None
+    return ext.flash_prefill(q.contiguous(), k, v, ks, int(start_pos), float(scale),
+                             int(tk) if tk is not None else 0)

@@ -223,3 +223,27 @@ def fused_adamw(
         denom = (v / bc2).sqrt_().add_(eps)
         mp.addcdiv_(m, denom, value=-lr / bc1)
         p.copy_(mp.to(p.dtype))
+
+
+def flash_prefill(q, k, v, key_starts=None, start_pos: int = 0, scale: float = 1.0, tk=None):
+    """fp32 reference for the flash prefill kernel (masking semantics match
+    causal_softmax: key j visible to query i iff key_start <= j <= start_pos+i)."""
+    B, Hq, T, D = q.shape
+    Hkv, Sk = k.shape[1], k.shape[2]
+    tkv = int(tk) if tk else Sk
+    kk = k[:, :, :tkv].float()
+    vv = v[:, :, :tkv].float()
+    if Hq != Hkv:
+        rep = Hq // Hkv
+        kk = kk.repeat_interleave(rep, dim=1)
+        vv = vv.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(q.float() * scale, kk.transpose(-1, -2))
+    i = torch.arange(T, device=q.device).view(1, 1, T, 1)
+    j = torch.arange(tkv, device=q.device).view(1, 1, 1, tkv)
+    mask = j <= (start_pos + i)
+    if key_starts is not None:
+        mask = mask & (j >= key_starts.view(B, 1, 1, 1).to(q.device))
+    scores = scores.masked_fill(~mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    probs = torch.nan_to_num(probs, nan=0.0)
+    return torch.matmul(probs, vv).to(q.dtype)
