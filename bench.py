@@ -98,14 +98,17 @@ def run_cycle(trainer, config, phase_times=None):
     trainer.store.clear_history()
     trainer.make_experience(config.method.num_rollouts)
     t1 = tick()
+    trainer.model.eval()  # learn()'s convention: PPO train-step forward sans dropout
     for _ in range(config.method.ppo_epochs):
         loader = trainer.store.create_loader(config.train.batch_size, shuffle=True)
         for minibatch in MiniBatchIterator(loader, trainer.mb_size, trainer.num_mb):
             for microbatch in minibatch:
                 with trainer._accumulate():
-                    loss, _ = trainer.loss(microbatch)
-                    trainer.model.train()
-                    loss.backward()
+                    # same body as NativeRLTrainer.learn: the captured
+                    # fwd+bwd hipGraph when eligible, eager otherwise
+                    if trainer._graphed_loss_backward(microbatch) is None:
+                        loss, _ = trainer.loss(microbatch)
+                        loss.backward()
             trainer.reducer.finalize()
             trainer.opt.step()
             trainer.opt.zero_grad()
