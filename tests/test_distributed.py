@@ -710,14 +710,22 @@ def test_pp_hydra_model_equivalence():
     _spawn(_worker_pp_hydra_model, 29525)
 
 
-def _ppo_cfg_for_pp(rank, pp, tp=1):
+def _ppo_cfg_for_pp(rank, pp, tp=1, arch="gpt2"):
     import trlx_amd  # noqa: F401
     from trlx_amd.data.default_configs import default_ppo_config
     from trlx_amd.models.nn.config import TransformerConfig
 
+    if arch == "gpt_neox":
+        # the BASELINE config #5 architecture shape: parallel residual + RoPE
+        tiny = TransformerConfig(vocab_size=300, hidden_size=32, num_layers=4, num_heads=2,
+                                 max_position_embeddings=128, arch_name="gpt_neox",
+                                 position_encoding="rope", rope_pct=0.25,
+                                 parallel_residual=True, activation="gelu_new",
+                                 tie_word_embeddings=False)
+    else:
+        tiny = TransformerConfig(vocab_size=300, hidden_size=32, num_layers=4, num_heads=2,
+                                 max_position_embeddings=128, arch_name="gpt2")
     cfg = default_ppo_config()
-    tiny = TransformerConfig(vocab_size=300, hidden_size=32, num_layers=4, num_heads=2,
-                             max_position_embeddings=128, arch_name="gpt2")
     cfg.model.model_path = "tiny"
     cfg.model.model_extra_configs = {"config": tiny.to_dict()}
     cfg.model.num_layers_unfrozen = 1
@@ -930,3 +938,30 @@ def _worker_cross_tp_reshard(rank):
 
 def test_cross_tp_size_resharding():
     _spawn_n(_worker_cross_tp_reshard, 29530, 4)
+
+
+def _worker_pp_neox_ppo(rank):
+    """PP=2 PPO on a NeoX-shaped arch (parallel residual + partial RoPE) —
+    the BASELINE config #5 architecture class at test scale."""
+    import trlx_amd
+    from trlx_amd.parallel import topo
+
+    cfg = _ppo_cfg_for_pp(rank, pp=2, arch="gpt_neox")
+
+    def reward_fn(samples, prompts, outputs, **kw):
+        return [float(len(o)) for o in outputs]
+
+    try:
+        trainer = trlx_amd.train(
+            reward_fn=reward_fn,
+            prompts=["aa", "bb", "cc", "dd"],
+            eval_prompts=["aa"],
+            config=cfg,
+        )
+        assert trainer.iter_count == 2
+    finally:
+        topo.reset()
+
+
+def test_pp_neox_arch_ppo():
+    _spawn(_worker_pp_neox_ppo, 29531)
