@@ -128,7 +128,7 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
     }
     __syncthreads();
   }
-  for (int i = threadIdx.x; i < H; i += BLOCK) atomicAdd(&dw[i], dwacc[i]);
+  for (int i = threadIdx.x; i < H; i += BLOCK) dw[(size_t)blockIdx.x * H + i] = dwacc[i];
 }
 
 template <typename T, bool HAS_BIAS, bool HAS_RES>
@@ -274,9 +274,11 @@ __global__ void layernorm_bwd_kernel(const T* __restrict__ x, const T* __restric
     }
     __syncthreads();
   }
+  // per-block partial slabs (plain stores; v1's atomicAdd tail serialized
+  // ~2048 adds per address = ~60 us/launch — profile r02)
   for (int i = threadIdx.x; i < H; i += BLOCK) {
-    atomicAdd(&dw[i], dwacc[i]);
-    atomicAdd(&db[i], dbacc[i]);
+    dw[(size_t)blockIdx.x * H + i] = dwacc[i];
+    db[(size_t)blockIdx.x * H + i] = dbacc[i];
   }
 }
 
@@ -339,11 +341,12 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
   at::Tensor drc;
   if (has_dres) drc = dres->contiguous();
   auto dx = at::empty_like(x);
-  auto dwf = at::zeros({H}, x.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
   auto dyc = dy.contiguous();
   const int grid = pick_grid(N);
+  auto dwp = at::empty({std::max(grid, 1), H}, x.options().dtype(at::kFloat));
+  if (N == 0) dwp.zero_();
   const size_t lds = (size_t)H * sizeof(float);
   if (N > 0) {
     if (x.dtype() == at::kBFloat16) {
@@ -354,23 +357,24 @@ std::vector<at::Tensor> rmsnorm_bwd(const at::Tensor& x, const at::Tensor& w,
       if (has_dres)
         rmsnorm_bwd_kernel<bf16_t, true><<<grid, BLOCK, lds, stream>>>(
             xp, wp, invr.data_ptr<float>(), dyp,
-            reinterpret_cast<const bf16_t*>(drc.data_ptr()), dxp, dwf.data_ptr<float>(), H, N);
+            reinterpret_cast<const bf16_t*>(drc.data_ptr()), dxp, dwp.data_ptr<float>(), H, N);
       else
         rmsnorm_bwd_kernel<bf16_t, false><<<grid, BLOCK, lds, stream>>>(
-            xp, wp, invr.data_ptr<float>(), dyp, nullptr, dxp, dwf.data_ptr<float>(), H, N);
+            xp, wp, invr.data_ptr<float>(), dyp, nullptr, dxp, dwp.data_ptr<float>(), H, N);
     } else {
       if (has_dres)
         rmsnorm_bwd_kernel<float, true><<<grid, BLOCK, lds, stream>>>(
             x.data_ptr<float>(), wc.data_ptr<float>(), invr.data_ptr<float>(),
             dyc.data_ptr<float>(), drc.data_ptr<float>(), dx.data_ptr<float>(),
-            dwf.data_ptr<float>(), H, N);
+            dwp.data_ptr<float>(), H, N);
       else
         rmsnorm_bwd_kernel<float, false><<<grid, BLOCK, lds, stream>>>(
             x.data_ptr<float>(), wc.data_ptr<float>(), invr.data_ptr<float>(),
-            dyc.data_ptr<float>(), nullptr, dx.data_ptr<float>(), dwf.data_ptr<float>(), H, N);
+            dyc.data_ptr<float>(), nullptr, dx.data_ptr<float>(), dwp.data_ptr<float>(), H, N);
     }
     HIP_CHECK_LAST();
   }
+  auto dwf = dwp.sum(0);
   return {dx, dwf.to(w.dtype())};
 }
 
@@ -442,12 +446,16 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x, const at::Tensor& w,
   at::Tensor drc;
   if (has_dres) drc = dres->contiguous();
   auto dx = at::empty_like(x);
-  auto dwf = at::zeros({H}, x.options().dtype(at::kFloat));
-  auto dbf = at::zeros({H}, x.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
   auto dyc = dy.contiguous();
   const int grid = pick_grid(N);
+  auto dwp = at::empty({std::max(grid, 1), H}, x.options().dtype(at::kFloat));
+  auto dbp = at::empty({std::max(grid, 1), H}, x.options().dtype(at::kFloat));
+  if (N == 0) {
+    dwp.zero_();
+    dbp.zero_();
+  }
   const size_t lds = 2 * (size_t)H * sizeof(float);
   if (N > 0) {
     if (x.dtype() == at::kBFloat16) {
@@ -458,25 +466,27 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& x, const at::Tensor& w,
       if (has_dres)
         layernorm_bwd_kernel<bf16_t, true><<<grid, BLOCK, lds, stream>>>(
             xp, wp, mean.data_ptr<float>(), invstd.data_ptr<float>(), dyp,
-            reinterpret_cast<const bf16_t*>(drc.data_ptr()), dxp, dwf.data_ptr<float>(),
-            dbf.data_ptr<float>(), H, N);
+            reinterpret_cast<const bf16_t*>(drc.data_ptr()), dxp, dwp.data_ptr<float>(),
+            dbp.data_ptr<float>(), H, N);
       else
         layernorm_bwd_kernel<bf16_t, false><<<grid, BLOCK, lds, stream>>>(
             xp, wp, mean.data_ptr<float>(), invstd.data_ptr<float>(), dyp, nullptr, dxp,
-            dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
+            dwp.data_ptr<float>(), dbp.data_ptr<float>(), H, N);
     } else {
       if (has_dres)
         layernorm_bwd_kernel<float, true><<<grid, BLOCK, lds, stream>>>(
             x.data_ptr<float>(), wc.data_ptr<float>(), mean.data_ptr<float>(),
             invstd.data_ptr<float>(), dyc.data_ptr<float>(), drc.data_ptr<float>(),
-            dx.data_ptr<float>(), dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
+            dx.data_ptr<float>(), dwp.data_ptr<float>(), dbp.data_ptr<float>(), H, N);
       else
         layernorm_bwd_kernel<float, false><<<grid, BLOCK, lds, stream>>>(
             x.data_ptr<float>(), wc.data_ptr<float>(), mean.data_ptr<float>(),
             invstd.data_ptr<float>(), dyc.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
-            dwf.data_ptr<float>(), dbf.data_ptr<float>(), H, N);
+            dwp.data_ptr<float>(), dbp.data_ptr<float>(), H, N);
     }
     HIP_CHECK_LAST();
   }
+  auto dwf = dwp.sum(0);
+  auto dbf = dbp.sum(0);
   return {dx, dwf.to(w.dtype()), dbf.to(w.dtype())};
 }
