@@ -86,6 +86,9 @@ def test_generation_prefill_flash_consistent():
                             swiglu=True, attn_bias=False, mlp_bias=False,
                             intermediate_size=256, tie_word_embeddings=False)
     model = CausalTransformer(cfg).cuda().bfloat16().eval()
+    # rope tables must stay fp32 (cast_compute semantics)
+    model.rope_cos = model.rope_cos.float()
+    model.rope_sin = model.rope_sin.float()
     ids = torch.randint(3, 400, (4, 21), device="cuda")
     mask = torch.ones_like(ids)
     mask[1, :4] = 0

@@ -542,6 +542,8 @@ def qkv_prep(
     ext = _require_ext("qkv_prep")
     if positions is not None:
         positions = positions.to(torch.int32).contiguous()
+    if cos is not None and cos.dtype != torch.float32:
+        cos, sin = cos.float(), sin.float()  # kernels require fp32 trig tables
     return _QKVPrep.apply(qkv.contiguous(), num_heads, num_kv_heads, head_dim, cos, sin,
                           positions, qscale, rot, interleaved, ext)
 
