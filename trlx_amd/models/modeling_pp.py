@@ -268,12 +268,13 @@ class PipelinedPPOModel(nn.Module):
     # --- generation ----------------------------------------------------------
 
     @torch.no_grad()
-    def generate(self, input_ids, attention_mask=None, **kwargs):
+    def generate(self, input_ids, attention_mask=None, shaping_fn=None, **kwargs):
         """Token-level pipelined decode (reference NeMo generate parity,
         modeling_nemo_ppo.py:1158-1222): prefill fills per-stage KV caches in
         one pipeline pass; each subsequent token makes one pipeline trip and
         the sampled id is broadcast from the last stage.  All PP ranks return
-        the same [B, T+new] samples."""
+        the same [B, T+new] samples.  ``shaping_fn(logits, hidden, last_tok)``
+        runs on the LAST stage before sampling (ILQL logit shaping)."""
         gen = GenerateConfig.from_kwargs(**kwargs)
         st = self.stage
         device = self.device
@@ -303,9 +304,12 @@ class PipelinedPPOModel(nn.Module):
             group = topo.pp_group()
             finished = torch.zeros(B, dtype=torch.bool, device=device)
             generated = []
+            last_tok = input_ids[:, -1]
             for step in range(gen.max_new_tokens):
                 if st.is_last:
                     logits = st.project(h[:, -1:, :])[:, 0].float()
+                    if shaping_fn is not None:
+                        logits = shaping_fn(logits, h[:, -1], last_tok)
                     if eos_id is not None and step < gen.min_new_tokens:
                         logits[:, eos_id] = float("-inf")
                     if gen.do_sample:
@@ -319,6 +323,7 @@ class PipelinedPPOModel(nn.Module):
                 if dist.is_initialized():
                     dist.broadcast(tok, src=src, group=group)
                 generated.append(tok)
+                last_tok = tok
                 if eos_id is not None:
                     finished = finished | (tok == eos_id)
                     if bool(finished.all()):
@@ -431,3 +436,88 @@ def merge_pp_checkpoint(directory: str) -> Dict[str, torch.Tensor]:
     # merge_state_dicts_tp pattern-matches on submodule names, so the
     # wrapper-prefixed frozen_head.blocks.* keys merge like base layers
     return merge_state_dicts_tp(per_tp, cfg, saved_tp)
+
+
+class PipelinedILQLModel(PipelinedPPOModel):
+    """PP-sharded ILQL model: the Q/V/target-Q heads live on the LAST stage
+    (reference NeMo ILQLGPT parity, modeling_nemo_ilql.py:255-785 — heads
+    only on the post_process stage, shaped generation inside the pipeline
+    inference fn)."""
+
+    def __init__(self, config: TransformerConfig, two_qs: bool = True, alpha: float = 0.99,
+                 **kwargs):
+        kwargs.pop("with_value_head", None)
+        kwargs.setdefault("num_layers_unfrozen", -1)
+        super().__init__(config, with_value_head=False, **kwargs)
+        self.two_qs = two_qs
+        self.alpha = alpha
+        self.ilql_heads = None
+        if self.stage.is_last:
+            from .modeling_ilql import ILQLHeads
+
+            self.ilql_heads = ILQLHeads(config.hidden_size, config.vocab_size, two_qs, alpha)
+
+    def sync_target_q_heads(self):
+        if self.ilql_heads is not None:
+            self.ilql_heads.sync_target_q_heads()
+
+    @torch.no_grad()
+    def generate(self, input_ids, attention_mask=None, beta: float = 1.0,
+                 max_new_tokens: int = 32, max_length: int = 1024, temperature: float = 1.0,
+                 top_k: int = 20, logit_mask=None, pad_token_id=None, eos_token_id=None,
+                 **kwargs):
+        """Shaped pipelined generation: log pi + beta*(minQ - V) computed on
+        the last stage's hidden (reference modeling_nemo_ilql.py:685-739 —
+        the shaping lives inside the pipeline inference fn)."""
+        import torch.nn.functional as F
+
+        if attention_mask is None and pad_token_id is not None:
+            attention_mask = input_ids.not_equal(pad_token_id).long()
+
+        shaping = None
+        if self.stage.is_last:
+            def shaping(logits, hidden, last_tokens):
+                hs = hidden.unsqueeze(1)
+                qs, target_qs, vs = self.ilql_heads(hs)
+                if self.two_qs:
+                    qv = torch.minimum(target_qs[0][:, -1, :], target_qs[1][:, -1, :])
+                else:
+                    qv = target_qs[0][:, -1, :]
+                v = vs[:, -1, :]
+                if logit_mask is not None:
+                    m = logit_mask[last_tokens.to(logit_mask.device)]
+                    logits = logits.masked_fill(m.to(logits.device), float("-inf"))
+                adv = (qv - v).to(logits.dtype)
+                return F.log_softmax(logits, -1) + beta * adv
+
+        max_new_tokens = min(max_new_tokens, max_length - input_ids.shape[1])
+        return super().generate(
+            input_ids, attention_mask, shaping_fn=shaping,
+            max_new_tokens=max_new_tokens, do_sample=temperature > 0,
+            temperature=temperature if temperature > 0 else 1.0, top_k=top_k,
+            eos_token_id=eos_token_id,
+            pad_token_id=eos_token_id if eos_token_id is not None else pad_token_id,
+        )
+
+    def generate_eval(self, *args, **kwargs):
+        return self.generate(*args, **kwargs)
+
+    @classmethod
+    def from_any(cls, path_or_config, two_qs=True, alpha=0.99, **kwargs):
+        heads = None
+        if isinstance(path_or_config, TransformerConfig):
+            cfg = path_or_config
+            full_sd = None
+        elif isinstance(path_or_config, str) and os.path.isdir(path_or_config):
+            from .nn.convert import load_hf_dir
+
+            cfg, full_sd = load_hf_dir(path_or_config)
+        elif isinstance(path_or_config, str) and path_or_config in PRESETS:
+            cfg = preset(path_or_config)
+            full_sd = None
+        else:
+            raise OSError(f"cannot build a pipelined ILQL model from {path_or_config!r}")
+        model = cls(cfg, two_qs=two_qs, alpha=alpha, **kwargs)
+        if full_sd is not None:
+            model.load_full_base_state_dict(full_sd)
+        return model

@@ -102,6 +102,16 @@ class ILQLTrainer(NativeRLTrainer):
         path = config.model.model_path
         kwargs = dict(two_qs=config.method.two_qs, alpha=config.method.alpha,
                       peft_config=config.model.peft_config)
+        if config.train.pipeline_parallel_size > 1:
+            if config.model.model_arch_type == "seq2seq":
+                raise NotImplementedError("seq2seq + pipeline parallelism is not supported")
+            from ..models.modeling_pp import PipelinedILQLModel
+            from ..models.nn.config import TransformerConfig
+
+            kwargs.pop("peft_config")
+            if isinstance(path, str) and config.model.model_extra_configs.get("config"):
+                path = TransformerConfig.from_dict(config.model.model_extra_configs["config"])
+            return PipelinedILQLModel.from_any(path, **kwargs)
         if config.model.model_arch_type == "seq2seq":
             from ..models.modeling_seq2seq import AutoModelForSeq2SeqLMWithILQLHeads
             from ..models.nn.seq2seq import Seq2SeqConfig
@@ -121,6 +131,28 @@ class ILQLTrainer(NativeRLTrainer):
     def post_backward_callback(self):
         if self.iter_count % self.config.method.steps_for_target_q_sync == 0:
             self.unwrapped_model.sync_target_q_heads()
+
+    def pp_train_minibatch(self, microbatches):
+        """ILQL loss under pipeline parallelism: logits + Q/V heads on the
+        last stage (reference NeMo ILQLGPT get_forward_output_and_loss_func,
+        modeling_nemo_ilql.py:612-683)."""
+        model = self.model
+        device = self.device
+        mbs = []
+        for batch in microbatches:
+            b = to_device(batch, device)
+            mbs.append({"input_ids": b.input_ids, "attention_mask": b.attention_mask,
+                        "_batch": b})
+
+        def loss_fn(h, mb):
+            b = mb["_batch"]
+            logits = model.stage.project(h)
+            qs, target_qs, vs = model.ilql_heads(h, states_ixs=b.states_ixs,
+                                                 actions_ixs=b.actions_ixs)
+            loss, stats = self.ilql.loss((logits, (qs, target_qs, vs)), b)
+            return loss, {k: float(v) for k, v in stats.items()}
+
+        return model.forward_backward(mbs, loss_fn)
 
     def loss(self, batch: ILQLBatch):
         batch = to_device(batch, self.device)
