@@ -16,8 +16,9 @@ end-to-end RLHF pipeline throughput, nothing skipped.
 
 Data: synthetic prompts over a model-vocab-sized synthetic vocabulary with perfect
 decode/encode round-trip; weights: random-init GPT-2-small; reward: cheap
-deterministic function of the sample string on rank 0 (scatter to ranks, as
-the real protocol does).
+deterministic stateless function of the sample string, scored on EVERY rank
+(method.local_rewards — the NeMo-path protocol; the gather-to-rank-0 +
+scatter protocol is implemented and gloo-tested but serializes at DP=8).
 
 Launch (the driver's contract):
   python bench.py --gpus N --steps K --warmup W
@@ -69,6 +70,10 @@ def build_trainer(args):
     config.method.gen_kwargs = dict(
         max_new_tokens=args.max_new_tokens, top_k=0, top_p=1.0, do_sample=True
     )
+    # the synthetic reward is stateless + deterministic: score on every rank
+    # (the NeMo-path protocol) instead of serializing through rank 0 — at
+    # DP=8 the gather->rank-0-score->scatter round trip idles 7 ranks
+    config.method.local_rewards = True
 
     def reward_fn(samples, prompts, outputs, **kwargs):
         # cheap deterministic stand-in for the sentiment classifier
@@ -219,6 +224,7 @@ def main():
                 "ppo_epochs": args.ppo_epochs,
                 "num_layers_unfrozen": args.num_layers_unfrozen,
                 "parallelism": f"dp{world}",
+                "local_rewards": True,
             },
             "phases_ms_per_step": {k: round(1000 * v / args.steps, 1)
                                    for k, v in phase_times.items()},

@@ -1010,3 +1010,48 @@ def _worker_pp_ilql_train(rank):
 
 def test_pp_ilql_end_to_end():
     _spawn(_worker_pp_ilql_train, 29532)
+
+
+def _worker_ppo_local_rewards(rank):
+    """DP=2 PPO with method.local_rewards: every rank scores its own rollouts
+    (no rank-0 gather/scatter); DP replicas still converge identically."""
+    import trlx_amd
+    from trlx_amd.data.default_configs import default_ppo_config
+    from trlx_amd.models.nn.config import TransformerConfig
+
+    cfg = default_ppo_config()
+    tiny = TransformerConfig(vocab_size=300, hidden_size=32, num_layers=2, num_heads=2,
+                             max_position_embeddings=128, arch_name="gpt2")
+    cfg.model.model_path = "tiny"
+    cfg.model.model_extra_configs = {"config": tiny.to_dict()}
+    cfg.model.num_layers_unfrozen = 1
+    cfg.tokenizer.tokenizer_path = "byte"
+    cfg.train.seq_length = 32
+    cfg.train.batch_size = 2
+    cfg.train.total_steps = 2
+    cfg.train.eval_interval = 2
+    cfg.train.checkpoint_interval = 100
+    cfg.train.tracker = None
+    cfg.train.save_best = False
+    cfg.train.checkpoint_dir = f"/tmp/dist_ppo_lr_{rank}"
+    cfg.method.num_rollouts = 4
+    cfg.method.chunk_size = 2
+    cfg.method.ppo_epochs = 1
+    cfg.method.local_rewards = True
+    cfg.method.gen_kwargs = dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True)
+
+    def reward_fn(samples, prompts, outputs, **kw):
+        return [float(len(o)) for o in outputs]  # stateless + deterministic
+
+    trainer = trlx_amd.train(reward_fn=reward_fn, prompts=["aa", "bb", "cc", "dd"],
+                             eval_prompts=["aa"], config=cfg)
+    assert trainer.iter_count == 2
+    for p in trainer.model.parameters():
+        if p.requires_grad:
+            buf = [torch.empty_like(p) for _ in range(WORLD)]
+            dist.all_gather(buf, p.detach())
+            assert torch.allclose(buf[0], buf[1], atol=1e-6)
+
+
+def test_ppo_local_rewards_dp():
+    _spawn(_worker_ppo_local_rewards, 29533)

@@ -94,6 +94,12 @@ class PPOConfig(MethodConfig):
     ref_std: Optional[float] = None
     cliprange_reward: float = 10.0
     num_value_layers_unfrozen: int = 0
+    # score rollouts on EVERY rank instead of the reference's gather-to-rank-0
+    # + scatter protocol (accelerate_ppo_trainer.py:292-338).  Correct only
+    # for stateless/deterministic reward fns; the NeMo path always scores
+    # locally (nemo_ppo_trainer.py:195-197).  At DP=8 the rank-0 round trip
+    # serializes ~10 ms per chunk while 7 ranks idle.
+    local_rewards: bool = False
     gen_kwargs: Dict[str, Any] = field(default_factory=lambda: dict(max_new_tokens=40, top_k=0, top_p=1.0, do_sample=True))
     gen_experience_kwargs: Optional[Dict[str, Any]] = None
 

@@ -340,10 +340,12 @@ class PPOTrainer(NativeRLTrainer):
             metadata_local = {k: v for k, v in batch.items()
                               if k not in ("input_ids", "attention_mask")}
 
-            if topo.tp_size() > 1 or topo.pp_size() > 1:
-                # model-parallel mode: every rank scores its own samples
-                # locally (TP/PP peers hold identical rollouts) — the
-                # NeMo-path protocol (reference nemo_ppo_trainer.py:195-197)
+            if (topo.tp_size() > 1 or topo.pp_size() > 1
+                    or getattr(self.config.method, "local_rewards", False)):
+                # model-parallel mode (or method.local_rewards): every rank
+                # scores its own samples locally (TP/PP peers hold identical
+                # rollouts; under DP this requires a stateless reward fn) —
+                # the NeMo-path protocol (reference nemo_ppo_trainer.py:195-197)
                 l_samples, l_prompts, l_outputs = self.decode(
                     prompt_tensors, samples, append_eos_token=True)
                 rollout_score_time = time()
@@ -394,7 +396,8 @@ class PPOTrainer(NativeRLTrainer):
                     scores = all_scores[0].clone().detach()
             scores_mask = scores != -np.inf
 
-            if topo.tp_size() <= 1 and topo.pp_size() <= 1 and self.world_size == 1:
+            if (topo.tp_size() <= 1 and topo.pp_size() <= 1 and self.world_size == 1
+                    and not getattr(self.config.method, "local_rewards", False)):
                 # single process: the gathered batch IS the local batch
                 str_samples, str_prompts, str_outputs = (
                     all_str_samples, all_str_prompts, all_str_outputs)
