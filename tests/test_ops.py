@@ -545,3 +545,17 @@ def test_gpu_lm_logprobs_train(N):
     _assert_close(h.grad.cpu(), hr.grad, atol=5e-2, name="lmlp_train dh")
     # dW accumulates over N rows of bf16 products; tolerance scales with N
     _assert_close(w.grad.cpu(), wr.grad, atol=0.3, name="lmlp_train dw")
+
+
+def test_flash_attention_cpu_fallback_differentiable():
+    """ops.flash_attention falls back to the fp32 reference off-GPU and stays
+    differentiable (autograd handles the backward there)."""
+    torch.manual_seed(11)
+    q = torch.randn(2, 2, 12, 64, requires_grad=True)
+    k = torch.randn(2, 2, 12, 64, requires_grad=True)
+    v = torch.randn(2, 2, 12, 64, requires_grad=True)
+    ks = torch.tensor([0, 3], dtype=torch.int32)
+    out = ops.flash_attention(q, k, v, ks, 0.125)
+    out.sum().backward()
+    assert q.grad is not None and k.grad is not None and v.grad is not None
+    assert torch.all(k.grad[1, :, :3] == 0)

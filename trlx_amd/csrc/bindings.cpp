@@ -101,6 +101,16 @@ void stage_gemm_v3(const at::Tensor& a, const at::Tensor& w, const c10::optional
 at::Tensor flash_prefill(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
                          const c10::optional<at::Tensor>& key_starts, long start_pos,
                          double scale, long tk);
+std::vector<at::Tensor> flash_prefill_lse(const at::Tensor& q, const at::Tensor& k,
+                                          const at::Tensor& v,
+                                          const c10::optional<at::Tensor>& key_starts,
+                                          long start_pos, double scale, long tk,
+                                          bool want_lse);
+std::vector<at::Tensor> flash_prefill_bwd(const at::Tensor& q, const at::Tensor& k,
+                                          const at::Tensor& v, const at::Tensor& out,
+                                          const at::Tensor& dout, const at::Tensor& lse,
+                                          const c10::optional<at::Tensor>& key_starts,
+                                          double scale);
 void lm_sample_v3(const at::Tensor& x, const at::Tensor& wlm,
                   const c10::optional<at::Tensor>& blm, const at::Tensor& pstats, long nparts,
                   const at::Tensor& nw, const c10::optional<at::Tensor>& nb, at::Tensor& packed,
@@ -147,4 +157,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("stage_gemm_v3", &stage_gemm_v3);
   mod.def("lm_sample_v3", &lm_sample_v3);
   mod.def("flash_prefill", &flash_prefill);
+  mod.def("flash_prefill_lse", &flash_prefill_lse);
+  mod.def("flash_prefill_bwd", &flash_prefill_bwd);
 }

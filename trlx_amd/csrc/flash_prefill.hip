@@ -39,9 +39,10 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_fp;
 template <int D>
 __global__ __launch_bounds__(FBLOCK, 2) void flash_prefill_kernel(
     const bf16_t* __restrict__ q, const bf16_t* __restrict__ k, const bf16_t* __restrict__ v,
-    const int* __restrict__ key_starts, bf16_t* __restrict__ out, int B, int Hq, int Hkv,
-    int T, int Tk, int Sk /* allocated token stride of k/v (cache prefill) */,
-    int start_pos, float scale) {
+    const int* __restrict__ key_starts, bf16_t* __restrict__ out,
+    float* __restrict__ lse /* [B,Hq,T] logsumexp of the SCALED scores; may be null */,
+    int B, int Hq, int Hkv, int T, int Tk,
+    int Sk /* allocated token stride of k/v (cache prefill) */, int start_pos, float scale) {
   constexpr int DPAD = D + 8;    // LDS row stride (shorts), bank-staggered
   constexpr int KPAD = TK + 8;
 
@@ -205,8 +206,19 @@ __global__ __launch_bounds__(FBLOCK, 2) void flash_prefill_kernel(
     }
   }
 
-  // ---- epilogue: O / l -------------------------------------------------------
+  // ---- epilogue: O / l (+ optional logsumexp for the backward) --------------
   const int c = lane & 15;
+  if (lse != nullptr && c == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + qrow0 + (lane >> 4) * 4 + r;
+      if (qrow < T) {
+        // rows with no visible keys store +inf so exp(S - L) == 0 in bwd
+        const float L = l_row[r] > 0.f ? m_row[r] + __logf(l_row[r]) : INFINITY;
+        lse[((size_t)b * Hq + h) * T + qrow] = L;
+      }
+    }
+  }
 #pragma unroll
   for (int d = 0; d < D / 16; ++d) {
 #pragma unroll
@@ -222,9 +234,22 @@ __global__ __launch_bounds__(FBLOCK, 2) void flash_prefill_kernel(
 
 }  // namespace
 
+std::vector<at::Tensor> flash_prefill_lse(const at::Tensor& q, const at::Tensor& k,
+                                          const at::Tensor& v,
+                                          const c10::optional<at::Tensor>& key_starts,
+                                          long start_pos, double scale, long tk, bool want_lse);
+
 at::Tensor flash_prefill(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
                          const c10::optional<at::Tensor>& key_starts, long start_pos,
                          double scale, long tk) {
+  return flash_prefill_lse(q, k, v, key_starts, start_pos, scale, tk, false)[0];
+}
+
+std::vector<at::Tensor> flash_prefill_lse(const at::Tensor& q, const at::Tensor& k,
+                                          const at::Tensor& v,
+                                          const c10::optional<at::Tensor>& key_starts,
+                                          long start_pos, double scale, long tk,
+                                          bool want_lse) {
   TORCH_CHECK(q.is_cuda() && q.dtype() == at::kBFloat16 && q.dim() == 4 && q.is_contiguous());
   TORCH_CHECK(k.is_contiguous() && v.is_contiguous());
   const int B = q.size(0), Hq = q.size(1), T = q.size(2), D = q.size(3);
@@ -233,7 +258,8 @@ at::Tensor flash_prefill(const at::Tensor& q, const at::Tensor& k, const at::Ten
   TORCH_CHECK(Tk <= Sk);
   TORCH_CHECK(k.size(3) == D && Hq % Hkv == 0);
   auto out = at::empty_like(q);
-  if (q.numel() == 0) return out;
+  auto lse = want_lse ? at::empty({B, Hq, T}, q.options().dtype(at::kFloat)) : at::Tensor();
+  if (q.numel() == 0) return {out, lse};
   const int* ks = nullptr;
   at::Tensor ksc;
   if (key_starts.has_value()) {
@@ -247,18 +273,19 @@ at::Tensor flash_prefill(const at::Tensor& q, const at::Tensor& k, const at::Ten
   auto kp = reinterpret_cast<const bf16_t*>(k.data_ptr());
   auto vp = reinterpret_cast<const bf16_t*>(v.data_ptr());
   auto op = reinterpret_cast<bf16_t*>(out.data_ptr());
+  float* lp = want_lse ? lse.data_ptr<float>() : nullptr;
   switch (D) {
     case 64:
-      flash_prefill_kernel<64><<<grid, FBLOCK, 0, stream>>>(qp, kp, vp, ks, op, B, Hq, Hkv, T,
-                                                            Tk, Sk, (int)start_pos, (float)scale);
+      flash_prefill_kernel<64><<<grid, FBLOCK, 0, stream>>>(
+          qp, kp, vp, ks, op, lp, B, Hq, Hkv, T, Tk, Sk, (int)start_pos, (float)scale);
       break;
     case 128:
-      flash_prefill_kernel<128><<<grid, FBLOCK, 0, stream>>>(qp, kp, vp, ks, op, B, Hq, Hkv, T,
-                                                             Tk, Sk, (int)start_pos, (float)scale);
+      flash_prefill_kernel<128><<<grid, FBLOCK, 0, stream>>>(
+          qp, kp, vp, ks, op, lp, B, Hq, Hkv, T, Tk, Sk, (int)start_pos, (float)scale);
       break;
     default:
       TORCH_CHECK(false, "flash_prefill: head dim must be 64 or 128, got ", D);
   }
   HIP_CHECK_LAST();
-  return out;
+  return {out, lse};
 }

@@ -249,6 +249,25 @@ class Attention(nn.Module):
             out = out.transpose(1, 2).reshape(B, T, -1)
             return dense(self.o_proj, out)
 
+        flash_train_ok = (
+            T > 1 and x.is_cuda and q.dtype == torch.bfloat16
+            and self.head_dim in (64, 128) and ctx.alibi is None
+            and torch.is_grad_enabled() and kv_cache is None and ctx.start_pos == 0
+            and (self.attn_pdrop == 0 or not self.training)
+            and ops.extension_available()
+            and os.environ.get("TRLX_AMD_NO_FLASH_PREFILL") != "1"
+            and os.environ.get("TRLX_AMD_NO_FLASH_TRAIN") != "1"
+        )
+        if flash_train_ok:
+            # TRAINING forward: differentiable flash attention — forward saves
+            # the row logsumexp, backward recomputes P tile-by-tile
+            # (csrc/flash_backward.hip); [B, H, T, T] never materializes in
+            # either direction
+            out = ops.flash_attention(q, k, v, ctx.key_starts,
+                                      1.0 if pre_scaled else self.scale)
+            out = out.transpose(1, 2).reshape(B, T, -1)
+            return dense(self.o_proj, out)
+
         # prefill / training: rocBLAS batched GEMMs + fused causal softmax
         if self.num_kv_heads != self.num_heads:
             rep = self.num_heads // self.num_kv_heads

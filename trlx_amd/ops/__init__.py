@@ -704,3 +704,39 @@ def flash_prefill(q, k, v, key_starts=None, start_pos: int = 0, scale: float = 1
     ks = key_starts.to(torch.int32).contiguous() if key_starts is not None else None
     return ext.flash_prefill(q.contiguous(), k, v, ks, int(start_pos), float(scale),
                              int(tk) if tk is not None else 0)
+
+
+class _FlashAttention(torch.autograd.Function):
+    """Differentiable flash attention (training prefill, start_pos == 0,
+    Tk == T): forward saves the per-row logsumexp, backward recomputes P from
+    it tile-by-tile (csrc/flash_backward.hip) — the [B,H,T,T] scores never
+    materialize in either direction."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, key_starts, scale, ext):
+        out, lse = ext.flash_prefill_lse(q, k, v, key_starts, 0, float(scale), 0, True)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.key_starts = key_starts
+        ctx.scale = scale
+        ctx.ext = ext
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        dq, dk, dv = ctx.ext.flash_prefill_bwd(q, k, v, out, dout, lse, ctx.key_starts,
+                                               float(ctx.scale))
+        return dq, dk, dv, None, None, None
+
+
+def flash_attention(q, k, v, key_starts=None, scale: float = 1.0):
+    """Differentiable flash attention for the TRAINING forward (full causal
+    self-attention, no cache): q [B,Hq,T,D] bf16, k/v [B,Hkv,T,D].  On CPU or
+    without the extension, falls back to the fp32 reference (autograd handles
+    the backward there)."""
+    ext = _require_ext("flash_attention")
+    if ext is None or not q.is_cuda:
+        return reference.flash_prefill(q, k, v, key_starts, 0, scale).to(q.dtype)
+    ks = key_starts.to(torch.int32).contiguous() if key_starts is not None else None
+    return _FlashAttention.apply(q.contiguous(), k.contiguous(), v.contiguous(), ks,
+                                 float(scale), ext)
