@@ -422,6 +422,13 @@ constexpr int WAVE_H_MAX = 4096;
 int pick_wave_grid(long n) {
   return (int)std::min<long>((n + NWAVES - 1) / NWAVES, 2048);
 }
+bool wave_norm_enabled() {
+  static const bool on = [] {
+    const char* e = getenv("TRLX_AMD_NO_WAVE_NORM");
+    return !(e && e[0] == '1');
+  }();
+  return on;
+}
 
 }  // namespace
 
@@ -445,7 +452,7 @@ std::vector<at::Tensor> rmsnorm_fwd(const at::Tensor& x, const at::Tensor& w, do
   auto stream = c10::hip::getCurrentHIPStream();
   auto wc = w.contiguous();
   const int grid = pick_grid(N);
-  const bool wavep = H <= WAVE_H_MAX;
+  const bool wavep = H <= WAVE_H_MAX && wave_norm_enabled();
   const int wgrid = pick_wave_grid(N);
 #define LAUNCH_RMS_FWD(T, HR, XP, RP, WP, YP, SP)                                            \
   do {                                                                                       \
@@ -549,7 +556,7 @@ std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x, const at::Tensor& w,
   at::Tensor bc;
   if (has_bias) bc = b->contiguous();
   const int grid = pick_grid(N);
-  const bool wavep = H <= WAVE_H_MAX;
+  const bool wavep = H <= WAVE_H_MAX && wave_norm_enabled();
   const int wgrid = pick_wave_grid(N);
 #define LAUNCH_LN_FWD(T, HB, HR, XP, RP, WP, BP, YP, SP)                                    \
   do {                                                                                      \
