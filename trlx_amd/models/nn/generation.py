@@ -372,6 +372,11 @@ class DecodeEngine:
         self.seq_lens.fill_(T)
         self.cache_idx.fill_(T - 1)
         self.step_col.zero_()
+        if self.gen.seed is not None:
+            # explicit seed: same (seed, inputs) -> same tokens, like eager.
+            # With no seed the offset keeps advancing across calls so reused
+            # engines stay fresh (the capture bakes the random seed in).
+            self.rng_offset.zero_()
         self.finished.zero_()
         self.out_tokens.fill_(self.pad_id)
 

@@ -596,17 +596,38 @@ def sample_token(
                 return torch.argmax(logits, dim=-1)
             if seed is None:
                 seed = int(torch.randint(0, 2**31 - 1, (1,), generator=generator).item())
+            dev_offset = isinstance(offset, torch.Tensor)
             if top_p is not None and 0.0 < top_p < 1.0:
                 # top-p needs a sorted scan: do the filter with library sort,
-                # then fused gumbel sampling on the filtered logits.
+                # then fused gumbel sampling on the filtered logits.  Every op
+                # here is hipGraph-capturable (device-offset RNG included), so
+                # top-p decode stays on the captured-graph path.
                 # HF order: top-k first, then top-p (matches reference.sample_token)
+                def _gumbel(t):
+                    if dev_offset:
+                        return ext.gumbel_sample_dev(t, 1.0, None, seed, offset)
+                    return ext.gumbel_sample(t, 1.0, None, seed, offset)
+
                 lg = logits.float() / max(temperature, 1e-6)
                 if top_k and 0 < top_k < logits.shape[-1]:
-                    kth = torch.topk(lg, top_k, dim=-1).values[..., -1, None]
-                    lg = lg.masked_fill(lg < kth, float("-inf"))
-                lg = _top_p_filter(lg, top_p)
-                return ext.gumbel_sample(lg.contiguous(), 1.0, None, seed, offset)
-            dev_offset = isinstance(offset, torch.Tensor)
+                    # both filters confined to the [B, top_k] subset: softmax
+                    # over the top-k masked vocab equals softmax over the
+                    # subset, so the p-filter never touches [B, V]
+                    vals, idx = torch.topk(lg, top_k, dim=-1)  # sorted desc
+                    probs = torch.softmax(vals, dim=-1)
+                    cum = probs.cumsum(dim=-1)
+                    vals = vals.masked_fill(cum - probs > top_p, float("-inf"))
+                    pos = _gumbel(vals.contiguous())
+                    return idx.gather(-1, pos.view(-1, 1)).view(pos.shape)
+                # V-wide: per-row threshold from the sorted scan, elementwise
+                # mask — no [B, V] scatter (was a 520 us kernel at [128, 50257])
+                sorted_logits, _ = torch.sort(lg, descending=True, dim=-1)
+                probs = torch.softmax(sorted_logits, dim=-1)
+                cum = probs.cumsum(dim=-1)
+                keep = (cum - probs <= top_p).sum(-1, keepdim=True).clamp(min=1) - 1
+                thr = sorted_logits.gather(-1, keep)
+                lg = lg.masked_fill(lg < thr, float("-inf"))
+                return _gumbel(lg.contiguous())
             if top_k and 0 < top_k < logits.shape[-1]:
                 lg = (logits.float() / max(temperature, 1e-6)).contiguous()
                 thr = torch.topk(lg, top_k, dim=-1).values[:, -1].contiguous()
