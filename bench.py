@@ -129,8 +129,8 @@ def run_cycle(trainer, config, phase_times=None):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=3)
-    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--steps", type=int, default=5)
+    p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--model", type=str, default="gpt2")
     p.add_argument("--seq-len", type=int, default=1024)
     p.add_argument("--prompt-len", type=int, default=64)
