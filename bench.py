@@ -43,7 +43,7 @@ def model_config(args):
     return preset(args.model)
 
 
-def build_trainer(args):
+def build_trainer(args, reward_fn=None):
     from trlx_amd.data.default_configs import default_ppo_config
     from trlx_amd.pipeline.offline_pipeline import PromptPipeline
     from trlx_amd.utils.loading import get_trainer
@@ -75,9 +75,10 @@ def build_trainer(args):
     # DP=8 the gather->rank-0-score->scatter round trip idles 7 ranks
     config.method.local_rewards = True
 
-    def reward_fn(samples, prompts, outputs, **kwargs):
-        # cheap deterministic stand-in for the sentiment classifier
-        return [float((len(s) * 2654435761) % 1000) / 1000.0 - 0.5 for s in samples]
+    if reward_fn is None:
+        def reward_fn(samples, prompts, outputs, **kwargs):
+            # cheap deterministic stand-in for the sentiment classifier
+            return [float((len(s) * 2654435761) % 1000) / 1000.0 - 0.5 for s in samples]
 
     trainer = get_trainer(config.train.trainer)(config=config, reward_fn=reward_fn)
 
@@ -126,7 +127,7 @@ def run_cycle(trainer, config, phase_times=None):
         phase_times["train"] += t2 - t1
 
 
-def main():
+def parse_args(argv=None):
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=5)
@@ -147,7 +148,11 @@ def main():
                    help="tiny model for the torchrun CPU contract dry-run (not a benchmark)")
     p.add_argument("--no-secondary", action="store_true",
                    help="skip the secondary ILQL / model-ladder measurements")
-    args = p.parse_args()
+    return p.parse_args(argv)
+
+
+def main():
+    args = parse_args()
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     rank = int(os.environ.get("RANK", 0))

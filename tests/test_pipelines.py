@@ -171,3 +171,24 @@ def test_rollout_json_export(tmp_path):
     assert data[0]["query_tensor"] == [1, 2]
     assert data[0]["response_tensor"] == [3, 4]
     assert "logprobs" not in data[0]  # only_text=True
+
+
+def test_running_moments_matches_batch_statistics():
+    """RunningMoments over chunks converges to the full-population statistics
+    (reference tests/test_utils.py test_running_moments)."""
+    import torch
+    from trlx_amd.utils.modeling import RunningMoments
+
+    torch.manual_seed(0)
+    rm = RunningMoments()
+    all_chunks = []
+    for _ in range(10):
+        xs = torch.randn(100) * 3 + 1.5
+        all_chunks.append(xs)
+        b_mean, b_std = rm.update(xs)
+        # per-batch return values are the batch's own stats
+        assert abs(b_mean - xs.mean().item()) < 1e-4
+        assert abs(b_std - xs.std().item()) < 1e-2
+    full = torch.cat(all_chunks)
+    assert abs(rm.mean - full.mean().item()) < 1e-4
+    assert abs(rm.std - full.std().item()) < 1e-3
