@@ -370,3 +370,26 @@ def test_ppo_update_direction_raises_best_advantage_logprob(tmp_path):
             trainer.opt.zero_grad()
     after = best_logprob()
     assert after > before + 0.3, (before, after)
+
+
+def test_ppo_end_to_end_with_ref_offload(tmp_path):
+    """model.ref_offload=True (K15): PPO e2e with the frozen reference branch
+    offloaded to host memory."""
+    cfg = _tiny_model_cfg(default_ppo_config(), tmp_path)
+    cfg.model.num_layers_unfrozen = 1
+    cfg.model.ref_offload = True
+    cfg.method.num_rollouts = 8
+    cfg.method.chunk_size = 4
+    cfg.method.ppo_epochs = 1
+
+    def reward_fn(samples, prompts, outputs, **kw):
+        return [float(len(o)) for o in outputs]
+
+    trainer = trlx_amd.train(
+        reward_fn=reward_fn,
+        prompts=["hello", "world", "foo", "bar"],
+        eval_prompts=["hello"],
+        config=cfg,
+    )
+    assert trainer.model.frozen_head._offloaded
+    assert trainer.iter_count == 2

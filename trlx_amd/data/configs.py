@@ -61,6 +61,11 @@ class ModelConfig:
     # activations_checkpoint_method, megatron_20b.yaml:77-79) — trades ~30%
     # step time for activation memory on long-context runs
     gradient_checkpointing: bool = False
+    # K15: keep the hydra reference branch's block weights in pinned host
+    # memory and stream them to the GPU just in time (parity: NeMo
+    # offload_reference_model, megatron_65b.yaml:5-6).  Only worth it on
+    # 65B-class replicas; 288 GB HBM keeps <=20B references resident.
+    ref_offload: bool = False
 
     @classmethod
     def from_dict(cls, config: Dict[str, Any]) -> "ModelConfig":

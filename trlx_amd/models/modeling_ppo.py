@@ -332,8 +332,11 @@ class FrozenBranch(nn.Module):
         fused lm_logprobs path consumes these directly."""
         with torch.no_grad():
             h, res = hidden, None
-            for block in self.blocks:
-                h, res = block(h, ctx, rope_tables, res=res)
+            if self._offloaded and h.is_cuda:
+                h, res = self._forward_blocks_offloaded(h, ctx, rope_tables)
+            else:
+                for block in self.blocks:
+                    h, res = block(h, ctx, rope_tables, res=res)
             if logits_slice is not None:
                 h = h[:, logits_slice[0] : logits_slice[1]]
                 if res is not None:
@@ -341,6 +344,66 @@ class FrozenBranch(nn.Module):
             if res is None:
                 return self.final_norm(h)
             return self.final_norm.forward_add(h, res)[0]
+
+    # ---- K15: reference-weight CPU offload ---------------------------------
+    # (reference modeling_nemo_ppo.py:228-244 offload_reference_model for the
+    # 65B config, megatron_65b.yaml:5-6).  On MI355X the 288 GB HBM keeps
+    # <=20B hydras resident, so this is OPT-IN (model.ref_offload) for
+    # 65B-class replicas: block weights live in pinned host memory and stream
+    # to two GPU shadow blocks just in time, the H2D of block i+1 overlapping
+    # block i's compute on a side stream.
+
+    _offloaded = False
+
+    def offload(self):
+        """Move the frozen blocks' weights to pinned host memory; forwards
+        stream them back block-by-block (double-buffered).  final_norm and
+        lm_head stay resident (tiny next to the blocks)."""
+        if self._offloaded or len(self.blocks) == 0:
+            return self
+        dev = next(self.blocks[0].parameters()).device
+        if dev.type == "cuda":
+            # two GPU shadow blocks, structure-identical to a frozen block
+            self._shadows = [copy.deepcopy(self.blocks[0]), copy.deepcopy(self.blocks[0])]
+            self._copy_stream = torch.cuda.Stream()
+            self._ready = [torch.cuda.Event(), torch.cuda.Event()]
+            self._free = [torch.cuda.Event(), torch.cuda.Event()]
+        pin = torch.cuda.is_available()
+        for block in self.blocks:
+            for p in list(block.parameters()) + list(block.buffers()):
+                host = p.data.detach().cpu()
+                p.data = host.pin_memory() if pin else host
+        self._offloaded = True
+        return self
+
+    def _forward_blocks_offloaded(self, h, ctx, rope_tables):
+        shadows, ready, free = self._shadows, self._ready, self._free
+        cs = self._copy_stream
+        main = torch.cuda.current_stream()
+
+        def load(slot, block):
+            with torch.cuda.stream(cs):
+                cs.wait_event(free[slot])
+                for dst, src in zip(
+                    list(shadows[slot].parameters()) + list(shadows[slot].buffers()),
+                    list(block.parameters()) + list(block.buffers()),
+                ):
+                    dst.data.copy_(src.data, non_blocking=True)
+                ready[slot].record(cs)
+
+        # both shadows start free (record on main so the first waits are no-ops)
+        free[0].record(main)
+        free[1].record(main)
+        load(0, self.blocks[0])
+        res = None
+        for i, block in enumerate(self.blocks):
+            slot = i % 2
+            if i + 1 < len(self.blocks):
+                load((i + 1) % 2, self.blocks[i + 1])
+            main.wait_event(ready[slot])
+            h, res = shadows[slot](h, ctx, rope_tables, res=res)
+            free[slot].record(main)  # shadow reusable once this compute drains
+        return h, res
 
 
 class AutoModelForCausalLMWithHydraValueHead(AutoModelForCausalLMWithValueHead):

@@ -136,6 +136,11 @@ class NativeRLTrainer(BaseRLTrainer):
             process_group=self.dp_group,
         )
         self.reducer.broadcast_parameters(self.model)
+        if (getattr(self.config.model, "ref_offload", False)
+                and getattr(self.model, "frozen_head", None) is not None):
+            # K15 ref-weight CPU offload — AFTER the init broadcast (RCCL
+            # can't broadcast the pinned-host tensors this creates)
+            self.model.frozen_head.offload()
 
         script_name = os.path.basename(sys.argv[0]).rsplit(".", 1)[0]
         run_time = int(time())
