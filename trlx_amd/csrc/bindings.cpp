@@ -51,6 +51,8 @@ at::Tensor lm_logprobs_v2(const at::Tensor& hidden, const at::Tensor& weight,
                           const at::Tensor& labels);
 at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
                        const c10::optional<at::Tensor>& bias, long act);
+at::Tensor skinny_gemm_fp8(const at::Tensor& a, const at::Tensor& w8, const at::Tensor& wscale,
+                           const c10::optional<at::Tensor>& bias, long act);
 std::vector<at::Tensor> lm_logprobs_v2_with_lse(const at::Tensor& hidden,
                                                 const at::Tensor& weight,
                                                 const at::Tensor& labels, bool want_lse);
@@ -144,6 +146,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("lm_logprobs", &lm_logprobs);
   mod.def("lm_logprobs_v2", &lm_logprobs_v2);
   mod.def("skinny_gemm", &skinny_gemm);
+  mod.def("skinny_gemm_fp8", &skinny_gemm_fp8);
   mod.def("lm_logprobs_v2_with_lse", &lm_logprobs_v2_with_lse);
   mod.def("ce_dlogits", &ce_dlogits);
   mod.def("decode_advance", &decode_advance);

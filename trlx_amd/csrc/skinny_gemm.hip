@@ -154,6 +154,119 @@ __global__ void skinny_epilogue_kernel(const float* __restrict__ ws,
   ScalarIO<bf16_t>::store(C + i, sk_act(v, act));
 }
 
+// ---- fp8 (e4m3) weight-only variant ---------------------------------------
+// The decode GEMMs are pinned at an L2-access-rate wall proportional to the
+// UNIQUE bytes demanded (profiles/r02_notes.md): fp8 weights halve the weight
+// stream.  W is stored as e4m3 bytes [N, K] with one fp32 scale per output
+// row; fragments dequantize in registers to feed the same bf16 MFMA chain
+// (A stays bf16 — weight-only quantization, activations untouched), and the
+// row scale multiplies the fp32 accumulator once at the epilogue (exact).
+
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_sk;
+typedef __attribute__((ext_vector_type(2))) float f32x2_sk;
+
+// hardware v_cvt_pk_f32_fp8 (gfx940+): 2 e4m3 bytes -> 2 floats per
+// instruction; the first cut's scalar ldexpf chain made the kernel
+// ALU-bound and erased the bandwidth win
+DEV bf16x8_sk dequant8(u32x2_sk raw) {
+  bf16x8_sk r;
+#pragma unroll
+  for (int h = 0; h < 2; ++h) {
+    f32x2_sk lo = __builtin_amdgcn_cvt_pk_f32_fp8(raw[h], false);
+    f32x2_sk hi = __builtin_amdgcn_cvt_pk_f32_fp8(raw[h], true);
+    r[4 * h + 0] = (short)f2bf(lo[0]);
+    r[4 * h + 1] = (short)f2bf(lo[1]);
+    r[4 * h + 2] = (short)f2bf(hi[0]);
+    r[4 * h + 3] = (short)f2bf(hi[1]);
+  }
+  return r;
+}
+
+template <bool SPLIT>
+__global__ __launch_bounds__(SK_BLOCK) void skinny_gemm_fp8_kernel(
+    const bf16_t* __restrict__ A, const unsigned char* __restrict__ W,
+    const float* __restrict__ wscale, const bf16_t* __restrict__ bias, void* __restrict__ Cout,
+    int M, int N, int K, int act, int nsplit) {
+  const int lane = threadIdx.x % WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int nM = (M + 15) >> 4;
+  const int nN = (N + 15) >> 4;
+  int tile = blockIdx.x * 4 + wid;
+  const int ntile2 = nM * nN;
+  if (tile >= ntile2 * nsplit) return;
+  const int ks = tile / ntile2;
+  tile -= ks * ntile2;
+  const int nt = tile / nM;
+  const int mt = tile % nM;
+
+  const int arow = min(mt * 16 + (lane & 15), M - 1);
+  const int wrow = min(nt * 16 + (lane & 15), N - 1);
+  const int k8 = (lane >> 4) * 8;
+  const int kchunk = ((K / 32) / nsplit) * 32;
+  const int k0 = ks * kchunk;
+  const int k1 = (ks == nsplit - 1) ? K : k0 + kchunk;
+  const bf16_t* ap = A + (size_t)arow * K + k8;
+  const unsigned char* wp = W + (size_t)wrow * K + k8;
+
+  f32x4_sk acc = {0.f, 0.f, 0.f, 0.f};
+  int k = k0;
+  if (k + 128 <= k1) {
+    bf16x8_sk a0 = *reinterpret_cast<const bf16x8_sk*>(ap + k);
+    u32x2_sk b0 = *reinterpret_cast<const u32x2_sk*>(wp + k);
+    bf16x8_sk a1 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 32);
+    u32x2_sk b1 = *reinterpret_cast<const u32x2_sk*>(wp + k + 32);
+    bf16x8_sk a2 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 64);
+    u32x2_sk b2 = *reinterpret_cast<const u32x2_sk*>(wp + k + 64);
+    bf16x8_sk a3 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 96);
+    u32x2_sk b3 = *reinterpret_cast<const u32x2_sk*>(wp + k + 96);
+    for (k += 128; k + 128 <= k1; k += 128) {
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, dequant8(b0), acc, 0, 0, 0);
+      a0 = *reinterpret_cast<const bf16x8_sk*>(ap + k);
+      b0 = *reinterpret_cast<const u32x2_sk*>(wp + k);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, dequant8(b1), acc, 0, 0, 0);
+      a1 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 32);
+      b1 = *reinterpret_cast<const u32x2_sk*>(wp + k + 32);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, dequant8(b2), acc, 0, 0, 0);
+      a2 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 64);
+      b2 = *reinterpret_cast<const u32x2_sk*>(wp + k + 64);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, dequant8(b3), acc, 0, 0, 0);
+      a3 = *reinterpret_cast<const bf16x8_sk*>(ap + k + 96);
+      b3 = *reinterpret_cast<const u32x2_sk*>(wp + k + 96);
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, dequant8(b0), acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, dequant8(b1), acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, dequant8(b2), acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, dequant8(b3), acc, 0, 0, 0);
+  }
+  for (; k < k1; k += 32) {
+    bf16x8_sk a = *reinterpret_cast<const bf16x8_sk*>(ap + k);
+    u32x2_sk b = *reinterpret_cast<const u32x2_sk*>(wp + k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, dequant8(b), acc, 0, 0, 0);
+  }
+
+  const int ccol = nt * 16 + (lane & 15);
+  if (ccol >= N) return;
+  const float sc = wscale[ccol];
+  if (SPLIT) {
+    float* c = reinterpret_cast<float*>(Cout) + (size_t)ks * M * N;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int crow = mt * 16 + (lane >> 4) * 4 + r;
+      if (crow < M) c[(size_t)crow * N + ccol] = acc[r] * sc;
+    }
+  } else {
+    bf16_t* c = reinterpret_cast<bf16_t*>(Cout);
+    float bv = bias ? ScalarIO<bf16_t>::load(bias + ccol) : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int crow = mt * 16 + (lane >> 4) * 4 + r;
+      if (crow < M) {
+        ScalarIO<bf16_t>::store(c + (size_t)crow * N + ccol, sk_act(acc[r] * sc + bv, act));
+      }
+    }
+  }
+}
+
 }  // namespace
 
 at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
@@ -191,6 +304,50 @@ at::Tensor skinny_gemm(const at::Tensor& a, const at::Tensor& w,
     const int grid = (ntiles * nsplit + 3) / 4;
     skinny_gemm_kernel<true><<<grid, SK_BLOCK, 0, stream>>>(
         ap, wp, nullptr, ws.data_ptr(), M, N, K, (int)act, nsplit);
+    const long MN = (long)M * N;
+    skinny_epilogue_kernel<<<(int)((MN + 255) / 256), 256, 0, stream>>>(
+        ws.data_ptr<float>(), bp, reinterpret_cast<bf16_t*>(c.data_ptr()), MN, N, (int)act,
+        nsplit);
+  }
+  HIP_CHECK_LAST();
+  return c;
+}
+
+at::Tensor skinny_gemm_fp8(const at::Tensor& a, const at::Tensor& w8, const at::Tensor& wscale,
+                           const c10::optional<at::Tensor>& bias, long act) {
+  TORCH_CHECK(a.is_cuda() && a.dtype() == at::kBFloat16 && a.dim() == 2 && a.is_contiguous());
+  TORCH_CHECK(w8.dtype() == at::kByte && w8.dim() == 2 && w8.is_contiguous(),
+              "skinny_gemm_fp8: weight must be contiguous e4m3 bytes viewed as uint8");
+  const int M = a.size(0);
+  const int K = a.size(1);
+  const int N = w8.size(0);
+  TORCH_CHECK(w8.size(1) == K && K % 32 == 0, "skinny_gemm_fp8: K must be a multiple of 32");
+  TORCH_CHECK(wscale.dtype() == at::kFloat && wscale.numel() == N && wscale.is_contiguous());
+  auto c = at::empty({M, N}, a.options());
+  if (M == 0) return c;
+  const bf16_t* bp = nullptr;
+  at::Tensor bc;
+  if (bias.has_value() && bias->defined()) {
+    bc = bias->contiguous();
+    TORCH_CHECK(bc.numel() == N && bc.dtype() == at::kBFloat16);
+    bp = reinterpret_cast<const bf16_t*>(bc.data_ptr());
+  }
+  const int ntiles = ((M + 15) / 16) * ((N + 15) / 16);
+  int nsplit = 1;
+  while (nsplit < 8 && ntiles * nsplit < 4096 && (K / 32) / (nsplit * 2) >= 2) nsplit *= 2;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const auto ap = reinterpret_cast<const bf16_t*>(a.data_ptr());
+  const auto wp = w8.data_ptr<unsigned char>();
+  const auto sp = wscale.data_ptr<float>();
+  if (nsplit == 1) {
+    const int grid = (ntiles + 3) / 4;
+    skinny_gemm_fp8_kernel<false><<<grid, SK_BLOCK, 0, stream>>>(
+        ap, wp, sp, bp, c.data_ptr(), M, N, K, (int)act, 1);
+  } else {
+    auto ws = at::empty({nsplit, (long)M, (long)N}, a.options().dtype(at::kFloat));
+    const int grid = (ntiles * nsplit + 3) / 4;
+    skinny_gemm_fp8_kernel<true><<<grid, SK_BLOCK, 0, stream>>>(
+        ap, wp, sp, nullptr, ws.data_ptr(), M, N, K, (int)act, nsplit);
     const long MN = (long)M * N;
     skinny_epilogue_kernel<<<(int)((MN + 255) / 256), 256, 0, stream>>>(
         ws.data_ptr<float>(), bp, reinterpret_cast<bf16_t*>(c.data_ptr()), MN, N, (int)act,

@@ -181,12 +181,29 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Te
     K = x.shape[-1]
     M = x.numel() // K
     if (x.is_cuda and not torch.is_grad_enabled() and x.dtype == torch.bfloat16
-            and weight.dtype == torch.bfloat16 and M <= 256 and K % 32 == 0
-            and os.environ.get("TRLX_AMD_SKINNY") == "1"):
-        ext = _require_ext("skinny_gemm")
-        if ext is not None and hasattr(ext, "skinny_gemm"):
-            y = ext.skinny_gemm(x.reshape(M, K).contiguous(), weight.contiguous(), bias, act)
-            return y.view(*x.shape[:-1], weight.shape[0])
+            and weight.dtype == torch.bfloat16 and M <= 256 and K % 32 == 0):
+        if os.environ.get("TRLX_AMD_FP8_DECODE") == "1":
+            # fp8 (e4m3) WEIGHT-ONLY decode GEMM: halves the weight stream
+            # that the decode L2-access wall is proportional to.  Weight-only
+            # quantization; activations and accumulation stay bf16/fp32;
+            # per-output-row scales applied exactly on the accumulator.
+            # OPT-IN: ~0.4% relative logit noise shifts sampled trajectories.
+            ext = _require_ext("skinny_gemm_fp8")
+            if ext is not None and hasattr(ext, "skinny_gemm_fp8"):
+                q = getattr(weight, "_fp8_cache", None)
+                if q is None or q[0].shape != weight.shape:
+                    weight._fp8_cache = quantize_fp8_rows(weight)
+                    q = weight._fp8_cache
+                bb = bias.bfloat16() if bias is not None and bias.dtype != torch.bfloat16 \
+                    else bias
+                y = ext.skinny_gemm_fp8(x.reshape(M, K).contiguous(), q[0], q[1], bb, act)
+                return y.view(*x.shape[:-1], weight.shape[0])
+        if os.environ.get("TRLX_AMD_SKINNY") == "1":
+            ext = _require_ext("skinny_gemm")
+            if ext is not None and hasattr(ext, "skinny_gemm"):
+                y = ext.skinny_gemm(x.reshape(M, K).contiguous(), weight.contiguous(), bias,
+                                    act)
+                return y.view(*x.shape[:-1], weight.shape[0])
     if (act in (1, 2, 3) and bias is not None and x.is_cuda
             and not torch.is_grad_enabled() and x.dtype == torch.bfloat16
             and _addmm_act_ok()):
@@ -203,6 +220,20 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor, bias: Optional[torch.Te
             _ADDMM_ACT = False
     y = torch.nn.functional.linear(x, weight, bias)
     return _ACT_FNS[act](y)
+
+
+def quantize_fp8_rows(w: torch.Tensor):
+    """Per-output-row e4m3 weight quantization for the fp8 decode GEMM:
+    returns (bytes [N, K] uint8, scales [N] fp32) with scale = amax/448."""
+    wf = w.detach().float()
+    s = (wf.abs().amax(dim=1) / 448.0).clamp(min=1e-12)
+    q = (wf / s[:, None]).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).contiguous(), s.contiguous()
+
+
+def dequantize_fp8_rows(q8: torch.Tensor, s: torch.Tensor) -> torch.Tensor:
+    """Reference inverse of quantize_fp8_rows (fp32)."""
+    return q8.view(torch.float8_e4m3fn).float() * s[:, None]
 
 
 _ADDMM_ACT: Optional[bool] = None
