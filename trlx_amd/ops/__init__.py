@@ -236,6 +236,24 @@ def dequantize_fp8_rows(q8: torch.Tensor, s: torch.Tensor) -> torch.Tensor:
     return q8.view(torch.float8_e4m3fn).float() * s[:, None]
 
 
+def refresh_fp8_caches(module: "torch.nn.Module") -> int:
+    """Requantize every cached fp8 decode weight IN PLACE (same buffers, so
+    captured decode graphs see the fresh values).  Needed between training
+    steps and generation: FusedAdamW writes the parameter arenas from a
+    kernel, which torch's version counter never sees, so the caches cannot
+    self-invalidate.  Returns the number of refreshed weights."""
+    n = 0
+    for m in module.modules():
+        w = getattr(m, "weight", None)
+        q = getattr(w, "_fp8_cache", None) if w is not None else None
+        if q is not None:
+            q8, s = quantize_fp8_rows(w)
+            q[0].copy_(q8)
+            q[1].copy_(s)
+            n += 1
+    return n
+
+
 _ADDMM_ACT: Optional[bool] = None
 
 
