@@ -539,3 +539,31 @@ def test_frozen_branch_offload_gpu_streams_and_matches():
     with torch.no_grad():
         out2 = m.forward_hydra(ids, attention_mask=mask).logits.float()
     torch.testing.assert_close(out2, ref, atol=0, rtol=0)
+
+
+@pytest.mark.gpu
+def test_gpu_decode_without_cache_idx_matches_full_forward():
+    """PP generation drives decode steps through AttentionContext(seq_lens=..)
+    WITHOUT a device cache_idx; on GPU that path must match the full forward
+    (it double-applied the attention scale on top of qkv_prep's pre-scaled q
+    before this test existed)."""
+    from trlx_amd.models.nn.transformer import AttentionContext
+
+    torch.manual_seed(0)
+    cfg = tiny_config(vocab_size=400, hidden_size=128, num_layers=2, num_heads=2)
+    m = CausalTransformer(cfg).cuda().bfloat16().eval()
+    B, T = 2, 9
+    ids = torch.randint(3, 400, (B, T), device="cuda")
+    mask = torch.ones_like(ids)
+    with torch.no_grad():
+        full = m(ids, attention_mask=mask).logits.float()
+        kv = m.new_kv_cache(B, T + 2, device="cuda")
+        m(ids[:, :-1], attention_mask=mask[:, :-1], kv_cache=kv, start_pos=0)
+        key_starts = torch.zeros(B, dtype=torch.int32, device="cuda")
+        pos = torch.full((B, 1), T - 1, dtype=torch.int32, device="cuda")
+        seq_lens = torch.full((B,), T, dtype=torch.int32, device="cuda")
+        # the PP-style call: seq_lens set, cache_idx absent
+        out = m(ids[:, -1:], kv_cache=kv, start_pos=T - 1, position_ids=pos,
+                seq_lens=seq_lens, key_starts=key_starts)
+        step = out.logits[:, -1].float()
+    torch.testing.assert_close(step, full[:, -1], atol=6e-2, rtol=6e-2)

@@ -85,6 +85,9 @@ class AttentionContext:
     # device scalar cache write index: enables the fused decode_prep kernel
     # and hipGraph capture (no host-side position state)
     cache_idx: Optional[torch.Tensor] = None
+    # PER-ROW cache write positions [B] (continuous-batching decode: slots
+    # sit at different depths); mutually exclusive with cache_idx
+    cache_rows: Optional[torch.Tensor] = None
     # ALiBi additive bias [B, H_local, 1, T_keys] (Bloom)
     alibi: Optional[torch.Tensor] = None
     # PREFIX_TUNING: per-layer K/V params + the cache positions they occupy
@@ -115,6 +118,15 @@ class KVCache:
         assert start + T <= self.max_len, "KV cache overflow"
         self.k[layer][:, :, start : start + T] = k
         self.v[layer][:, :, start : start + T] = v
+        return self.k[layer], self.v[layer]
+
+    def update_rows(self, layer: int, k: torch.Tensor, v: torch.Tensor,
+                    rows: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Single-token append at PER-ROW positions (continuous batching:
+        each slot decodes at its own depth).  k/v [B, Hkv, 1, D], rows [B]."""
+        b = torch.arange(k.shape[0], device=k.device)
+        self.k[layer][b, :, rows.long()] = k[:, :, 0]
+        self.v[layer][b, :, rows.long()] = v[:, :, 0]
         return self.k[layer], self.v[layer]
 
 
@@ -221,11 +233,17 @@ class Attention(nn.Module):
             if pre_scaled:
                 # cache stores unscaled k (decode kernel scales q itself)
                 pass
-            k_full, v_full = kv_cache.update(self.layer_idx, k, v, ctx.start_pos)
+            if ctx.cache_rows is not None:
+                k_full, v_full = kv_cache.update_rows(self.layer_idx, k, v, ctx.cache_rows)
+            else:
+                k_full, v_full = kv_cache.update(self.layer_idx, k, v, ctx.start_pos)
             if (T == 1 and ctx.seq_lens is not None
                     and (not x.is_cuda or self.head_dim in (32, 64, 128, 256))):
-                # fused flash-decode kernel
-                out = ops.attention_decode(q, k_full, v_full, ctx.seq_lens, self.scale,
+                # fused flash-decode kernel (qkv_prep may already have scaled
+                # q: passing self.scale again double-scaled the PP/direct
+                # decode path on GPU — caught by the ctx-level decode test)
+                out = ops.attention_decode(q, k_full, v_full, ctx.seq_lens,
+                                           1.0 if pre_scaled else self.scale,
                                            seq_starts=ctx.key_starts)
                 out = out.transpose(1, 2).reshape(B, T, -1)
                 return dense(self.o_proj, out)
@@ -563,6 +581,7 @@ class CausalTransformer(nn.Module):
         hidden_at_layer: Optional[int] = None,
         return_logits: bool = True,
         cache_idx: Optional[torch.Tensor] = None,
+        cache_rows: Optional[torch.Tensor] = None,
         logits_slice: Optional[Tuple[int, int]] = None,
     ) -> TransformerOutput:
         """hidden_at_layer=k stashes the hidden state FED INTO layer k
@@ -587,6 +606,7 @@ class CausalTransformer(nn.Module):
             ctx = self.make_context(input_ids, attention_mask, start_pos, seq_lens, key_starts,
                                     position_ids)
             ctx.cache_idx = cache_idx
+            ctx.cache_rows = cache_rows
             h = self.embed_tokens(input_ids)
         if self.embed_norm is not None:
             h = self.embed_norm(h)
