@@ -41,3 +41,26 @@ def test_serve_loads_saved_checkpoint(tmp_path):
     app = create_app(model, tok, device=torch.device("cpu"))
     r = TestClient(app).post("/generate", json={"prompts": ["ab"], "max_new_tokens": 2})
     assert r.status_code == 200 and len(r.json()["completions"]) == 1
+
+
+def test_serve_continuous_batching_mode():
+    from fastapi.testclient import TestClient
+
+    from trlx_amd.models.nn.transformer import CausalTransformer
+    from trlx_amd.serve import create_app
+    from trlx_amd.utils.tokenizer import get_tokenizer
+
+    import torch
+    from conftest import tiny_config
+
+    torch.manual_seed(0)
+    model = CausalTransformer(tiny_config(vocab_size=300)).eval()
+    tok = get_tokenizer("byte")
+    app = create_app(model, tok, continuous_slots=2, cache_len=128)
+    with TestClient(app) as client:
+        r = client.post("/generate", json={"prompts": ["hello", "hi there", "x"],
+                                           "max_new_tokens": 5})
+        assert r.status_code == 200
+        outs = r.json()["completions"]
+        assert len(outs) == 3
+    app.state.batcher.close()

@@ -38,8 +38,13 @@ class GenerateRequest(BaseModel):
     seed: Optional[int] = None
 
 
-def create_app(model, tokenizer, device=None) -> FastAPI:
-    """Wrap a loaded model (wrapper or bare CausalTransformer) + tokenizer."""
+def create_app(model, tokenizer, device=None, continuous_slots: int = 0,
+               cache_len: int = 2048) -> FastAPI:
+    """Wrap a loaded model (wrapper or bare CausalTransformer) + tokenizer.
+
+    ``continuous_slots > 0`` serves through the continuous batcher
+    (trlx_amd/serving.py): concurrent requests share one slot pool and
+    admit/retire per token step instead of padding into one fixed batch."""
     app = FastAPI()
     base = model.base_model if hasattr(model, "base_model") else model
     if device is None:
@@ -53,8 +58,25 @@ def create_app(model, tokenizer, device=None) -> FastAPI:
     def health():
         return {"ok": True, "model": name, "device": str(device)}
 
+    batcher = None
+    if continuous_slots > 0:
+        from .serving import ContinuousBatcher
+
+        batcher = ContinuousBatcher(base, slots=continuous_slots, cache_len=cache_len,
+                                    gen=GenerateConfig(
+                                        do_sample=True,
+                                        eos_token_id=tokenizer.eos_token_id)).start()
+        app.state.batcher = batcher
+
     @app.post("/generate")
     def gen(req: GenerateRequest):
+        if batcher is not None:
+            # unpadded per-prompt admission; futures resolve as slots finish
+            futs = [batcher.submit(tokenizer(p, return_tensors="pt")["input_ids"][0],
+                                   max_new_tokens=req.max_new_tokens)
+                    for p in req.prompts]
+            return {"completions": [
+                tokenizer.decode(f.result(), skip_special_tokens=True) for f in futs]}
         tokenizer.padding_side = "left"
         enc = tokenizer(req.prompts, padding=True, return_tensors="pt")
         ids = enc["input_ids"].to(device)
@@ -90,9 +112,14 @@ def main():
     p.add_argument("--tokenizer", default=None)
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8720)
+    p.add_argument("--slots", type=int, default=0,
+                   help="enable continuous batching with this many KV slots")
+    p.add_argument("--cache-len", type=int, default=2048)
     args = p.parse_args()
     model, tok = load(args.model, args.tokenizer)
-    uvicorn.run(create_app(model, tok), host=args.host, port=args.port)
+    uvicorn.run(create_app(model, tok, continuous_slots=args.slots,
+                           cache_len=args.cache_len),
+                host=args.host, port=args.port)
 
 
 if __name__ == "__main__":
