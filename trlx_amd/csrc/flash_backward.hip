@@ -63,11 +63,15 @@ DEV void stage_tileT(const bf16_t* __restrict__ src, long row0, long nrows,
 }
 
 // per-lane: load this wave's 16-row A-operand fragments (row = lane&15,
-// k-slice = (lane>>4)*8) for the full head dim into registers
+// k-slice = (lane>>4)*8) for the full head dim into registers.  rowmax is
+// the ABSOLUTE last valid row (inclusive): a wave whose 16-row group starts
+// past the end must clamp globally, not relatively — a relative clamp read
+// up to 60 rows past the tensor and page-faulted the llama (D=128) bench
+// when the allocation ended at the tensor.
 template <int D>
-DEV void load_afrag(const bf16_t* __restrict__ src, long row0, long nrows, int lane,
+DEV void load_afrag(const bf16_t* __restrict__ src, long row0, long rowmax, int lane,
                     bf16x8_fb* frag, float mul) {
-  const long row = row0 + min((long)(lane & 15), nrows - 1);
+  const long row = min(row0 + (long)(lane & 15), rowmax);
   const int g8 = (lane >> 4) * 8;
 #pragma unroll
   for (int dch = 0; dch < D / 32; ++dch) {
@@ -117,10 +121,10 @@ __global__ __launch_bounds__(BBLOCK, 2) void flash_bwd_dkv_kernel(
 
   // K/V A-fragments for this wave's 16 keys, resident for the whole kernel
   bf16x8_fb k_frag[D / 32], v_frag[D / 32];
-  load_afrag<D>(k + ((size_t)b * Hkv + hkv) * (size_t)T * D, kt + krow0,
-                (long)T - (kt + krow0) > 0 ? (long)T - (kt + krow0) : 1, lane, k_frag, 1.f);
-  load_afrag<D>(v + ((size_t)b * Hkv + hkv) * (size_t)T * D, kt + krow0,
-                (long)T - (kt + krow0) > 0 ? (long)T - (kt + krow0) : 1, lane, v_frag, 1.f);
+  load_afrag<D>(k + ((size_t)b * Hkv + hkv) * (size_t)T * D, kt + krow0, (long)T - 1, lane,
+                k_frag, 1.f);
+  load_afrag<D>(v + ((size_t)b * Hkv + hkv) * (size_t)T * D, kt + krow0, (long)T - 1, lane,
+                v_frag, 1.f);
 
   f32x4_fb dv_acc[D / 16], dk_acc[D / 16];
 #pragma unroll
@@ -241,10 +245,10 @@ __global__ __launch_bounds__(BBLOCK, 2) void flash_bwd_dq_kernel(
 
   // scaled-Q / dO A-fragments for this wave's 16 queries, plus their L / Drow
   bf16x8_fb q_frag[D / 32], do_frag[D / 32];
-  const long nq = (long)T - (qt + qrow0) > 0 ? (long)T - (qt + qrow0) : 1;
-  load_afrag<D>(q + ((size_t)b * Hq + h) * (size_t)T * D, qt + qrow0, nq, lane, q_frag, scale);
-  load_afrag<D>(dout + ((size_t)b * Hq + h) * (size_t)T * D, qt + qrow0, nq, lane, do_frag,
-                1.f);
+  load_afrag<D>(q + ((size_t)b * Hq + h) * (size_t)T * D, qt + qrow0, (long)T - 1, lane,
+                q_frag, scale);
+  load_afrag<D>(dout + ((size_t)b * Hq + h) * (size_t)T * D, qt + qrow0, (long)T - 1, lane,
+                do_frag, 1.f);
   float L_r[4], D_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
