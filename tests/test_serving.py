@@ -107,3 +107,34 @@ def test_gpu_continuous_batching_matches_single_generation():
     got = [f.result(timeout=10) for f in futs]
     agree = sum(int(g == w) for g, w in zip(got, want))
     assert agree >= 3, (got, want)  # bf16 near-ties may flip one trajectory
+
+
+@pytest.mark.gpu
+def test_gpu_continuous_batching_graph_vs_eager():
+    """The hipGraph-captured step must produce the same greedy tokens as the
+    eager step path (TRLX_AMD_NO_GRAPHS=1)."""
+    import os
+
+    torch.manual_seed(0)
+    m = CausalTransformer(tiny_config(vocab_size=300, hidden_size=64, num_layers=2,
+                                      num_heads=2)).cuda().bfloat16().eval()
+    torch.manual_seed(1)
+    prompts = [torch.randint(3, 300, (t,)) for t in (5, 9, 3, 7, 4)]
+
+    def run():
+        cb = ContinuousBatcher(m, slots=2, cache_len=64,
+                               gen=GenerateConfig(do_sample=False, eos_token_id=None))
+        futs = [cb.submit(p, max_new_tokens=6) for p in prompts]
+        cb.run_until_idle()
+        return [f.result(timeout=10) for f in futs], cb
+
+    graph_out, cb = run()
+    assert cb.graph is not None, "graph step was not captured"
+    os.environ["TRLX_AMD_NO_GRAPHS"] = "1"
+    try:
+        eager_out, cb2 = run()
+        assert cb2.graph is None
+    finally:
+        del os.environ["TRLX_AMD_NO_GRAPHS"]
+    agree = sum(int(g == e) for g, e in zip(graph_out, eager_out))
+    assert agree >= 4, (graph_out, eager_out)

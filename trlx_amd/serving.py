@@ -23,6 +23,7 @@ graph capture of the steady-state step is an optimization left open.
 """
 
 import collections
+import os
 import threading
 from concurrent.futures import Future
 from typing import List, Optional
@@ -30,7 +31,7 @@ from typing import List, Optional
 import torch
 
 from . import ops
-from .models.nn.generation import GenerateConfig
+from .models.nn.generation import GenerateConfig, _ENGINES
 
 
 class _SlotView:
@@ -86,6 +87,20 @@ class ContinuousBatcher:
         self.seed = self.gen.seed if self.gen.seed is not None else 0
         self._thread = None
         self._stop = False
+        # graph-captured step state (GPU): the whole per-token step — forward
+        # + sampling + per-slot position/row/length advance — replays as ONE
+        # hipGraph; admission rewrites these persistent tensors between
+        # replays (same discipline as DecodeEngine)
+        self.pos = torch.ones(slots, 1, dtype=torch.int32, device=self.device)
+        self.rows = torch.zeros(slots, dtype=torch.long, device=self.device)
+        self.lens = torch.ones(slots, dtype=torch.int32, device=self.device)
+        self.active_i = torch.zeros(slots, dtype=torch.int32, device=self.device)
+        self.ks = torch.zeros(slots, dtype=torch.int32, device=self.device)
+        self.rng_offset = torch.zeros(1, dtype=torch.long, device=self.device)
+        self.graph = None
+        self._use_graph = (self.device.type == "cuda" and ops.extension_available()
+                           and os.environ.get("TRLX_AMD_NO_GRAPHS") != "1")
+        _ENGINES.add(self)  # trlx_amd.release_graphs() tears the step graph down
 
     # ---- client API -------------------------------------------------------
 
@@ -182,28 +197,81 @@ class ContinuousBatcher:
             self._step(active)
             return True
 
-    def _step(self, active: List[int]):
-        # per-slot depths: token at position seq_len, cache row seq_len,
-        # attention over rows [0, seq_len]
-        pos = torch.ones(self.B, dtype=torch.int32, device=self.device)
-        rows = torch.zeros(self.B, dtype=torch.long, device=self.device)
-        lens = torch.ones(self.B, dtype=torch.int32, device=self.device)
+    def _sync_slot_state(self, active: List[int]):
+        """Host -> persistent device tensors (between graph replays)."""
+        pos = torch.ones(self.B, dtype=torch.int32)
+        rows = torch.zeros(self.B, dtype=torch.long)
+        lens = torch.ones(self.B, dtype=torch.int32)
+        act = torch.zeros(self.B, dtype=torch.int32)
         for i in active:
             pos[i] = self.slots[i].seq_len
             rows[i] = self.slots[i].seq_len
             lens[i] = self.slots[i].seq_len + 1
-        with torch.no_grad():
-            out = self.model(self.cur_tok, kv_cache=self.kv, start_pos=0,
-                             position_ids=pos.unsqueeze(1), seq_lens=lens,
-                             key_starts=torch.zeros(self.B, dtype=torch.int32,
-                                                    device=self.device),
-                             cache_rows=rows, return_logits=False)
-            logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
-        toks = self._sample(logits)
+            act[i] = 1
+        self.pos.copy_(pos.unsqueeze(1))
+        self.rows.copy_(rows)
+        self.lens.copy_(lens)
+        self.active_i.copy_(act)
+
+    def _step_body(self):
+        """One decode step over all slots, expressed entirely in device ops
+        (hipGraph-capturable): forward + sample + per-slot state advance."""
+        out = self.model(self.cur_tok, kv_cache=self.kv, start_pos=0,
+                         position_ids=self.pos, seq_lens=self.lens,
+                         key_starts=self.ks, cache_rows=self.rows,
+                         return_logits=False)
+        logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
+        if self.gen.do_sample:
+            toks = ops.sample_token(logits, self.gen.temperature, self.gen.top_k,
+                                    self.gen.top_p, seed=self.seed,
+                                    offset=self.rng_offset)
+        else:
+            toks = logits.argmax(dim=-1)
+        self.cur_tok.copy_(toks.unsqueeze(1))
+        adv = self.active_i
+        self.pos.add_(adv.unsqueeze(1))
+        self.rows.add_(adv.long())
+        self.lens.add_(adv)
+        self.rng_offset.add_(1)
+
+    def _capture(self):
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                with torch.no_grad():
+                    self._step_body()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph), torch.no_grad():
+            self._step_body()
+
+    def _step(self, active: List[int]):
+        # per-slot depths: token at position seq_len, cache row seq_len,
+        # attention over rows [0, seq_len]
+        self._sync_slot_state(active)
+        if self._use_graph:
+            if self.graph is None:
+                # warmup/capture run the step 3x for real state: they write
+                # FUTURE cache rows (overwritten before ever read) and
+                # clobber cur_tok/pos — restore both, then replay this step
+                # for real
+                self._capture()
+                self._sync_slot_state(active)
+                cur = torch.zeros(self.B, dtype=torch.long)
+                for i in active:
+                    cur[i] = self.slots[i].tokens[-1]
+                self.cur_tok.copy_(cur.unsqueeze(1))
+            self.graph.replay()
+        else:
+            with torch.no_grad():
+                self._step_body()
+        toks = self.cur_tok[:, 0].tolist()
         for i in active:
             slot = self.slots[i]
             slot.tokens.append(int(toks[i]))
             slot.seq_len += 1
             slot.remaining -= 1
-            self.cur_tok[i, 0] = toks[i]
             self._maybe_finish(i)
