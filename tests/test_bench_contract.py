@@ -24,7 +24,7 @@ def _last_json_line(stdout: str) -> dict:
     return json.loads(lines[-1])
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_torchrun_bench_contract(world, tmp_path):
     env = dict(os.environ)
     env["HIP_VISIBLE_DEVICES"] = ""
