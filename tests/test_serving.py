@@ -138,3 +138,22 @@ def test_gpu_continuous_batching_graph_vs_eager():
         del os.environ["TRLX_AMD_NO_GRAPHS"]
     agree = sum(int(g == e) for g, e in zip(graph_out, eager_out))
     assert agree >= 4, (graph_out, eager_out)
+
+
+def test_continuous_batching_per_request_sampling():
+    """Greedy and sampled requests share the same batched step; the greedy
+    ones still match single-request generation exactly."""
+    m = _model()
+    torch.manual_seed(4)
+    p_greedy = torch.randint(3, 300, (6,))
+    p_sampled = torch.randint(3, 300, (5,))
+    want = generate(m, p_greedy.unsqueeze(0), gen=GenerateConfig(
+        max_new_tokens=5, do_sample=False, eos_token_id=None))[0, 6:].tolist()
+    cb = ContinuousBatcher(m, slots=2, cache_len=64,
+                           gen=GenerateConfig(do_sample=True, temperature=1.0,
+                                              eos_token_id=None, seed=3))
+    fg = cb.submit(p_greedy, max_new_tokens=5, do_sample=False)
+    fs = cb.submit(p_sampled, max_new_tokens=5, temperature=0.7)
+    cb.run_until_idle()
+    assert fg.result(timeout=5) == want
+    assert len(fs.result(timeout=5)) == 5

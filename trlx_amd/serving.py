@@ -52,7 +52,8 @@ class _SlotView:
 
 
 class _Slot:
-    __slots__ = ("active", "future", "tokens", "seq_len", "remaining")
+    __slots__ = ("active", "future", "tokens", "seq_len", "remaining", "temperature",
+                 "greedy")
 
     def __init__(self):
         self.active = False
@@ -60,6 +61,8 @@ class _Slot:
         self.tokens = []
         self.seq_len = 1
         self.remaining = 0
+        self.temperature = 1.0
+        self.greedy = True
 
 
 class ContinuousBatcher:
@@ -97,6 +100,10 @@ class ContinuousBatcher:
         self.active_i = torch.zeros(slots, dtype=torch.int32, device=self.device)
         self.ks = torch.zeros(slots, dtype=torch.int32, device=self.device)
         self.rng_offset = torch.zeros(1, dtype=torch.long, device=self.device)
+        # per-request sampling state (per-row temperature + greedy mask stay
+        # graph-safe: one elementwise scale + one where-select)
+        self.inv_temp = torch.ones(slots, 1, device=self.device)
+        self.greedy_m = torch.ones(slots, 1, dtype=torch.bool, device=self.device)
         self.graph = None
         self._use_graph = (self.device.type == "cuda" and ops.extension_available()
                            and os.environ.get("TRLX_AMD_NO_GRAPHS") != "1")
@@ -104,15 +111,21 @@ class ContinuousBatcher:
 
     # ---- client API -------------------------------------------------------
 
-    def submit(self, input_ids, max_new_tokens: int = 40) -> Future:
+    def submit(self, input_ids, max_new_tokens: int = 40,
+               temperature: Optional[float] = None,
+               do_sample: Optional[bool] = None) -> Future:
+        """``temperature``/``do_sample`` default to the batcher's
+        GenerateConfig; per-request overrides ride the same batched step."""
         ids = torch.as_tensor(input_ids, dtype=torch.long).view(-1)
         if ids.numel() + max_new_tokens > self.cache_len:
             raise ValueError(
                 f"prompt ({ids.numel()}) + max_new_tokens ({max_new_tokens}) exceeds "
                 f"the slot cache length {self.cache_len}")
+        temp = self.gen.temperature if temperature is None else float(temperature)
+        sample = self.gen.do_sample if do_sample is None else bool(do_sample)
         fut = Future()
         with self.lock:
-            self.pending.append((ids, max_new_tokens, fut))
+            self.pending.append((ids, max_new_tokens, temp, sample, fut))
         if self._thread is not None:
             self._wake.set()
         return fut
@@ -150,7 +163,8 @@ class ContinuousBatcher:
                 return i
         return None
 
-    def _admit(self, b: int, ids: torch.Tensor, max_new: int, fut: Future):
+    def _admit(self, b: int, ids: torch.Tensor, max_new: int, temp: float, sample: bool,
+               fut: Future):
         slot = self.slots[b]
         ids = ids.to(self.device).unsqueeze(0)
         T = ids.shape[1]
@@ -159,7 +173,17 @@ class ContinuousBatcher:
                              kv_cache=_SlotView(self.kv, b), start_pos=0,
                              return_logits=False)
             logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
-        tok = self._sample(logits)[0]
+        slot.temperature = temp
+        slot.greedy = (not sample) or temp == 0.0
+        self.inv_temp[b, 0] = 1.0 if slot.greedy else 1.0 / max(temp, 1e-6)
+        self.greedy_m[b, 0] = slot.greedy
+        if slot.greedy:
+            tok = logits.argmax(dim=-1)[0]
+        else:
+            self.step_count += 1
+            tok = ops.sample_token(logits / max(temp, 1e-6), 1.0, self.gen.top_k,
+                                   self.gen.top_p, seed=self.seed,
+                                   offset=self.step_count)[0]
         slot.active = True
         slot.future = fut
         slot.tokens = [int(tok)]
@@ -189,8 +213,8 @@ class ContinuousBatcher:
         remains (active slots or queued requests)."""
         with self.lock:
             while self.pending and (b := self._free_slot()) is not None:
-                ids, max_new, fut = self.pending.popleft()
-                self._admit(b, ids, max_new, fut)
+                ids, max_new, temp, sample, fut = self.pending.popleft()
+                self._admit(b, ids, max_new, temp, sample, fut)
             active = [i for i, s in enumerate(self.slots) if s.active]
             if not active:
                 return bool(self.pending)
@@ -221,12 +245,12 @@ class ContinuousBatcher:
                          key_starts=self.ks, cache_rows=self.rows,
                          return_logits=False)
         logits = self.model.lm_head(out.last_hidden_state[:, -1:, :])[:, 0].float()
-        if self.gen.do_sample:
-            toks = ops.sample_token(logits, self.gen.temperature, self.gen.top_k,
-                                    self.gen.top_p, seed=self.seed,
-                                    offset=self.rng_offset)
-        else:
-            toks = logits.argmax(dim=-1)
+        # per-request sampling: row temperature folds into the logits, the
+        # greedy mask selects argmax rows — both graph-safe
+        sampled = ops.sample_token(logits * self.inv_temp, 1.0, self.gen.top_k,
+                                   self.gen.top_p, seed=self.seed,
+                                   offset=self.rng_offset)
+        toks = torch.where(self.greedy_m[:, 0], logits.argmax(dim=-1), sampled)
         self.cur_tok.copy_(toks.unsqueeze(1))
         adv = self.active_i
         self.pos.add_(adv.unsqueeze(1))
