@@ -73,7 +73,9 @@ def create_app(model, tokenizer, device=None, continuous_slots: int = 0,
         if batcher is not None:
             # unpadded per-prompt admission; futures resolve as slots finish
             futs = [batcher.submit(tokenizer(p, return_tensors="pt")["input_ids"][0],
-                                   max_new_tokens=req.max_new_tokens)
+                                   max_new_tokens=req.max_new_tokens,
+                                   temperature=req.temperature,
+                                   do_sample=req.do_sample)
                     for p in req.prompts]
             return {"completions": [
                 tokenizer.decode(f.result(), skip_special_tokens=True) for f in futs]}
