@@ -284,3 +284,63 @@ def test_grounded_program_synthesis_run(tmp_path, monkeypatch):
         "method.gen_kwargs": dict(max_new_tokens=4, top_k=0, top_p=1.0, do_sample=True),
     }
     mod.main(overrides)
+
+
+def test_alpaca_sft_example(tmp_path):
+    import os
+    import subprocess
+    import sys
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "examples/alpaca/sft_alpaca.py",
+         "--hparams", '{"train.total_steps": 2, "train.eval_interval": 100, '
+                      '"train.checkpoint_interval": 1000000000, '
+                      '"train.checkpoint_dir": "%s"}' % tmp_path],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+
+
+def test_t5_translation_and_ilql_examples(tmp_path):
+    import os
+    import subprocess
+    import sys
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for script, hp in [
+        ("examples/ppo_translation_t5.py",
+         '{"train.total_steps": 2, "method.num_rollouts": 8, "method.chunk_size": 8, '
+         '"train.eval_interval": 100, "train.checkpoint_interval": 1000000000, '
+         '"train.checkpoint_dir": "%s"}' % tmp_path),
+        ("examples/ilql_sentiments_t5.py",
+         '{"train.total_steps": 2, "train.eval_interval": 100, '
+         '"train.checkpoint_interval": 1000000000, '
+         '"train.checkpoint_dir": "%s"}' % tmp_path),
+    ]:
+        r = subprocess.run([sys.executable, script, hp], cwd=REPO, env=dict(os.environ),
+                           capture_output=True, text=True, timeout=600)
+        assert r.returncode == 0, (script, r.stderr[-2000:])
+
+
+def test_inference_checkpoint_example(tmp_path):
+    import os
+    import subprocess
+    import sys
+
+    import torch
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    from conftest import tiny_config
+    from trlx_amd.models.modeling_ppo import AutoModelForCausalLMWithValueHead
+
+    torch.manual_seed(0)
+    m = AutoModelForCausalLMWithValueHead.from_config(tiny_config())
+    m.save_pretrained(str(tmp_path / "ckpt"))
+    r = subprocess.run(
+        [sys.executable, "examples/inference_checkpoint.py", "--model",
+         str(tmp_path / "ckpt"), "--tokenizer", "byte", "--max-new-tokens", "6",
+         "--greedy", "hello world"],
+        cwd=REPO, env=dict(os.environ), capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "=== 'hello world'" in r.stdout
