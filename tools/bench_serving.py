@@ -9,7 +9,10 @@ from trlx_amd.models.nn.transformer import CausalTransformer
 from trlx_amd.serving import ContinuousBatcher
 
 torch.manual_seed(0)
-m = CausalTransformer(preset("gpt2")).cuda().bfloat16().eval()
+import sys as _sys
+_model_name = _sys.argv[1] if len(_sys.argv) > 1 else "gpt2"
+m = CausalTransformer(preset(_model_name)).cuda().bfloat16().eval()
+print("model:", _model_name)
 N, MAXNEW = 32, 40
 prompts = [torch.randint(3, 50257, (int(t),)) for t in torch.randint(16, 64, (N,))]
 gen = GenerateConfig(do_sample=True, temperature=1.0, eos_token_id=None, seed=7)
