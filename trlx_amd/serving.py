@@ -210,16 +210,24 @@ class ContinuousBatcher:
 
     def _pump(self) -> bool:
         """Admit what fits, then one decode step.  Returns True if work
-        remains (active slots or queued requests)."""
-        with self.lock:
-            while self.pending and (b := self._free_slot()) is not None:
+        remains (active slots or queued requests).
+
+        The lock guards only the pending queue: every slot mutation happens
+        in the single pump thread, so the (long) model forwards run outside
+        the lock and submit() never waits on a decode step."""
+        while True:
+            with self.lock:
+                b = self._free_slot()
+                if not self.pending or b is None:
+                    break
                 ids, max_new, temp, sample, fut = self.pending.popleft()
-                self._admit(b, ids, max_new, temp, sample, fut)
-            active = [i for i, s in enumerate(self.slots) if s.active]
-            if not active:
+            self._admit(b, ids, max_new, temp, sample, fut)
+        active = [i for i, s in enumerate(self.slots) if s.active]
+        if not active:
+            with self.lock:
                 return bool(self.pending)
-            self._step(active)
-            return True
+        self._step(active)
+        return True
 
     def _sync_slot_state(self, active: List[int]):
         """Host -> persistent device tensors (between graph replays)."""
