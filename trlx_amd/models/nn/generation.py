@@ -75,6 +75,16 @@ class GenerateConfig:
         kwargs = dict(kwargs)
         kwargs.pop("max_length", None)
         known = {f for f in cls.__dataclass_fields__}
+        unknown = [k for k in kwargs if k not in known]
+        if unknown:
+            # HF generate kwargs we do not implement (num_beams,
+            # repetition_penalty, ...) must not be dropped silently: the
+            # run would quietly sample a different distribution
+            import warnings
+
+            warnings.warn(
+                f"gen_kwargs not supported by the native generate and IGNORED: {unknown} "
+                f"(supported: {sorted(known)})", stacklevel=3)
         return cls(**{k: v for k, v in kwargs.items() if k in known})
 
     def sample_key(self):
