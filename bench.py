@@ -234,6 +234,8 @@ def main():
             "phases_ms_per_step": {k: round(1000 * v / args.steps, 1)
                                    for k, v in phase_times.items()},
         }
+        if torch.cuda.is_available():
+            result["peak_mem_gb"] = round(torch.cuda.max_memory_allocated() / 2**30, 2)
     else:
         result = None
 
