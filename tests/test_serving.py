@@ -157,3 +157,27 @@ def test_continuous_batching_per_request_sampling():
     cb.run_until_idle()
     assert fg.result(timeout=5) == want
     assert len(fs.result(timeout=5)) == 5
+
+
+def test_continuous_batching_randomized_stress_matches_generation():
+    """Randomized admission stress: many requests with random lengths and
+    budgets through few slots, every greedy result exactly equal to its
+    single-request generation."""
+    m = _model()
+    rng = torch.Generator().manual_seed(9)
+    prompts, budgets = [], []
+    for _ in range(12):
+        t = int(torch.randint(2, 12, (1,), generator=rng))
+        prompts.append(torch.randint(3, 300, (t,), generator=rng))
+        budgets.append(int(torch.randint(1, 8, (1,), generator=rng)))
+    want = []
+    for p, mn in zip(prompts, budgets):
+        out = generate(m, p.unsqueeze(0), gen=GenerateConfig(
+            max_new_tokens=mn, do_sample=False, eos_token_id=None))
+        want.append(out[0, p.numel():].tolist())
+    cb = ContinuousBatcher(m, slots=3, cache_len=64,
+                           gen=GenerateConfig(do_sample=False, eos_token_id=None))
+    futs = [cb.submit(p, max_new_tokens=mn) for p, mn in zip(prompts, budgets)]
+    cb.run_until_idle()
+    got = [f.result(timeout=10) for f in futs]
+    assert got == want
