@@ -537,7 +537,14 @@ class NativeRLTrainer(BaseRLTrainer):
         key = tuple((k, tuple(v.shape), v.dtype) for k, v in sorted(fields.items()))
         entry = self._train_graphs.get(key)
         if entry is None:
-            if len(self._train_graphs) >= 8:
+            # PPO minibatches trim to the minibatch max response width, so as
+            # training shifts response lengths new widths appear; with the
+            # old cap of 8 a 20-step run degraded train from 85 to 108
+            # ms/step once later shapes ran permanently eager.  Width count
+            # is bounded by max_new_tokens; each graph holds ~100 MB of
+            # static+pool at GPT-2 bench scale (288 GB HBM).
+            cap = int(os.environ.get("TRLX_AMD_TRAIN_GRAPH_CAP", "32"))
+            if len(self._train_graphs) >= cap:
                 return None  # too many distinct shapes; stay eager
             static = {k: v.to(self.device).clone() for k, v in fields.items()}
             static_batch = type(microbatch)(**static)
