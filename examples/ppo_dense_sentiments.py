@@ -6,6 +6,9 @@ make_experience."""
 import json
 import sys
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import trlx_amd as trlx
 
 from ppo_sentiments import EVAL_PROMPTS, PROMPTS, default_config, sentiment_reward

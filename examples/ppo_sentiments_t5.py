@@ -7,6 +7,9 @@ Offline adaptation: random-init tiny-T5 + lexicon sentiment reward; point
 import json
 import sys
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import trlx_amd as trlx
 from trlx_amd.data.default_configs import default_ppo_config
 from trlx_amd.models.nn.seq2seq import Seq2SeqConfig

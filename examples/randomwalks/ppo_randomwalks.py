@@ -4,6 +4,9 @@ ppo_randomwalks.py) — offline, random-init model, byte tokenizer."""
 import json
 import sys
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))))
 import trlx_amd as trlx
 from examples.randomwalks import generate_random_walks
 from trlx_amd.data.configs import (

@@ -4,6 +4,9 @@ sampling fine-tuning against the sentiment reward."""
 import json
 import sys
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import trlx_amd as trlx
 from trlx_amd.data.default_configs import default_sft_config
 from trlx_amd.models.nn.config import preset

@@ -22,6 +22,9 @@ from typing import Optional, Union
 import torch
 from torch import nn
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))))
 from trlx_amd.models.modeling_base import load_hf_dir
 from trlx_amd.models.nn.config import TransformerConfig, preset
 from trlx_amd.models.nn.transformer import CausalTransformer
