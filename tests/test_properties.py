@@ -122,3 +122,48 @@ def test_ppo_collate_shapes(lengths):
     # queries left-padded: the LAST q tokens of each row are the original
     for i, (q, r) in enumerate(lengths):
         assert batch.query_tensors[i, qmax - q:].tolist() == (torch.arange(q) + 3).tolist()
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(2, 30), st.floats(0.1, 0.99), st.integers(0, 2**31 - 1))
+def test_top_p_filter_keeps_nucleus(v, top_p, seed):
+    """CPU reference sampling under top-p only ever emits tokens from the
+    nucleus: the smallest prefix of the sorted distribution whose exclusive
+    cumulative mass is <= top_p (HF semantics)."""
+    import torch
+
+    from trlx_amd.ops import reference
+
+    g = torch.Generator().manual_seed(seed)
+    logits = torch.randn(1, v, generator=g) * 3
+    probs = torch.softmax(logits[0], -1)
+    sp, si = torch.sort(probs, descending=True)
+    cum = sp.cumsum(0)
+    keep = {int(si[i]) for i in range(v) if float(cum[i] - sp[i]) <= top_p}
+    for off in range(20):
+        tok = int(reference.sample_token(logits, 1.0, 0, top_p,
+                                         generator=torch.Generator().manual_seed(off)))
+        assert tok in keep, (tok, keep, top_p)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 6), st.integers(2, 12), st.integers(3, 40), st.integers(1, 8),
+       st.integers(0, 2**31 - 1))
+def test_batched_index_select_matches_expanded_gather(b, t, h, na, seed):
+    """The advanced-indexing fast path equals the reference expanded-index
+    gather for any shape, forward and backward."""
+    import torch
+
+    from trlx_amd.models.modeling_ilql import batched_index_select
+
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(b, t, h, generator=g, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    idx = torch.randint(0, t, (b, na), generator=g)
+    fast = batched_index_select(x, idx)
+    ref = x2.gather(1, idx.unsqueeze(-1).expand(b, na, h))
+    assert torch.equal(fast, ref)
+    go = torch.randn(b, na, h, generator=g)
+    fast.backward(go)
+    ref.backward(go)
+    assert torch.equal(x.grad, x2.grad)
